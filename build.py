@@ -1,0 +1,107 @@
+#!/usr/bin/env python3
+"""In-tree build of the gats_amd._core extension with hipcc (gfx950).
+
+hipcc cross-compiles on CPU-only machines; the resulting .so is committed to
+the working tree (git-ignored) and travels with gpurun snapshots.
+"""
+import concurrent.futures as cf
+import os
+import subprocess
+import sys
+import sysconfig
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+SRC = os.path.join(ROOT, "src")
+OBJ = os.path.join(ROOT, "build", "obj")
+HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+ARCH = os.environ.get("GATS_GPU_ARCH", "gfx950")
+
+SOURCES = [
+    "taillard.cpp",
+    "bounds.cpp",
+    "search_host.cpp",
+    "kernels.hip",
+    "engine_gpu.cpp",
+    "engine_multi.cpp",
+    "bindings.cpp",
+]
+
+
+def _includes():
+    import pybind11
+
+    return [
+        sysconfig.get_path("include"),
+        pybind11.get_include(),
+        SRC,
+    ]
+
+
+def _flags():
+    f = [
+        "-O3",
+        "-std=c++17",
+        "-fPIC",
+        f"--offload-arch={ARCH}",
+        "-ffast-math",
+        "-fvisibility=hidden",
+        "-Wno-unused-result",
+    ]
+    for inc in _includes():
+        f.append(f"-I{inc}")
+    return f
+
+
+def _needs_build(obj, src, headers_mtime):
+    if not os.path.exists(obj):
+        return True
+    om = os.path.getmtime(obj)
+    return om < os.path.getmtime(src) or om < headers_mtime
+
+
+def _compile_one(src_name, headers_mtime, flags):
+    src = os.path.join(SRC, src_name)
+    obj = os.path.join(OBJ, src_name.replace("/", "_") + ".o")
+    if not _needs_build(obj, src, headers_mtime):
+        return obj, False
+    cmd = [HIPCC, *flags, "-x", "hip", "-c", src, "-o", obj]
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    if r.returncode != 0:
+        raise RuntimeError(f"hipcc failed for {src_name}:\n{r.stdout}\n{r.stderr}")
+    return obj, True
+
+
+def ext_path():
+    suffix = sysconfig.get_config_var("EXT_SUFFIX")
+    return os.path.join(ROOT, "gats_amd", f"_core{suffix}")
+
+
+def build(verbose=True):
+    os.makedirs(OBJ, exist_ok=True)
+    headers = [os.path.join(SRC, h) for h in os.listdir(SRC) if h.endswith(".hpp")]
+    headers_mtime = max(os.path.getmtime(h) for h in headers)
+    flags = _flags()
+    objs = []
+    rebuilt = False
+    with cf.ThreadPoolExecutor(max_workers=os.cpu_count()) as ex:
+        futs = {ex.submit(_compile_one, s, headers_mtime, flags): s for s in SOURCES}
+        for fut in cf.as_completed(futs):
+            obj, did = fut.result()
+            objs.append(obj)
+            rebuilt |= did
+            if verbose and did:
+                print(f"  [hipcc] {futs[fut]}")
+    out = ext_path()
+    if rebuilt or not os.path.exists(out):
+        cmd = [HIPCC, f"--offload-arch={ARCH}", "-shared", "-fPIC", *sorted(objs), "-o", out]
+        r = subprocess.run(cmd, capture_output=True, text=True)
+        if r.returncode != 0:
+            raise RuntimeError(f"hipcc link failed:\n{r.stdout}\n{r.stderr}")
+        if verbose:
+            print(f"  [link] {os.path.relpath(out, ROOT)}")
+    return out
+
+
+if __name__ == "__main__":
+    build()
+    print("build OK")

@@ -1,0 +1,290 @@
+// pybind11 bindings for the gats_amd core (built directly with hipcc; no
+// libtorch dependency — torch is only used by the Python distributed tier).
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstring>
+#include <stdexcept>
+
+#include "engine_gpu.hpp"
+#include "nodes.hpp"
+#include "pool.hpp"
+#include "search_host.hpp"
+#include "taillard.hpp"
+
+namespace py = pybind11;
+using namespace gats;
+
+namespace {
+
+py::dict result_to_dict(const Result& r) {
+  py::dict d;
+  d["tree"] = r.tree;
+  d["sol"] = r.sol;
+  d["optimum"] = r.optimum;
+  d["time"] = r.time;
+  py::list phases;
+  for (const auto& p : r.phases) {
+    py::dict pd;
+    pd["tree"] = p.tree;
+    pd["sol"] = p.sol;
+    pd["time"] = p.time;
+    phases.append(pd);
+  }
+  d["phases"] = phases;
+  py::dict diag;
+  diag["kernel_launch"] = r.kernel_launch;
+  diag["host_to_device"] = r.h2d;
+  diag["device_to_host"] = r.d2h;
+  diag["h2d_bytes"] = r.h2d_bytes;
+  diag["d2h_bytes"] = r.d2h_bytes;
+  diag["gpu_iters"] = r.gpu_iters;
+  diag["gpu_time"] = r.gpu_time;
+  d["diag"] = diag;
+  return d;
+}
+
+template <typename NodeT>
+std::vector<NodeT> nodes_from_bytes(const py::bytes& b) {
+  std::string s = b;  // copy; fine for frontier-scale data
+  if (s.size() % sizeof(NodeT) != 0)
+    throw std::invalid_argument("node buffer size not a multiple of 24");
+  std::vector<NodeT> v(s.size() / sizeof(NodeT));
+  std::memcpy(v.data(), s.data(), s.size());
+  return v;
+}
+
+template <typename NodeT>
+py::bytes nodes_to_bytes(const NodeT* p, size_t n) {
+  return py::bytes(reinterpret_cast<const char*>(p), n * sizeof(NodeT));
+}
+
+// CPU oracle twins of the GPU eval kernels (same output layout) for numerics
+// tests: GPU labels/bounds must match these exactly.
+py::bytes nq_cpu_labels(int N, int g, const py::bytes& pool_bytes) {
+  auto nodes = nodes_from_bytes<NQNode>(pool_bytes);
+  std::vector<uint8_t> labels(nodes.size() * N, 255);
+  for (size_t i = 0; i < nodes.size(); i++) {
+    const NQNode& p = nodes[i];
+    if (p.depth >= N) continue;
+    for (int k = p.depth; k < N; k++)
+      labels[i * N + k] = nq_is_safe(p.board, p.depth, p.board[k], g) ? 1 : 0;
+  }
+  return py::bytes(reinterpret_cast<const char*>(labels.data()), labels.size());
+}
+
+std::vector<int32_t> pfsp_cpu_bounds(int inst, const std::string& lb_str,
+                                     const py::bytes& pool_bytes, int best) {
+  const LbKind lb = lb_from_string(lb_str);
+  PfspInstance I = make_pfsp_instance(inst, 1);
+  auto nodes = nodes_from_bytes<PFSPNode>(pool_bytes);
+  std::vector<int32_t> bounds(nodes.size() * I.jobs, -1);
+  for (size_t i = 0; i < nodes.size(); i++) {
+    const PFSPNode& p = nodes[i];
+    if (lb == LbKind::LB1_D) {
+      int lb_begin[MAX_JOBS];
+      lb1_children_bounds(I.lb1, p.prmu, p.limit1, I.jobs, lb_begin);
+      for (int k = p.limit1 + 1; k < I.jobs; k++)
+        bounds[i * I.jobs + k] = lb_begin[p.prmu[k]];
+    } else {
+      for (int k = p.limit1 + 1; k < I.jobs; k++) {
+        PFSPNode child = p;
+        child.depth = static_cast<int8_t>(p.depth + 1);
+        child.limit1 = static_cast<int8_t>(p.limit1 + 1);
+        child.prmu[p.depth] = p.prmu[k];
+        child.prmu[k] = p.prmu[p.depth];
+        bounds[i * I.jobs + k] =
+            (lb == LbKind::LB1)
+                ? lb1_bound(I.lb1, child.prmu, child.limit1, I.jobs)
+                : lb2_bound(I.lb1, I.lb2, child.prmu, child.limit1, I.jobs, best);
+      }
+    }
+  }
+  return bounds;
+}
+
+py::tuple nq_bfs_frontier(int N, int g, size_t target) {
+  Pool<NQNode> pool;
+  pool.pushBack(nq_root());
+  uint64_t tree = 0, sol = 0;
+  nq_bfs_until(N, g, target, pool, tree, sol);
+  return py::make_tuple(nodes_to_bytes(pool.data(), pool.size()), tree, sol);
+}
+
+py::tuple pfsp_bfs_frontier(int inst, const std::string& lb_str, int ub, size_t target) {
+  const LbKind lb = lb_from_string(lb_str);
+  PfspInstance I = make_pfsp_instance(inst, ub);
+  Pool<PFSPNode> pool;
+  pool.pushBack(pfsp_root());
+  uint64_t tree = 0, sol = 0;
+  int best = I.init_ub;
+  pfsp_bfs_until(I, lb, target, pool, tree, sol, best);
+  return py::make_tuple(nodes_to_bytes(pool.data(), pool.size()), tree, sol, best);
+}
+
+// Exercises pool push/pop/bulk semantics from C++ (unit-test helper).
+py::dict pool_selftest() {
+  py::dict d;
+  Pool<NQNode> p;
+  NQNode n = nq_root();
+  for (int i = 0; i < 3000; i++) {
+    n.depth = static_cast<uint8_t>(i % 20);
+    p.pushBack(n);
+  }
+  d["size_after_push"] = p.size();
+  NQNode out;
+  bool ok = p.popFront(out);
+  d["pop_front_ok"] = ok;
+  d["front_depth"] = static_cast<int>(out.depth);
+  ok = p.popBack(out);
+  d["pop_back_ok"] = ok;
+  d["back_depth"] = static_cast<int>(out.depth);
+  std::vector<NQNode> buf(5000);
+  d["bulk_below_m"] = p.popBackBulk(100000, 5000, buf.data());  // size < m -> 0
+  d["bulk"] = p.popBackBulk(10, 500, buf.data());               // -> 500
+  d["size_final"] = p.size();
+  return d;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_core, mod) {
+  mod.doc() = "MI355X-native tree-search core (N-Queens + PFSP B&B)";
+
+  mod.def("taillard_nb_jobs", &taillard_nb_jobs);
+  mod.def("taillard_nb_machines", &taillard_nb_machines);
+  mod.def("taillard_best_ub", &taillard_best_ub);
+  mod.def("taillard_processing_times", &taillard_processing_times);
+
+  mod.def("nqueens_seq",
+          [](int N, int g) {
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = nqueens_seq(N, g);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("N") = 14, py::arg("g") = 1);
+  mod.def("pfsp_seq",
+          [](int inst, const std::string& lb, int ub) {
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = pfsp_seq(inst, lb, ub);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("inst") = 14, py::arg("lb") = "lb1", py::arg("ub") = 1);
+
+  mod.def("gpu_device_count", &gpu_device_count);
+
+  mod.def("nqueens_multigpu",
+          [](int N, int g, int m, int M, int D, const std::string& eval) {
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = nqueens_multigpu(N, g, m, M, D, eval);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("N") = 14, py::arg("g") = 1, py::arg("m") = 25, py::arg("M") = 50000,
+          py::arg("D") = 1, py::arg("eval") = "gpu");
+  mod.def("pfsp_multigpu",
+          [](int inst, const std::string& lb, int ub, int m, int M, int D,
+             const std::string& eval, bool share_best) {
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = pfsp_multigpu(inst, lb, ub, m, M, D, eval, share_best);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("inst") = 14, py::arg("lb") = "lb1", py::arg("ub") = 1, py::arg("m") = 25,
+          py::arg("M") = 50000, py::arg("D") = 1, py::arg("eval") = "gpu",
+          py::arg("share_best") = false);
+
+  mod.def("nqueens_gpu",
+          [](int N, int g, int m, int M, int device, const std::string& mode,
+             unsigned long long capacity) {
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = nqueens_gpu(N, g, m, M, device, mode, capacity);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("N") = 14, py::arg("g") = 1, py::arg("m") = 25, py::arg("M") = 50000,
+          py::arg("device") = 0, py::arg("mode") = "devpool",
+          py::arg("capacity") = (1ull << 27));
+
+  mod.def("pfsp_gpu",
+          [](int inst, const std::string& lb, int ub, int m, int M, int device,
+             const std::string& mode, unsigned long long capacity) {
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = pfsp_gpu(inst, lb, ub, m, M, device, mode, capacity);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("inst") = 14, py::arg("lb") = "lb1", py::arg("ub") = 1, py::arg("m") = 25,
+          py::arg("M") = 50000, py::arg("device") = 0, py::arg("mode") = "devpool",
+          py::arg("capacity") = (1ull << 27));
+
+  mod.def("nqueens_gpu_from_pool",
+          [](const py::bytes& nodes, int N, int g, int m, int M, int device,
+             const std::string& mode, unsigned long long capacity) {
+            auto v = nodes_from_bytes<NQNode>(nodes);
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = nqueens_gpu_from_pool(v, N, g, m, M, device, mode, capacity);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("nodes"), py::arg("N"), py::arg("g") = 1, py::arg("m") = 25,
+          py::arg("M") = 50000, py::arg("device") = 0, py::arg("mode") = "devpool",
+          py::arg("capacity") = (1ull << 27));
+
+  mod.def("pfsp_gpu_from_pool",
+          [](const py::bytes& nodes, int inst, const std::string& lb, int ub, int best0,
+             int m, int M, int device, const std::string& mode,
+             unsigned long long capacity) {
+            auto v = nodes_from_bytes<PFSPNode>(nodes);
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = pfsp_gpu_from_pool(v, inst, lb, ub, best0, m, M, device, mode, capacity);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("nodes"), py::arg("inst"), py::arg("lb") = "lb1", py::arg("ub") = 1,
+          py::arg("best0") = 0, py::arg("m") = 25, py::arg("M") = 50000,
+          py::arg("device") = 0, py::arg("mode") = "devpool",
+          py::arg("capacity") = (1ull << 27));
+
+  mod.def("nq_bfs_frontier", &nq_bfs_frontier, py::arg("N"), py::arg("g") = 1,
+          py::arg("target") = 1024);
+  mod.def("pfsp_bfs_frontier", &pfsp_bfs_frontier, py::arg("inst"), py::arg("lb") = "lb1",
+          py::arg("ub") = 1, py::arg("target") = 1024);
+
+  mod.def("nq_cpu_labels", &nq_cpu_labels);
+  mod.def("pfsp_cpu_bounds", &pfsp_cpu_bounds);
+  mod.def("nq_gpu_labels",
+          [](int N, int g, const py::bytes& nodes, int device) {
+            auto v = nodes_from_bytes<NQNode>(nodes);
+            auto labels = nq_gpu_labels(N, g, v, device);
+            return py::bytes(reinterpret_cast<const char*>(labels.data()), labels.size());
+          },
+          py::arg("N"), py::arg("g"), py::arg("nodes"), py::arg("device") = 0);
+  mod.def("pfsp_gpu_bounds",
+          [](int inst, const std::string& lb, const py::bytes& nodes, int best, int device) {
+            auto v = nodes_from_bytes<PFSPNode>(nodes);
+            return pfsp_gpu_bounds(inst, lb, v, best, device);
+          },
+          py::arg("inst"), py::arg("lb"), py::arg("nodes"), py::arg("best"),
+          py::arg("device") = 0);
+
+  mod.def("pool_selftest", &pool_selftest);
+}

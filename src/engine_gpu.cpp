@@ -1,0 +1,432 @@
+// Single-GPU 3-phase search engines for MI355X.
+//
+// Phase structure parity: reference pfsp_gpu_chpl.chpl:306-431 /
+// nqueens_gpu_chpl.chpl:152-248 (phase 1 CPU BFS until size >= m; phase 2
+// chunked m/M offload; phase 3 CPU DFS drain).
+//
+// Two phase-2 modes:
+//   "hostpool": the reference's architecture, done right for MI355X — pinned
+//       host buffers, async prefix-only copies (the Chapel version copies the
+//       full M-element arrays every iteration, SURVEY.md §8.4 — we don't),
+//       one HIP stream, host-side generate_children.
+//   "devpool": MI355X-native fast path — the pool lives in HBM3E, each
+//       iteration is a begin/copy/expand kernel triple with on-device pruning
+//       and appends; the host polls a 64 B control block every few iterations.
+#include <hip/hip_runtime.h>
+
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "gpu_api.hpp"
+#include "search_host.hpp"
+
+namespace gats {
+
+#define HIP_CHECK(expr)                                                              \
+  do {                                                                               \
+    hipError_t _e = (expr);                                                          \
+    if (_e != hipSuccess)                                                            \
+      throw std::runtime_error(std::string("HIP error: ") + hipGetErrorString(_e) +  \
+                               " at " #expr);                                        \
+  } while (0)
+
+int gpu_device_count() {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess) return 0;
+  return n;
+}
+
+namespace {
+
+template <typename T>
+T* dev_alloc(size_t n) {
+  void* p = nullptr;
+  HIP_CHECK(hipMalloc(&p, n * sizeof(T)));
+  return static_cast<T*>(p);
+}
+
+template <typename T>
+T* dev_upload(const T* src, size_t n) {
+  T* d = dev_alloc<T>(n);
+  HIP_CHECK(hipMemcpy(d, src, n * sizeof(T), hipMemcpyHostToDevice));
+  return d;
+}
+
+// Owns the int16/uint8-compressed PFSP bound tables on device.
+struct PfspTablesGuard {
+  PfspDevTables tb{};
+  std::vector<void*> allocs;
+
+  PfspTablesGuard(const PfspInstance& I) {
+    const int n = I.jobs, m = I.machines;
+    const int pairs = I.lb2.nb_pairs;
+    std::vector<int16_t> p16(static_cast<size_t>(m) * n);
+    for (size_t i = 0; i < p16.size(); i++) p16[i] = static_cast<int16_t>(I.lb1.p_times[i]);
+    std::vector<int32_t> mt(I.lb1.min_tails.begin(), I.lb1.min_tails.end());
+    std::vector<int16_t> lags16(static_cast<size_t>(pairs) * n);
+    std::vector<uint8_t> js8(static_cast<size_t>(pairs) * n);
+    for (size_t i = 0; i < lags16.size(); i++) {
+      lags16[i] = static_cast<int16_t>(I.lb2.lags[i]);
+      js8[i] = static_cast<uint8_t>(I.lb2.johnson_schedules[i]);
+    }
+    std::vector<uint8_t> p1(pairs), p2(pairs);
+    for (int i = 0; i < pairs; i++) {
+      p1[i] = static_cast<uint8_t>(I.lb2.pairs1[i]);
+      p2[i] = static_cast<uint8_t>(I.lb2.pairs2[i]);
+    }
+    tb.p_times = keep(dev_upload(p16.data(), p16.size()));
+    tb.min_tails = keep(dev_upload(mt.data(), mt.size()));
+    tb.lags = keep(dev_upload(lags16.data(), lags16.size()));
+    tb.johnson_schedules = keep(dev_upload(js8.data(), js8.size()));
+    tb.pairs1 = keep(dev_upload(p1.data(), p1.size()));
+    tb.pairs2 = keep(dev_upload(p2.data(), p2.size()));
+  }
+  template <typename T>
+  T* keep(T* p) {
+    allocs.push_back(p);
+    return p;
+  }
+  ~PfspTablesGuard() {
+    for (void* p : allocs) (void)hipFree(p);
+  }
+};
+
+struct StreamGuard {
+  hipStream_t s{};
+  StreamGuard() { HIP_CHECK(hipStreamCreate(&s)); }
+  ~StreamGuard() { (void)hipStreamDestroy(s); }
+};
+
+template <typename T>
+struct DevGuard {
+  T* p = nullptr;
+  explicit DevGuard(size_t n) { p = dev_alloc<T>(n); }
+  ~DevGuard() { (void)hipFree(p); }
+};
+
+template <typename T>
+struct PinnedGuard {
+  T* p = nullptr;
+  explicit PinnedGuard(size_t n) {
+    void* q = nullptr;
+    HIP_CHECK(hipHostMalloc(&q, n * sizeof(T)));
+    p = static_cast<T*>(q);
+  }
+  ~PinnedGuard() { (void)hipHostFree(p); }
+};
+
+int lbk_of(LbKind lb) {
+  switch (lb) {
+    case LbKind::LB1_D:
+      return 0;
+    case LbKind::LB1:
+      return 1;
+    default:
+      return 2;
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// N-Queens
+// ---------------------------------------------------------------------------
+
+// Runs phases 2+3 given a phase-1 pool; shared by the CLI engine and the
+// distributed tier (which builds the frontier itself and slices it).
+Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int device,
+                       const std::string& mode, uint64_t tree0, uint64_t sol0,
+                       double phase1_time, unsigned long long capacity) {
+  Result r;
+  r.phases.push_back({tree0, sol0, phase1_time});
+  uint64_t tree = tree0, sol = sol0;
+
+  HIP_CHECK(hipSetDevice(device));
+  StreamGuard stream;
+  const double t2 = now_sec();
+
+  if (mode == "hostpool") {
+    PinnedGuard<NQNode> parents(M);
+    PinnedGuard<uint8_t> labels(static_cast<size_t>(M) * N);
+    DevGuard<NQNode> parents_d(M);
+    DevGuard<uint8_t> labels_d(static_cast<size_t>(M) * N);
+    while (true) {
+      const size_t n = pool.popBackBulk(m, M, parents.p);
+      if (n == 0) break;
+      HIP_CHECK(hipMemcpyAsync(parents_d.p, parents.p, n * sizeof(NQNode),
+                               hipMemcpyHostToDevice, stream.s));
+      launch_nq_eval(parents_d.p, static_cast<int>(n), N, g, labels_d.p, stream.s);
+      HIP_CHECK(hipMemcpyAsync(labels.p, labels_d.p, n * N, hipMemcpyDeviceToHost, stream.s));
+      HIP_CHECK(hipStreamSynchronize(stream.s));
+      r.kernel_launch++;
+      r.h2d++;
+      r.d2h++;
+      r.h2d_bytes += n * sizeof(NQNode);
+      r.d2h_bytes += n * N;
+      r.gpu_iters++;
+      nq_generate_children(parents.p, n, N, labels.p, tree, sol, pool);
+    }
+  } else if (mode == "devpool") {
+    DevGuard<NQNode> pool_d(capacity);
+    DevGuard<NQNode> parents_d(M);
+    DevGuard<DevCtl> ctl_d(1);
+    const size_t init = pool.size();
+    if (init > capacity) throw std::runtime_error("devpool capacity too small");
+    HIP_CHECK(hipMemcpy(pool_d.p, pool.data(), init * sizeof(NQNode), hipMemcpyHostToDevice));
+    DevCtl ctl{};
+    ctl.size = init;
+    ctl.tree = tree;
+    ctl.sol = sol;
+    HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    r.h2d += 2;
+    r.h2d_bytes += init * sizeof(NQNode) + sizeof(DevCtl);
+
+    PinnedGuard<DevCtl> ctl_h(1);
+    const int BATCH = 8;
+    while (true) {
+      for (int b = 0; b < BATCH; b++) {
+        launch_begin(ctl_d.p, m, M, stream.s);
+        launch_copy_parents_nq(ctl_d.p, pool_d.p, parents_d.p, M, stream.s);
+        launch_nq_expand(ctl_d.p, parents_d.p, pool_d.p, capacity, M, N, g, stream.s);
+        r.kernel_launch += 3;
+      }
+      HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d.p, sizeof(DevCtl), hipMemcpyDeviceToHost,
+                               stream.s));
+      HIP_CHECK(hipStreamSynchronize(stream.s));
+      r.d2h++;
+      r.d2h_bytes += sizeof(DevCtl);
+      if (ctl_h.p->overflow) throw std::runtime_error("device pool overflow; raise capacity");
+      if (ctl_h.p->size < static_cast<unsigned long long>(m)) break;
+    }
+    tree = ctl_h.p->tree;
+    sol = ctl_h.p->sol;
+    r.gpu_iters = ctl_h.p->iters;
+    // bring the leftover (< m nodes) back for phase 3
+    pool.clear();
+    const size_t left = ctl_h.p->size;
+    if (left > 0) {
+      std::vector<NQNode> tmp(left);
+      HIP_CHECK(hipMemcpy(tmp.data(), pool_d.p, left * sizeof(NQNode), hipMemcpyDeviceToHost));
+      pool.pushBackBulk(tmp.data(), left);
+      r.d2h++;
+      r.d2h_bytes += left * sizeof(NQNode);
+    }
+  } else {
+    throw std::invalid_argument("mode must be hostpool or devpool");
+  }
+
+  const double t3 = now_sec();
+  r.gpu_time = t3 - t2;
+  r.phases.push_back({tree - tree0, sol - sol0, t3 - t2});
+
+  // Phase 3: CPU DFS drain (nqueens_gpu_chpl.chpl:226-245).
+  uint64_t tree_p2 = tree, sol_p2 = sol;
+  NQNode parent;
+  while (pool.popBack(parent)) nq_decompose(parent, N, g, tree, sol, pool);
+  const double t4 = now_sec();
+  r.phases.push_back({tree - tree_p2, sol - sol_p2, t4 - t3});
+
+  r.tree = tree;
+  r.sol = sol;
+  r.time = phase1_time + (t4 - t2);
+  return r;
+}
+
+Result nqueens_gpu(int N, int g, int m, int M, int device, const std::string& mode,
+                   unsigned long long capacity) {
+  Pool<NQNode> pool;
+  pool.pushBack(nq_root());
+  uint64_t tree = 0, sol = 0;
+  const double t0 = now_sec();
+  nq_bfs_until(N, g, static_cast<size_t>(m), pool, tree, sol);
+  const double p1 = now_sec() - t0;
+  return nqueens_gpu_run(pool, N, g, m, M, device, mode, tree, sol, p1, capacity);
+}
+
+Result nqueens_gpu_from_pool(const std::vector<NQNode>& nodes, int N, int g, int m, int M,
+                             int device, const std::string& mode,
+                             unsigned long long capacity) {
+  Pool<NQNode> pool;
+  pool.pushBackBulk(nodes.data(), nodes.size());
+  return nqueens_gpu_run(pool, N, g, m, M, device, mode, 0, 0, 0.0, capacity);
+}
+
+// ---------------------------------------------------------------------------
+// PFSP
+// ---------------------------------------------------------------------------
+
+Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int m, int M,
+                    int device, const std::string& mode, uint64_t tree0, uint64_t sol0,
+                    int best0, double phase1_time, unsigned long long capacity) {
+  Result r;
+  r.phases.push_back({tree0, sol0, phase1_time});
+  uint64_t tree = tree0, sol = sol0;
+  int best = best0;
+  const int jobs = I.jobs, machines = I.machines;
+  const int lbk = lbk_of(lb);
+
+  HIP_CHECK(hipSetDevice(device));
+  StreamGuard stream;
+  PfspTablesGuard tables(I);
+  const double t2 = now_sec();
+
+  if (mode == "hostpool") {
+    PinnedGuard<PFSPNode> parents(M);
+    PinnedGuard<int32_t> bounds(static_cast<size_t>(M) * jobs);
+    DevGuard<PFSPNode> parents_d(M);
+    DevGuard<int32_t> bounds_d(static_cast<size_t>(M) * jobs);
+    while (true) {
+      const size_t n = pool.popBackBulk(m, M, parents.p);
+      if (n == 0) break;
+      HIP_CHECK(hipMemcpyAsync(parents_d.p, parents.p, n * sizeof(PFSPNode),
+                               hipMemcpyHostToDevice, stream.s));
+      launch_pfsp_eval(parents_d.p, static_cast<int>(n), jobs, machines, lbk, tables.tb, best,
+                       bounds_d.p, stream.s);
+      HIP_CHECK(hipMemcpyAsync(bounds.p, bounds_d.p, n * jobs * sizeof(int32_t),
+                               hipMemcpyDeviceToHost, stream.s));
+      HIP_CHECK(hipStreamSynchronize(stream.s));
+      r.kernel_launch++;
+      r.h2d++;
+      r.d2h++;
+      r.h2d_bytes += n * sizeof(PFSPNode);
+      r.d2h_bytes += n * jobs * sizeof(int32_t);
+      r.gpu_iters++;
+      pfsp_generate_children(I, parents.p, n, bounds.p, tree, sol, best, pool);
+    }
+  } else if (mode == "devpool") {
+    DevGuard<PFSPNode> pool_d(capacity);
+    DevGuard<PFSPNode> parents_d(M);
+    DevGuard<DevCtl> ctl_d(1);
+    const size_t init = pool.size();
+    if (init > capacity) throw std::runtime_error("devpool capacity too small");
+    HIP_CHECK(
+        hipMemcpy(pool_d.p, pool.data(), init * sizeof(PFSPNode), hipMemcpyHostToDevice));
+    DevCtl ctl{};
+    ctl.size = init;
+    ctl.tree = tree;
+    ctl.sol = sol;
+    ctl.best = best;
+    HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    r.h2d += 2;
+    r.h2d_bytes += init * sizeof(PFSPNode) + sizeof(DevCtl);
+
+    PinnedGuard<DevCtl> ctl_h(1);
+    const int BATCH = 8;
+    while (true) {
+      for (int b = 0; b < BATCH; b++) {
+        launch_begin(ctl_d.p, m, M, stream.s);
+        launch_copy_parents_pfsp(ctl_d.p, pool_d.p, parents_d.p, M, stream.s);
+        launch_pfsp_expand(ctl_d.p, parents_d.p, pool_d.p, capacity, M, jobs, machines, lbk,
+                           tables.tb, stream.s);
+        r.kernel_launch += 3;
+      }
+      HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d.p, sizeof(DevCtl), hipMemcpyDeviceToHost,
+                               stream.s));
+      HIP_CHECK(hipStreamSynchronize(stream.s));
+      r.d2h++;
+      r.d2h_bytes += sizeof(DevCtl);
+      if (ctl_h.p->overflow) throw std::runtime_error("device pool overflow; raise capacity");
+      if (ctl_h.p->size < static_cast<unsigned long long>(m)) break;
+    }
+    tree = ctl_h.p->tree;
+    sol = ctl_h.p->sol;
+    best = ctl_h.p->best;
+    r.gpu_iters = ctl_h.p->iters;
+    pool.clear();
+    const size_t left = ctl_h.p->size;
+    if (left > 0) {
+      std::vector<PFSPNode> tmp(left);
+      HIP_CHECK(
+          hipMemcpy(tmp.data(), pool_d.p, left * sizeof(PFSPNode), hipMemcpyDeviceToHost));
+      pool.pushBackBulk(tmp.data(), left);
+      r.d2h++;
+      r.d2h_bytes += left * sizeof(PFSPNode);
+    }
+  } else {
+    throw std::invalid_argument("mode must be hostpool or devpool");
+  }
+
+  const double t3 = now_sec();
+  r.gpu_time = t3 - t2;
+  r.phases.push_back({tree - tree0, sol - sol0, t3 - t2});
+
+  uint64_t tree_p2 = tree, sol_p2 = sol;
+  PFSPNode parent;
+  while (pool.popBack(parent)) pfsp_decompose(I, lb, parent, tree, sol, best, pool);
+  const double t4 = now_sec();
+  r.phases.push_back({tree - tree_p2, sol - sol_p2, t4 - t3});
+
+  r.tree = tree;
+  r.sol = sol;
+  r.optimum = best;
+  r.time = phase1_time + (t4 - t2);
+  return r;
+}
+
+Result pfsp_gpu(int inst, const std::string& lb_str, int ub, int m, int M, int device,
+                const std::string& mode, unsigned long long capacity) {
+  const LbKind lb = lb_from_string(lb_str);
+  PfspInstance I = make_pfsp_instance(inst, ub);
+  Pool<PFSPNode> pool;
+  pool.pushBack(pfsp_root());
+  uint64_t tree = 0, sol = 0;
+  int best = I.init_ub;
+  const double t0 = now_sec();
+  pfsp_bfs_until(I, lb, static_cast<size_t>(m), pool, tree, sol, best);
+  const double p1 = now_sec() - t0;
+  return pfsp_gpu_run(I, lb, pool, m, M, device, mode, tree, sol, best, p1, capacity);
+}
+
+Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst,
+                          const std::string& lb_str, int ub, int best0, int m, int M,
+                          int device, const std::string& mode, unsigned long long capacity) {
+  const LbKind lb = lb_from_string(lb_str);
+  PfspInstance I = make_pfsp_instance(inst, ub);
+  Pool<PFSPNode> pool;
+  pool.pushBackBulk(nodes.data(), nodes.size());
+  const int best = (best0 > 0) ? best0 : I.init_ub;
+  return pfsp_gpu_run(I, lb, pool, m, M, device, mode, 0, 0, best, 0.0, capacity);
+}
+
+// ---------------------------------------------------------------------------
+// Eval-only entry points (HIP-kernel-vs-CPU-oracle numerics tests)
+// ---------------------------------------------------------------------------
+
+std::vector<uint8_t> nq_gpu_labels(int N, int g, const std::vector<NQNode>& nodes,
+                                   int device) {
+  HIP_CHECK(hipSetDevice(device));
+  const size_t n = nodes.size();
+  std::vector<uint8_t> labels(n * N, 255);
+  DevGuard<NQNode> parents_d(n);
+  DevGuard<uint8_t> labels_d(n * N);
+  HIP_CHECK(hipMemcpy(parents_d.p, nodes.data(), n * sizeof(NQNode), hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemset(labels_d.p, 255, n * N));
+  launch_nq_eval(parents_d.p, static_cast<int>(n), N, g, labels_d.p, nullptr);
+  HIP_CHECK(hipMemcpy(labels.data(), labels_d.p, n * N, hipMemcpyDeviceToHost));
+  return labels;
+}
+
+std::vector<int32_t> pfsp_gpu_bounds(int inst, const std::string& lb_str,
+                                     const std::vector<PFSPNode>& nodes, int best,
+                                     int device) {
+  const LbKind lb = lb_from_string(lb_str);
+  PfspInstance I = make_pfsp_instance(inst, 1);
+  HIP_CHECK(hipSetDevice(device));
+  PfspTablesGuard tables(I);
+  const size_t n = nodes.size();
+  std::vector<int32_t> bounds(n * I.jobs, -1);
+  DevGuard<PFSPNode> parents_d(n);
+  DevGuard<int32_t> bounds_d(n * I.jobs);
+  HIP_CHECK(hipMemcpy(parents_d.p, nodes.data(), n * sizeof(PFSPNode), hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemset(bounds_d.p, 255, n * I.jobs * sizeof(int32_t)));
+  launch_pfsp_eval(parents_d.p, static_cast<int>(n), I.jobs, I.machines, lbk_of(lb),
+                   tables.tb, best, bounds_d.p, nullptr);
+  HIP_CHECK(hipMemcpy(bounds.data(), bounds_d.p, n * I.jobs * sizeof(int32_t),
+                      hipMemcpyDeviceToHost));
+  return bounds;
+}
+
+}  // namespace gats

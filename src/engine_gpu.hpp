@@ -1,0 +1,36 @@
+// Public API of the single-GPU engines (implemented in engine_gpu.cpp).
+#pragma once
+#include <string>
+#include <vector>
+
+#include "nodes.hpp"
+#include "search_host.hpp"
+
+namespace gats {
+
+int gpu_device_count();
+
+Result nqueens_gpu(int N, int g, int m, int M, int device, const std::string& mode,
+                   unsigned long long capacity);
+Result nqueens_gpu_from_pool(const std::vector<NQNode>& nodes, int N, int g, int m, int M,
+                             int device, const std::string& mode,
+                             unsigned long long capacity);
+
+Result pfsp_gpu(int inst, const std::string& lb, int ub, int m, int M, int device,
+                const std::string& mode, unsigned long long capacity);
+Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst, const std::string& lb,
+                          int ub, int best0, int m, int M, int device,
+                          const std::string& mode, unsigned long long capacity);
+
+// engine_multi.cpp: in-process multi-GPU tier (eval = "gpu" or "cpu").
+Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& eval);
+Result pfsp_multigpu(int inst, const std::string& lb, int ub, int m, int M, int D,
+                     const std::string& eval, bool share_best);
+
+std::vector<uint8_t> nq_gpu_labels(int N, int g, const std::vector<NQNode>& nodes,
+                                   int device);
+std::vector<int32_t> pfsp_gpu_bounds(int inst, const std::string& lb,
+                                     const std::vector<PFSPNode>& nodes, int best,
+                                     int device);
+
+}  // namespace gats
