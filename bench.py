@@ -1,0 +1,117 @@
+#!/usr/bin/env python3
+"""Driver benchmark contract: flagship tree-search throughput on N MI355X GPUs.
+
+Headline metric (BASELINE.json): Mnodes/sec (whole node), N-Queens N=17 with
+the reference's default offload window m=25, M=50000. A "step" is one complete
+search of the instance (the non-neural analog of a training pass over a fixed
+input). Strong scaling: the frontier of ONE search is round-robin partitioned
+across ranks (gats_amd.dist), so total work is fixed as N grows.
+
+Launched by the driver as
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+"""
+import argparse
+import json
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch  # noqa: E402
+
+import gats_amd  # noqa: E402
+from gats_amd import dist as gdist  # noqa: E402
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--problem", default="nqueens", choices=["nqueens", "pfsp"])
+    ap.add_argument("--N", type=int, default=17)
+    ap.add_argument("--g", type=int, default=1)
+    ap.add_argument("--inst", type=int, default=14)
+    ap.add_argument("--lb", default="lb1")
+    ap.add_argument("--ub", type=int, default=1)
+    ap.add_argument("--m", type=int, default=25)
+    ap.add_argument("--M", type=int, default=50000)
+    ap.add_argument("--mode", default="devpool")
+    return ap.parse_args()
+
+
+def main():
+    args = parse_args()
+    c = gats_amd.core()
+    rank, world = gdist.init_dist()
+    ndev = c.gpu_device_count()
+    if ndev == 0:
+        raise RuntimeError("bench.py needs an MI355X (no HIP device visible)")
+    local = int(os.environ.get("LOCAL_RANK", rank)) % ndev
+    torch.cuda.set_device(local)
+
+    def step():
+        if args.problem == "nqueens":
+            return gdist.run_nqueens(args.N, args.g, args.m, args.M, args.mode)
+        return gdist.run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode)
+
+    def sync():
+        if world > 1:
+            torch.distributed.barrier()
+        torch.cuda.synchronize()
+
+    nodes_per_step = None
+    for _ in range(args.warmup):
+        r = step()
+        nodes_per_step = r["tree"]
+    if nodes_per_step is None:  # warmup 0
+        r = step()
+        nodes_per_step = r["tree"]
+
+    import time
+
+    sync()
+    start = time.perf_counter()
+    for _ in range(args.steps):
+        r = step()
+    sync()
+    elapsed = time.perf_counter() - start
+
+    # MAX over ranks (contract): reduce elapsed
+    if world > 1:
+        dev = torch.device(f"cuda:{local}")
+        e = torch.tensor([elapsed], dtype=torch.float32, device=dev)
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(e[0].item())
+
+    if rank == 0:
+        total_nodes = nodes_per_step * args.steps
+        value = total_nodes / elapsed / 1e6  # Mnodes/s, whole job
+        if args.problem == "nqueens":
+            cfg = {"model": f"nqueens-N{args.N}", "N": args.N, "g": args.g, "m": args.m,
+                   "M": args.M, "parallelism": f"multipool-dp{world}"}
+        else:
+            cfg = {"model": f"pfsp-ta{args.inst:03d}-{args.lb}", "inst": args.inst,
+                   "lb": args.lb, "ub": args.ub, "m": args.m, "M": args.M,
+                   "parallelism": f"multipool-dp{world}"}
+        print(json.dumps({
+            "metric": "Mnodes_per_sec",
+            "value": value,
+            "unit": "Mnodes/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "int32",
+            "data": "synthetic (deterministic N-Queens board / Taillard generator)",
+            "config": cfg,
+            "explored_tree_per_step": nodes_per_step,
+        }))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,137 @@
+"""Distributed multi-GPU tier: one process per GPU, torch.distributed over
+RCCL (backend "nccl" on ROCm) across the xGMI links of one MI355X node.
+
+Capability parity with the reference's distributed drivers
+(pfsp_dist_multigpu_chpl.chpl / pfsp_dist_multigpu_cuda.c):
+  - every rank runs the deterministic phase-1 BFS redundantly and keeps its
+    round-robin slice — the broadcast-free partition trick of the MPI baseline
+    (pfsp_dist_multigpu_cuda.c:372-378, 416-426)
+  - phase-1 counts are attributed to rank 0 only (no double counting)
+  - each rank runs the single-GPU engine (devpool by default) on its slice
+  - end-of-search collectives: all_reduce SUM on tree/sol, MIN on the PFSP
+    incumbent, MAX on elapsed time (MPI parity: MPI_Reduce calls at
+    pfsp_dist_multigpu_cuda.c:681-694)
+No inter-rank work stealing yet — the reference's own MPI/CUDA baseline has
+none either; the round-robin interleave of a deep frontier
+keeps the slices statistically balanced.
+
+The RCCL collectives are tiny (a few int64 scalars); latency-bound, far from
+any xGMI bandwidth limit, so RCCL defaults are the right choice (SURVEY.md
+§2.4 "MI355X-native equivalent").
+"""
+import os
+
+import torch
+import torch.distributed as td
+
+import gats_amd
+
+NODE_BYTES = 24
+
+
+def init_dist():
+    """Initialize torch.distributed from torchrun env; returns (rank, world)."""
+    if td.is_initialized():
+        return td.get_rank(), td.get_world_size()
+    if "RANK" not in os.environ:
+        return 0, 1
+    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    td.init_process_group(backend=backend)
+    return td.get_rank(), td.get_world_size()
+
+
+def slice_frontier(nodes: bytes, rank: int, world: int) -> bytes:
+    """Round-robin node slice (rank, rank+world, ...), the reference's static
+    interleaved partition (nqueens_dist_multigpu_chpl.chpl:223-227)."""
+    n = len(nodes) // NODE_BYTES
+    out = bytearray()
+    for i in range(rank, n, world):
+        out += nodes[i * NODE_BYTES:(i + 1) * NODE_BYTES]
+    return bytes(out)
+
+
+def _reduce_stats(r, phase1, device, world, best=None):
+    """Global reductions; returns combined stats dict on every rank."""
+    dev = torch.device(device) if torch.cuda.is_available() else torch.device("cpu")
+    t = torch.tensor([r["tree"], r["sol"]], dtype=torch.int64, device=dev)
+    td.all_reduce(t, op=td.ReduceOp.SUM)
+    elapsed = torch.tensor([r["time"]], dtype=torch.float64 if dev.type == "cpu" else torch.float32,
+                           device=dev)
+    td.all_reduce(elapsed, op=td.ReduceOp.MAX)
+    out = {
+        "tree": int(t[0].item()) + phase1["tree"],
+        "sol": int(t[1].item()) + phase1["sol"],
+        "time": float(elapsed[0].item()) + phase1["time"],
+        "phases": [phase1,
+                   {"tree": int(t[0].item()), "sol": int(t[1].item()),
+                    "time": float(elapsed[0].item())}],
+        "diag": r.get("diag", {}),
+        "optimum": r.get("optimum", 0),
+    }
+    if best is not None:
+        b = torch.tensor([r["optimum"]], dtype=torch.int64, device=dev)
+        td.all_reduce(b, op=td.ReduceOp.MIN)
+        out["optimum"] = int(b[0].item())
+    return out
+
+
+def run_nqueens(N, g=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
+                frontier_target=None, engine="gpu"):
+    c = gats_amd.core()
+    rank, world = init_dist()
+    if frontier_target is None:
+        frontier_target = max(65536, 8192 * world)
+    nodes, tree1, sol1 = c.nq_bfs_frontier(N, g, frontier_target)
+    my = slice_frontier(nodes, rank, world)
+    phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
+    if engine == "gpu":
+        local = rank % max(1, c.gpu_device_count())
+        r = c.nqueens_gpu_from_pool(my, N, g, m, M, local, mode, capacity)
+    else:  # CPU path for gloo CI
+        r = c.nqueens_seq_from_pool(my, N, g)
+    if world == 1:
+        r = dict(r)
+        r["tree"] += phase1["tree"]
+        r["sol"] += phase1["sol"]
+        return r
+    phase1["tree"], phase1["sol"] = tree1, sol1  # counted once globally
+    local_dev = f"cuda:{rank % max(1, c.gpu_device_count())}" if engine == "gpu" else "cpu"
+    return _reduce_stats(r, phase1, local_dev, world)
+
+
+def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
+             frontier_target=None, engine="gpu"):
+    c = gats_amd.core()
+    rank, world = init_dist()
+    if frontier_target is None:
+        frontier_target = max(65536, 8192 * world)
+    nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
+    my = slice_frontier(nodes, rank, world)
+    phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
+    if engine == "gpu":
+        local = rank % max(1, c.gpu_device_count())
+        r = c.pfsp_gpu_from_pool(my, inst, lb, ub, best, m, M, local, mode, capacity)
+    else:
+        r = c.pfsp_seq_from_pool(my, inst, lb, ub, best)
+    if world == 1:
+        r = dict(r)
+        r["tree"] += phase1["tree"]
+        r["sol"] += phase1["sol"]
+        return r
+    phase1["tree"], phase1["sol"] = tree1, sol1
+    local_dev = f"cuda:{rank % max(1, c.gpu_device_count())}" if engine == "gpu" else "cpu"
+    return _reduce_stats(r, phase1, local_dev, world, best=True)
+
+
+def run_from_cli(args):
+    """Entry for `gats-amd ... --tier dist` under torchrun; rank 0 returns the
+    combined stats dict, other ranks return None."""
+    rank, world = init_dist()
+    if args.problem == "nqueens":
+        r = run_nqueens(args.N, args.g, args.m, args.M, args.mode, args.capacity)
+    else:
+        r = run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode, args.capacity)
+    if world > 1:
+        td.barrier()
+    return r if rank == 0 else None

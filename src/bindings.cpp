@@ -122,6 +122,45 @@ py::tuple pfsp_bfs_frontier(int inst, const std::string& lb_str, int ub, size_t 
   return py::make_tuple(nodes_to_bytes(pool.data(), pool.size()), tree, sol, best);
 }
 
+// Sequential drain of an explicit frontier (distributed tier's CPU path and
+// the phase-3 semantics oracle).
+py::dict nqueens_seq_from_pool(const py::bytes& nodes, int N, int g) {
+  auto v = nodes_from_bytes<NQNode>(nodes);
+  Result r;
+  {
+    py::gil_scoped_release rel;
+    Pool<NQNode> pool;
+    pool.pushBackBulk(v.data(), v.size());
+    const double t0 = now_sec();
+    NQNode parent;
+    while (pool.popBack(parent)) nq_decompose(parent, N, g, r.tree, r.sol, pool);
+    r.time = now_sec() - t0;
+    r.phases.push_back({r.tree, r.sol, r.time});
+  }
+  return result_to_dict(r);
+}
+
+py::dict pfsp_seq_from_pool(const py::bytes& nodes, int inst, const std::string& lb_str,
+                            int ub, int best0) {
+  auto v = nodes_from_bytes<PFSPNode>(nodes);
+  const LbKind lb = lb_from_string(lb_str);
+  Result r;
+  {
+    py::gil_scoped_release rel;
+    PfspInstance I = make_pfsp_instance(inst, ub);
+    int best = (best0 > 0) ? best0 : I.init_ub;
+    Pool<PFSPNode> pool;
+    pool.pushBackBulk(v.data(), v.size());
+    const double t0 = now_sec();
+    PFSPNode parent;
+    while (pool.popBack(parent)) pfsp_decompose(I, lb, parent, r.tree, r.sol, best, pool);
+    r.time = now_sec() - t0;
+    r.optimum = best;
+    r.phases.push_back({r.tree, r.sol, r.time});
+  }
+  return result_to_dict(r);
+}
+
 // Exercises pool push/pop/bulk semantics from C++ (unit-test helper).
 py::dict pool_selftest() {
   py::dict d;
@@ -263,6 +302,11 @@ PYBIND11_MODULE(_core, mod) {
           py::arg("best0") = 0, py::arg("m") = 25, py::arg("M") = 50000,
           py::arg("device") = 0, py::arg("mode") = "devpool",
           py::arg("capacity") = (1ull << 27));
+
+  mod.def("nqueens_seq_from_pool", &nqueens_seq_from_pool, py::arg("nodes"), py::arg("N"),
+          py::arg("g") = 1);
+  mod.def("pfsp_seq_from_pool", &pfsp_seq_from_pool, py::arg("nodes"), py::arg("inst"),
+          py::arg("lb") = "lb1", py::arg("ub") = 1, py::arg("best0") = 0);
 
   mod.def("nq_bfs_frontier", &nq_bfs_frontier, py::arg("N"), py::arg("g") = 1,
           py::arg("target") = 1024);

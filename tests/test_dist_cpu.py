@@ -1,0 +1,33 @@
+"""Distributed tier logic on CPU: gloo backend, world_size 2 (no GPU needed).
+Verifies the redundant-BFS + round-robin-slice partition and the end-of-search
+collectives reproduce sequential counts exactly."""
+import os
+import socket
+import subprocess
+import sys
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def test_dist_gloo_world2():
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node=2",
+        "--master-addr", "127.0.0.1", "--master-port", str(free_port()),
+        os.path.join(HERE, "helpers", "dist_check.py"),
+    ]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env,
+                       cwd=os.path.dirname(HERE))
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "DIST_CHECK_OK" in r.stdout

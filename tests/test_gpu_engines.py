@@ -1,0 +1,54 @@
+"""Single-GPU engines: exact count parity with the sequential engine in both
+phase-2 modes (hostpool = reference-shaped offload, devpool = device-resident
+pool), plus PFSP optimum checks."""
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.mark.parametrize("mode", ["hostpool", "devpool"])
+def test_nqueens_gpu_matches_seq(gpu, mode):
+    seq = gpu.nqueens_seq(13, 1)
+    r = gpu.nqueens_gpu(13, 1, 25, 50000, 0, mode, 1 << 24)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    assert r["diag"]["gpu_iters"] > 0  # offload actually happened
+
+
+@pytest.mark.parametrize("mode", ["hostpool", "devpool"])
+@pytest.mark.parametrize("inst,lb", [(2, "lb1"), (14, "lb1_d"), (14, "lb2")])
+def test_pfsp_gpu_matches_seq_ub1(gpu, mode, inst, lb):
+    seq = gpu.pfsp_seq(inst, lb, 1)
+    r = gpu.pfsp_gpu(inst, lb, 1, 25, 50000, 0, mode, 1 << 24)
+    assert r["optimum"] == seq["optimum"]
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+
+
+@pytest.mark.parametrize("mode", ["hostpool", "devpool"])
+def test_pfsp_gpu_ub0_finds_optimum(gpu, mode):
+    r = gpu.pfsp_gpu(2, "lb2", 0, 25, 50000, 0, mode, 1 << 24)
+    assert r["optimum"] == 1359
+
+
+def test_nqueens_gpu_n15_golden(gpu):
+    # the reference's minimum end-to-end milestone: N=15 exact solution count
+    r = gpu.nqueens_gpu(15, 1, 25, 50000, 0, "devpool", 1 << 26)
+    assert r["sol"] == 2279184
+    assert r["tree"] == 171129071
+
+
+def test_multigpu_gpu_eval_matches_seq(gpu):
+    seq = gpu.nqueens_seq(13, 1)
+    r = gpu.nqueens_multigpu(13, 1, 25, 10000, 2, "gpu")  # 2 workers, 1 GPU ok
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+
+
+def test_gpu_from_pool_entry(gpu):
+    # distributed-tier building block: engine started from an explicit frontier
+    nodes, tree, sol = gpu.nq_bfs_frontier(13, 1, 4096)
+    r = gpu.nqueens_gpu_from_pool(nodes, 13, 1, 25, 50000, 0, "devpool", 1 << 24)
+    seq = gpu.nqueens_seq(13, 1)
+    assert tree + r["tree"] == seq["tree"]
+    assert sol + r["sol"] == seq["sol"]

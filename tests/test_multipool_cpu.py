@@ -1,0 +1,41 @@
+"""Multi-pool engine with work stealing + all-idle termination, run with the
+CPU evaluator (protocol parity with nqueens_multigpu_chpl.chpl:158-345 /
+pfsp_multigpu_chpl.chpl:312-535, testable without a GPU)."""
+
+
+def test_nqueens_multipool_matches_seq(core):
+    seq = core.nqueens_seq(12, 1)
+    for D in (1, 2, 4):
+        r = core.nqueens_multigpu(12, 1, 25, 5000, D, "cpu")
+        assert r["tree"] == seq["tree"]
+        assert r["sol"] == seq["sol"]
+
+
+def test_nqueens_multipool_small_M_forces_stealing(core):
+    # tiny M -> many chunks -> stealing definitely exercised
+    seq = core.nqueens_seq(11, 1)
+    r = core.nqueens_multigpu(11, 1, 5, 64, 4, "cpu")
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+
+
+def test_pfsp_multipool_ub1_matches_seq(core):
+    # ub=1 fixes pruning -> multi-pool counts are deterministic
+    seq = core.pfsp_seq(14, "lb1_d", 1)
+    r = core.pfsp_multigpu(14, "lb1_d", 1, 25, 5000, 4, "cpu", False)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    assert r["optimum"] == 1377
+
+
+def test_pfsp_multipool_ub0_finds_optimum(core):
+    for share in (False, True):
+        r = core.pfsp_multigpu(2, "lb2", 0, 5, 256, 4, "cpu", share)
+        assert r["optimum"] == 1359
+
+
+def test_pfsp_multipool_lb2_ub1_matches_seq(core):
+    seq = core.pfsp_seq(14, "lb2", 1)
+    r = core.pfsp_multigpu(14, "lb2", 1, 25, 2000, 3, "cpu", True)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
