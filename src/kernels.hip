@@ -15,10 +15,16 @@
 //                     wave-aggregated (wave64 ballot + one atomic per wave)
 //                     pool reservations; the host only polls a 64 B control
 //                     block every few iterations.
-//   * lb tables staged in LDS (p_times/lags as int16, Johnson schedules as
-//     uint8): ~12.5 KB for 20x20 vs 160 KB available per CU.
+//   * Bound tables staged in LDS (p_times/lags int16, Johnson schedules u8):
+//     ~12 KB for 20x20 vs 160 KB per CU.
+//   * No runtime-indexed per-thread arrays (they would spill to scratch on
+//     CDNA4): lb2's machine-pair-indexed `front` lives in LDS with a
+//     conflict-free padded stride; lb1_d iterates positions with uniform-
+//     length wave loops instead of a job-indexed local array.
 //   * Wavefront = 64 idioms throughout (__ballot is 64-bit).
 #include <hip/hip_runtime.h>
+
+#include <type_traits>
 
 #include "gpu_api.hpp"
 
@@ -30,9 +36,8 @@ namespace gats {
 // Wave64 helpers
 // ---------------------------------------------------------------------------
 
-// Order-preserving-free wave append: every lane with pred==true gets a unique
-// slot index from one atomicAdd per wave. Returns the lane's slot (valid only
-// when pred), and adds the wave's total to *counter.
+// Every lane with pred==true gets a unique slot from ONE atomicAdd per wave.
+// Must be executed by all lanes of the wave (uniform control flow).
 __device__ inline unsigned long long wave_reserve(bool pred, unsigned long long* counter) {
   const unsigned long long mask = __ballot(pred);
   const int lane = threadIdx.x & 63;
@@ -63,13 +68,73 @@ __device__ inline void copy_node(void* dst, const void* src) {
   d[2] = s[2];
 }
 
+// Block-cooperative staging of the parents this block touches into LDS.
+// One shared copy per parent (the reference's per-thread `var parent =
+// parents_d[parentId]` would be a 96 B/thread scratch/LDS spill on CDNA4
+// because board/prmu are runtime-indexed). Caller must __syncthreads() after.
+// Returns this thread's local parent index, or -1 if t >= total.
+template <class NodeT, int MAXN>
+__device__ inline int stage_parents(const NodeT* parents, unsigned long long total,
+                                    int per_parent, NodeT (&snodes)[MAXN],
+                                    unsigned long long t) {
+  const unsigned long long t0 = static_cast<unsigned long long>(blockIdx.x) * blockDim.x;
+  if (t0 >= total) return -1;
+  const unsigned int first = static_cast<unsigned int>(t0 / per_parent);
+  unsigned long long tlast = t0 + blockDim.x - 1;
+  if (tlast > total - 1) tlast = total - 1;
+  const unsigned int last = static_cast<unsigned int>(tlast / per_parent);
+  const int nblk = static_cast<int>(last - first + 1);
+  const uint32_t* src = reinterpret_cast<const uint32_t*>(parents + first);
+  uint32_t* dst = reinterpret_cast<uint32_t*>(&snodes[0]);
+  const int words = nblk * static_cast<int>(sizeof(NodeT) / 4);
+  for (int i = threadIdx.x; i < words; i += blockDim.x) dst[i] = src[i];
+  if (t >= total) return -1;
+  return static_cast<int>(t / per_parent - first);
+}
+
+// Emit a child node straight into its pool slot: three patched qword stores
+// plus two byte stores for the swapped permutation entries — avoids any
+// runtime-indexed private array (scratch) for the child.
+__device__ inline void emit_nq_child(NQNode* pool, unsigned long long slot,
+                                     const NQNode& parent, int depth, int k) {
+  const unsigned long long* s = reinterpret_cast<const unsigned long long*>(&parent);
+  unsigned long long* d = reinterpret_cast<unsigned long long*>(&pool[slot]);
+  // byte 0 = depth; board bytes at 1..20
+  d[0] = (s[0] & ~0xFFull) | static_cast<unsigned long long>(depth + 1);
+  d[1] = s[1];
+  d[2] = s[2];
+  uint8_t* db = reinterpret_cast<uint8_t*>(d);
+  db[1 + depth] = parent.board[k];
+  db[1 + k] = parent.board[depth];
+}
+
+__device__ inline void emit_pfsp_child(PFSPNode* pool, unsigned long long slot,
+                                       const PFSPNode& parent, int depth, int limit1,
+                                       int k) {
+  const unsigned long long* s = reinterpret_cast<const unsigned long long*>(&parent);
+  unsigned long long* d = reinterpret_cast<unsigned long long*>(&pool[slot]);
+  // byte 0 = depth, byte 1 = limit1; prmu bytes at 2..21
+  d[0] = (s[0] & ~0xFFFFull) |
+         static_cast<unsigned long long>(static_cast<uint8_t>(depth + 1)) |
+         (static_cast<unsigned long long>(static_cast<uint8_t>(limit1 + 1)) << 8);
+  d[1] = s[1];
+  d[2] = s[2];
+  uint8_t* db = reinterpret_cast<uint8_t*>(d);
+  db[2 + depth] = parent.prmu[k];
+  db[2 + k] = parent.prmu[depth];
+}
+
 // ---------------------------------------------------------------------------
-// Devpool control block + generic pool kernels (shared by both problems)
+// Devpool control kernels (shared by both problems)
 // ---------------------------------------------------------------------------
 
 // begin: decide this iteration's chunk (popBackBulk semantics, Pool.chpl:50-60):
 // pop min(size, M) from the back iff size >= m.
 __global__ void k_begin(DevCtl* ctl, unsigned long long m, unsigned long long M) {
+  if (ctl->overflow) {  // freeze the pool; host will abort at next readback
+    ctl->chunk = 0;
+    return;
+  }
   unsigned long long size = ctl->size;
   unsigned long long c = (size >= m) ? (size < M ? size : M) : 0;
   ctl->chunk = c;
@@ -80,13 +145,12 @@ __global__ void k_begin(DevCtl* ctl, unsigned long long m, unsigned long long M)
 // copy the popped parents out of the pool so expand can append over them.
 template <typename NodeT>
 __global__ void k_copy_parents(const DevCtl* ctl, const NodeT* pool, NodeT* parents) {
-  const unsigned long long c = ctl->chunk;
-  const unsigned long long words = c * (sizeof(NodeT) / 8);
+  const unsigned long long words = ctl->chunk * (sizeof(NodeT) / 8);
   const unsigned long long* src =
       reinterpret_cast<const unsigned long long*>(pool + ctl->size);
   unsigned long long* dst = reinterpret_cast<unsigned long long*>(parents);
   for (unsigned long long i = blockIdx.x * blockDim.x + threadIdx.x; i < words;
-       i += gridDim.x * blockDim.x)
+       i += static_cast<unsigned long long>(gridDim.x) * blockDim.x)
     dst[i] = src[i];
 }
 
@@ -107,11 +171,14 @@ __device__ inline uint8_t nq_safe(const uint8_t* board, int depth, int q, int g)
 // hostpool mode: one thread per (parent, k), labels out
 // (reference mapping nqueens_gpu_cuda.cu:137-164).
 __global__ void k_nq_eval(const NQNode* parents, int n, int N, int g, uint8_t* labels) {
-  const int t = blockIdx.x * blockDim.x + threadIdx.x;
-  if (t >= n * N) return;
-  const int pid = t / N;
-  const int k = t - pid * N;
-  const NQNode parent = parents[pid];
+  __shared__ NQNode snodes[BLOCK + 2];
+  const unsigned long long t =
+      static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int lp = stage_parents(parents, static_cast<unsigned long long>(n) * N, N, snodes, t);
+  __syncthreads();
+  if (lp < 0) return;
+  const int k = static_cast<int>(t % static_cast<unsigned long long>(N));
+  const NQNode& parent = snodes[lp];
   const int depth = parent.depth;
   if (k >= depth && depth < N)
     labels[t] = nq_safe(parent.board, depth, parent.board[k], g);
@@ -120,28 +187,23 @@ __global__ void k_nq_eval(const NQNode* parents, int n, int N, int g, uint8_t* l
 // devpool mode: evaluate + prune + append children on-device.
 __global__ void k_nq_expand(DevCtl* ctl, const NQNode* parents, NQNode* pool,
                             unsigned long long capacity, int N, int g) {
-  const unsigned long long c = ctl->chunk;
-  const unsigned long long total = c * N;
+  __shared__ NQNode snodes[BLOCK + 2];
+  const unsigned long long total = ctl->chunk * N;
   const unsigned long long t =
       static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int lp = stage_parents(parents, total, N, snodes, t);
+  __syncthreads();
 
   bool is_sol = false, has_child = false;
-  NQNode child;
-  if (t < total) {
-    const unsigned int pid = static_cast<unsigned int>(t / N);
-    const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * N);
-    const NQNode parent = parents[pid];
-    const int depth = parent.depth;
+  int depth = 0, k = 0;
+  if (lp >= 0) {
+    const NQNode& parent = snodes[lp];
+    k = static_cast<int>(t % static_cast<unsigned long long>(N));
+    depth = parent.depth;
     if (depth == N) {
       is_sol = (k == 0);  // leaf parent: counted once (nqueens_chpl.chpl:78-80)
     } else if (k >= depth) {
-      if (nq_safe(parent.board, depth, parent.board[k], g)) {
-        child = parent;
-        child.depth = static_cast<uint8_t>(depth + 1);
-        child.board[depth] = parent.board[k];
-        child.board[k] = parent.board[depth];
-        has_child = true;
-      }
+      has_child = nq_safe(parent.board, depth, parent.board[k], g) != 0;
     }
   }
   wave_count(is_sol, &ctl->sol);
@@ -150,79 +212,89 @@ __global__ void k_nq_expand(DevCtl* ctl, const NQNode* parents, NQNode* pool,
     if (slot >= capacity) {
       ctl->overflow = 1;
     } else {
-      copy_node(&pool[slot], &child);
+      emit_nq_child(pool, slot, snodes[lp], depth, k);
     }
   }
   wave_count(has_child, &ctl->tree);
 }
 
 // ---------------------------------------------------------------------------
-// PFSP device bound math (templated on machine count for full unrolling).
-// Tables live in LDS; p_times/lags int16, schedules/pairs uint8.
+// PFSP device bound math (templated on machine count MM for full unrolling;
+// every per-thread array index is compile-time so nothing spills to scratch).
 // ---------------------------------------------------------------------------
 
-struct PfspLds {
-  int16_t p[20 * 20];        // p_times[machine][job]
-  int32_t min_tails[20];
-  int16_t lags[190 * 20];    // lb2 only
-  uint8_t js[190 * 20];      // lb2 only: johnson schedules
-  uint8_t pair1[190], pair2[190];
+template <int MM>
+struct LdsLb1 {
+  int16_t p[MM * MAX_JOBS];  // p_times[machine][job]
+  int32_t min_tails[MM];
 };
 
-// Cooperative staging of the device-global tables into LDS.
-template <int MM, bool WITH_LB2>
-__device__ inline void stage_tables(PfspLds& lds, const PfspDevTables& tb, int jobs) {
-  const int n_p = MM * jobs;
-  for (int i = threadIdx.x; i < n_p; i += blockDim.x) lds.p[i] = tb.p_times[i];
-  for (int i = threadIdx.x; i < MM; i += blockDim.x) lds.min_tails[i] = tb.min_tails[i];
-  if (WITH_LB2) {
-    const int pairs = MM * (MM - 1) / 2;
-    for (int i = threadIdx.x; i < pairs * jobs; i += blockDim.x) {
-      lds.lags[i] = tb.lags[i];
-      lds.js[i] = tb.johnson_schedules[i];
-    }
-    for (int i = threadIdx.x; i < pairs; i += blockDim.x) {
-      lds.pair1[i] = tb.pairs1[i];
-      lds.pair2[i] = tb.pairs2[i];
-    }
+template <int MM>
+struct LdsLb2 {
+  static constexpr int PAIRS = MM * (MM - 1) / 2;
+  int16_t p[MM * MAX_JOBS];
+  int32_t min_tails[MM];
+  int16_t lags[PAIRS * MAX_JOBS];
+  uint8_t js[PAIRS * MAX_JOBS];  // per-pair Johnson schedules
+  uint8_t pair1[PAIRS], pair2[PAIRS];
+  // per-thread `front` scratch, runtime-indexed by machine-pair ids; padded
+  // stride MM+1 keeps the 32-bank groups conflict-free (stride odd vs 32)
+  int front[BLOCK * (MM + 1)];
+};
+
+template <int MM, class LDS>
+__device__ inline void stage_lb1_tables(LDS& lds, const PfspDevTables& tb, int jobs) {
+  for (int i = threadIdx.x; i < MM * jobs; i += blockDim.x) lds.p[i] = tb.p_times[i];
+  if (threadIdx.x < MM) lds.min_tails[threadIdx.x] = tb.min_tails[threadIdx.x];
+}
+
+template <int MM>
+__device__ inline void stage_lb2_tables(LdsLb2<MM>& lds, const PfspDevTables& tb, int jobs) {
+  stage_lb1_tables<MM>(lds, tb, jobs);
+  constexpr int PAIRS = LdsLb2<MM>::PAIRS;
+  for (int i = threadIdx.x; i < PAIRS * jobs; i += blockDim.x) {
+    lds.lags[i] = tb.lags[i];
+    lds.js[i] = tb.johnson_schedules[i];
   }
-  __syncthreads();
+  if (threadIdx.x < PAIRS) {
+    lds.pair1[threadIdx.x] = tb.pairs1[threadIdx.x];
+    lds.pair2[threadIdx.x] = tb.pairs2[threadIdx.x];
+  }
 }
 
 // front <- completion times of the child prefix (parent prefix + job k placed
-// at position depth); c_bound_simple.c:31-69 semantics without materializing
-// the swapped permutation.
-template <int MM>
-__device__ inline void child_front(const PfspLds& lds, const uint8_t* prmu, int depth,
-                                   int job_k, int jobs, int* front) {
+// at position depth); c_bound_simple.c:31-69 without materializing the swap.
+// `front` may be a register array (compile-time indexed here) or LDS.
+template <int MM, class LDS>
+__device__ inline void child_front(const LDS& lds, const uint8_t* prmu, int depth, int job_k,
+                                   int jobs, int* front, int stride) {
 #pragma unroll
-  for (int i = 0; i < MM; i++) front[i] = 0;
+  for (int i = 0; i < MM; i++) front[i * stride] = 0;
   for (int i = 0; i < depth; i++) {
     const int job = prmu[i];
     front[0] += lds.p[job];
 #pragma unroll
     for (int j = 1; j < MM; j++) {
-      const int prev = front[j - 1] > front[j] ? front[j - 1] : front[j];
-      front[j] = prev + lds.p[j * jobs + job];
+      const int prev = max(front[(j - 1) * stride], front[j * stride]);
+      front[j * stride] = prev + lds.p[j * jobs + job];
     }
   }
   front[0] += lds.p[job_k];
 #pragma unroll
   for (int j = 1; j < MM; j++) {
-    const int prev = front[j - 1] > front[j] ? front[j - 1] : front[j];
-    front[j] = prev + lds.p[j * jobs + job_k];
+    const int prev = max(front[(j - 1) * stride], front[j * stride]);
+    front[j * stride] = prev + lds.p[j * jobs + job_k];
   }
 }
 
 // lb1 bound of the child that schedules prmu[k] next (c_bound_simple.c:143-158;
-// limit2 == jobs so the back schedule is the constant min_tails row,
+// limit2 == jobs, so the back schedule is the constant min_tails row,
 // SURVEY.md §8.2).
 template <int MM>
-__device__ inline int lb1_child_bound(const PfspLds& lds, const uint8_t* prmu, int depth,
+__device__ inline int lb1_child_bound(const LdsLb1<MM>& lds, const uint8_t* prmu, int depth,
                                       int k, int jobs) {
   int front[MM];
-  const int job_k = prmu[k];
-  child_front<MM>(lds, prmu, depth, job_k, jobs, front);
+  child_front<MM>(lds, prmu, depth, prmu[k], jobs, front, 1);
 
   int remain[MM];
 #pragma unroll
@@ -238,29 +310,64 @@ __device__ inline int lb1_child_bound(const PfspLds& lds, const uint8_t* prmu, i
   int lb = tmp0 + lds.min_tails[0];
 #pragma unroll
   for (int i = 1; i < MM; i++) {
-    const int f = front[i] + remain[i];
-    const int tmp1 = tmp0 > f ? tmp0 : f;
-    const int v = tmp1 + lds.min_tails[i];
-    lb = lb > v ? lb : v;
+    const int tmp1 = max(tmp0, front[i] + remain[i]);
+    lb = max(lb, tmp1 + lds.min_tails[i]);
     tmp0 = tmp1;
   }
   return lb;
 }
 
-// lb2: Johnson two-machine relaxation over all machine pairs with early exit
-// (c_bound_johnson.c:211-254). Pair order is identity (LB2_FULL).
+// lb1_d setup: parent-prefix front + remain over unscheduled jobs
+// (c_bound_simple.c:52-69,109-124). Register arrays, compile-time indexed.
 template <int MM>
-__device__ inline int lb2_child_bound(const PfspLds& lds, const uint8_t* prmu, int depth,
-                                      int k, int jobs, int best) {
-  int front[MM];
+__device__ inline void lb1d_setup(const LdsLb1<MM>& lds, const uint8_t* prmu, int limit1,
+                                  int jobs, int* front, int* remain) {
+#pragma unroll
+  for (int i = 0; i < MM; i++) front[i] = remain[i] = 0;
+  for (int i = 0; i <= limit1; i++) {
+    const int job = prmu[i];
+    front[0] += lds.p[job];
+#pragma unroll
+    for (int j = 1; j < MM; j++) front[j] = max(front[j - 1], front[j]) + lds.p[j * jobs + job];
+  }
+  for (int i = limit1 + 1; i < jobs; i++) {
+    const int job = prmu[i];
+#pragma unroll
+    for (int j = 0; j < MM; j++) remain[j] += lds.p[j * jobs + job];
+  }
+}
+
+// O(m) incremental child bound from the parent's front/remain
+// (add_front_and_bound, c_bound_simple.c:218-244). remain still contains the
+// child job itself: lb1_d is a deliberately different bound than lb1.
+template <int MM>
+__device__ inline int lb1d_child_bound(const LdsLb1<MM>& lds, const int* front,
+                                       const int* remain, int job, int jobs) {
+  int lb = front[0] + remain[0] + lds.min_tails[0];
+  int tmp0 = front[0] + lds.p[job];
+#pragma unroll
+  for (int j = 1; j < MM; j++) {
+    const int tmp1 = max(tmp0, front[j]);
+    lb = max(lb, tmp1 + remain[j] + lds.min_tails[j]);
+    tmp0 = tmp1 + lds.p[j * jobs + job];
+  }
+  return lb;
+}
+
+// lb2: Johnson two-machine relaxation over all machine pairs with early exit
+// (c_bound_johnson.c:211-254). Pair order is identity (LB2_FULL). `front` is
+// this thread's LDS slice (runtime-indexed by pair machine ids).
+template <int MM>
+__device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu, int depth,
+                                      int k, int jobs, int best, int* front) {
   const int job_k = prmu[k];
-  child_front<MM>(lds, prmu, depth, job_k, jobs, front);
+  child_front<MM>(lds, prmu, depth, job_k, jobs, front, 1);
 
   unsigned int scheduled = 0;
   for (int i = 0; i < depth; i++) scheduled |= 1u << prmu[i];
   scheduled |= 1u << job_k;
 
-  constexpr int PAIRS = MM * (MM - 1) / 2;
+  constexpr int PAIRS = LdsLb2<MM>::PAIRS;
   int lb = 0;
   for (int l = 0; l < PAIRS; l++) {
     const int ma0 = lds.pair1[l];
@@ -273,87 +380,75 @@ __device__ inline int lb2_child_bound(const PfspLds& lds, const uint8_t* prmu, i
       const int job = js[j];
       if (!(scheduled >> job & 1u)) {
         tmp0 += lds.p[ma0 * jobs + job];
-        const int t = tmp0 + lag[job];
-        tmp1 = tmp1 > t ? tmp1 : t;
+        tmp1 = max(tmp1, tmp0 + lag[job]);
         tmp1 += lds.p[ma1 * jobs + job];
       }
     }
-    const int a = tmp1 + lds.min_tails[ma1];
-    const int b = tmp0 + lds.min_tails[ma0];
-    const int v = a > b ? a : b;
-    lb = lb > v ? lb : v;
+    lb = max(lb, max(tmp1 + lds.min_tails[ma1], tmp0 + lds.min_tails[ma0]));
     if (lb > best) break;
   }
   return lb;
-}
-
-// lb1_d: all children of one parent in O(m) each after one O(mn) setup
-// (c_bound_simple.c:160-244). Returns bounds indexed by JOB id in lb_begin.
-template <int MM>
-__device__ inline void lb1d_children(const PfspLds& lds, const uint8_t* prmu, int limit1,
-                                     int jobs, int* lb_begin) {
-  int front[MM], remain[MM];
-#pragma unroll
-  for (int i = 0; i < MM; i++) front[i] = remain[i] = 0;
-  for (int i = 0; i <= limit1; i++) {
-    const int job = prmu[i];
-    front[0] += lds.p[job];
-#pragma unroll
-    for (int j = 1; j < MM; j++) {
-      const int prev = front[j - 1] > front[j] ? front[j - 1] : front[j];
-      front[j] = prev + lds.p[j * jobs + job];
-    }
-  }
-  for (int i = limit1 + 1; i < jobs; i++) {
-    const int job = prmu[i];
-#pragma unroll
-    for (int j = 0; j < MM; j++) remain[j] += lds.p[j * jobs + job];
-  }
-  for (int i = limit1 + 1; i < jobs; i++) {
-    const int job = prmu[i];
-    int lb = front[0] + remain[0] + lds.min_tails[0];
-    int tmp0 = front[0] + lds.p[job];
-#pragma unroll
-    for (int j = 1; j < MM; j++) {
-      const int tmp1 = tmp0 > front[j] ? tmp0 : front[j];
-      const int v = tmp1 + remain[j] + lds.min_tails[j];
-      lb = lb > v ? lb : v;
-      tmp0 = tmp1 + lds.p[j * jobs + job];
-    }
-    lb_begin[job] = lb;
-  }
 }
 
 // ---------------------------------------------------------------------------
 // PFSP hostpool kernels: bounds out, host prunes (oracle-comparable).
 // ---------------------------------------------------------------------------
 
-template <int MM, int LB>
-__global__ void k_pfsp_eval(const PFSPNode* parents, int n, int jobs, PfspDevTables tb,
-                            int best, int32_t* bounds) {
-  __shared__ PfspLds lds;
-  stage_tables<MM, LB == 2>(lds, tb, jobs);
+template <int MM>
+__global__ void k_pfsp_eval_lb1(const PFSPNode* parents, int n, int jobs, PfspDevTables tb,
+                                int32_t* bounds) {
+  __shared__ LdsLb1<MM> lds;
+  __shared__ PFSPNode snodes[BLOCK / 5 + 2];
+  stage_lb1_tables<MM>(lds, tb, jobs);
+  const unsigned long long t =
+      static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int lp =
+      stage_parents(parents, static_cast<unsigned long long>(n) * jobs, jobs, snodes, t);
+  __syncthreads();
+  if (lp < 0) return;
+  const int k = static_cast<int>(t % static_cast<unsigned long long>(jobs));
+  const PFSPNode& parent = snodes[lp];
+  if (k >= parent.limit1 + 1)
+    bounds[t] = lb1_child_bound<MM>(lds, parent.prmu, parent.depth, k, jobs);
+}
 
-  if (LB == 1 || LB == 2) {
-    const int t = blockIdx.x * blockDim.x + threadIdx.x;
-    if (t >= n * jobs) return;
-    const int pid = t / jobs;
-    const int k = t - pid * jobs;
-    const PFSPNode parent = parents[pid];
-    const int depth = parent.depth;
-    if (k >= parent.limit1 + 1) {
-      bounds[t] = (LB == 1) ? lb1_child_bound<MM>(lds, parent.prmu, depth, k, jobs)
-                            : lb2_child_bound<MM>(lds, parent.prmu, depth, k, jobs, best);
-    }
-  } else {  // lb1_d: one thread per parent (reference evaluate.cu:51-70 mapping)
-    const int pid = blockIdx.x * blockDim.x + threadIdx.x;
-    if (pid >= n) return;
-    const PFSPNode parent = parents[pid];
-    int lb_begin[20];
-    lb1d_children<MM>(lds, parent.prmu, parent.limit1, jobs, lb_begin);
-    for (int k = parent.limit1 + 1; k < jobs; k++)
-      bounds[pid * jobs + k] = lb_begin[parent.prmu[k]];
-  }
+// lb1_d: one thread per parent (reference evaluate.cu:51-70 mapping); bounds
+// written per position, no job-indexed local array.
+template <int MM>
+__global__ void k_pfsp_eval_lb1d(const PFSPNode* parents, int n, int jobs, PfspDevTables tb,
+                                 int32_t* bounds) {
+  __shared__ LdsLb1<MM> lds;
+  __shared__ PFSPNode snodes[BLOCK];
+  stage_lb1_tables<MM>(lds, tb, jobs);
+  const unsigned long long t =
+      static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int lp = stage_parents(parents, static_cast<unsigned long long>(n), 1, snodes, t);
+  __syncthreads();
+  if (lp < 0) return;
+  const PFSPNode& parent = snodes[lp];
+  int front[MM], remain[MM];
+  lb1d_setup<MM>(lds, parent.prmu, parent.limit1, jobs, front, remain);
+  for (int k = parent.limit1 + 1; k < jobs; k++)
+    bounds[t * jobs + k] = lb1d_child_bound<MM>(lds, front, remain, parent.prmu[k], jobs);
+}
+
+template <int MM>
+__global__ void k_pfsp_eval_lb2(const PFSPNode* parents, int n, int jobs, PfspDevTables tb,
+                                int best, int32_t* bounds) {
+  __shared__ LdsLb2<MM> lds;
+  __shared__ PFSPNode snodes[BLOCK / 5 + 2];
+  stage_lb2_tables<MM>(lds, tb, jobs);
+  const unsigned long long t =
+      static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int lp =
+      stage_parents(parents, static_cast<unsigned long long>(n) * jobs, jobs, snodes, t);
+  __syncthreads();
+  if (lp < 0) return;
+  const int k = static_cast<int>(t % static_cast<unsigned long long>(jobs));
+  const PFSPNode& parent = snodes[lp];
+  int* front = &lds.front[threadIdx.x * (MM + 1)];
+  if (k >= parent.limit1 + 1)
+    bounds[t] = lb2_child_bound<MM>(lds, parent.prmu, parent.depth, k, jobs, best, front);
 }
 
 // ---------------------------------------------------------------------------
@@ -363,36 +458,41 @@ __global__ void k_pfsp_eval(const PFSPNode* parents, int n, int jobs, PfspDevTab
 template <int MM, int LB>
 __global__ void k_pfsp_expand(DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
                               unsigned long long capacity, int jobs, PfspDevTables tb) {
-  __shared__ PfspLds lds;
-  stage_tables<MM, LB == 2>(lds, tb, jobs);
+  using LDS = typename std::conditional<LB == 2, LdsLb2<MM>, LdsLb1<MM>>::type;
+  __shared__ LDS lds;
+  __shared__ PFSPNode snodes[BLOCK / 5 + 2];
+  if constexpr (LB == 2)
+    stage_lb2_tables<MM>(lds, tb, jobs);
+  else
+    stage_lb1_tables<MM>(lds, tb, jobs);
 
-  const unsigned long long c = ctl->chunk;
-  const unsigned long long total = c * jobs;
+  const unsigned long long total = ctl->chunk * jobs;
   const unsigned long long t =
       static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int lp = stage_parents(parents, total, jobs, snodes, t);
+  __syncthreads();
   const int best = ctl->best;  // fresher than the reference's per-launch scalar; still a
                                // valid incumbent, so pruning stays correct
 
   bool is_sol = false, has_child = false;
-  int lb = 0;
-  PFSPNode child;
-  if (t < total) {
-    const unsigned int pid = static_cast<unsigned int>(t / jobs);
-    const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * jobs);
-    const PFSPNode parent = parents[pid];
-    const int depth = parent.depth;
-    if (k >= parent.limit1 + 1) {
-      lb = (LB == 1) ? lb1_child_bound<MM>(lds, parent.prmu, depth, k, jobs)
-                     : lb2_child_bound<MM>(lds, parent.prmu, depth, k, jobs, best);
+  int depth = 0, limit1 = 0, k = 0;
+  if (lp >= 0) {
+    const PFSPNode& parent = snodes[lp];
+    k = static_cast<int>(t % static_cast<unsigned long long>(jobs));
+    depth = parent.depth;
+    limit1 = parent.limit1;
+    if (k >= limit1 + 1) {
+      int lb;
+      if constexpr (LB == 2) {
+        int* front = &lds.front[threadIdx.x * (MM + 1)];
+        lb = lb2_child_bound<MM>(lds, parent.prmu, depth, k, jobs, best, front);
+      } else {
+        lb = lb1_child_bound<MM>(lds, parent.prmu, depth, k, jobs);
+      }
       if (depth + 1 == jobs) {
         is_sol = true;
         if (lb < best) atomicMin(&ctl->best, lb);
       } else if (lb < best) {
-        child = parent;
-        child.depth = static_cast<int8_t>(depth + 1);
-        child.limit1 = static_cast<int8_t>(parent.limit1 + 1);
-        child.prmu[depth] = parent.prmu[k];
-        child.prmu[k] = parent.prmu[depth];
         has_child = true;
       }
     }
@@ -403,79 +503,60 @@ __global__ void k_pfsp_expand(DevCtl* ctl, const PFSPNode* parents, PFSPNode* po
     if (slot >= capacity) {
       ctl->overflow = 1;
     } else {
-      copy_node(&pool[slot], &child);
+      emit_pfsp_child(pool, slot, snodes[lp], depth, limit1, k);
     }
   }
   wave_count(has_child, &ctl->tree);
 }
 
-// lb1_d devpool: one thread per parent; per-lane child counts aggregated with a
-// wave scan, one pool reservation per wave.
+// lb1_d devpool: one thread per parent; the child loop is wave-uniform
+// (fixed bounds 0..jobs) so wave_reserve stays collective.
 template <int MM>
 __global__ void k_pfsp_expand_lb1d(DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
                                    unsigned long long capacity, int jobs, PfspDevTables tb) {
-  __shared__ PfspLds lds;
-  stage_tables<MM, false>(lds, tb, jobs);
+  __shared__ LdsLb1<MM> lds;
+  __shared__ PFSPNode snodes[BLOCK];
+  stage_lb1_tables<MM>(lds, tb, jobs);
 
   const unsigned long long c = ctl->chunk;
-  const unsigned long long pid =
+  const unsigned long long t =
       static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int lp = stage_parents(parents, c, 1, snodes, t);
+  __syncthreads();
   const int best0 = ctl->best;
+  const bool active = lp >= 0;
 
-  int lb_begin[20];
-  PFSPNode parent;
-  int n_children = 0, n_sols = 0;
-  if (pid < c) {
-    parent = parents[pid];
-    lb1d_children<MM>(lds, parent.prmu, parent.limit1, jobs, lb_begin);
-    const int depth = parent.depth;
-    for (int k = parent.limit1 + 1; k < jobs; k++) {
-      const int lb = lb_begin[parent.prmu[k]];
+  int front[MM], remain[MM];
+  int n_sols = 0;
+  int depth = 0, limit1 = 0;
+  if (active) {
+    const PFSPNode& parent = snodes[lp];
+    depth = parent.depth;
+    limit1 = parent.limit1;
+    lb1d_setup<MM>(lds, parent.prmu, limit1, jobs, front, remain);
+  }
+
+  for (int k = 0; k < jobs; k++) {  // uniform length across the wave
+    bool has_child = false;
+    if (active && k >= limit1 + 1) {
+      const int lb = lb1d_child_bound<MM>(lds, front, remain, snodes[lp].prmu[k], jobs);
       if (depth + 1 == jobs) {
         n_sols++;
         if (lb < best0) atomicMin(&ctl->best, lb);
       } else if (lb < best0) {
-        n_children++;
+        has_child = true;
       }
     }
-  }
-
-  // wave-exclusive scan of per-lane child counts (wave64: 6 shfl steps)
-  const int lane = threadIdx.x & 63;
-  int scan = n_children;
-#pragma unroll
-  for (int d = 1; d < 64; d <<= 1) {
-    const int v = __shfl_up(scan, d);
-    if (lane >= d) scan += v;
-  }
-  const int wave_total = __shfl(scan, 63);
-  const int my_off = scan - n_children;
-  unsigned long long base = 0;
-  if (lane == 0 && wave_total > 0)
-    base = atomicAdd(&ctl->size, static_cast<unsigned long long>(wave_total));
-  base = __shfl(base, 0);
-
-  if (pid < c && n_children > 0) {
-    if (base + wave_total > capacity) {
-      ctl->overflow = 1;
-    } else {
-      unsigned long long slot = base + my_off;
-      const int depth = parent.depth;
-      for (int k = parent.limit1 + 1; k < jobs; k++) {
-        const int lb = lb_begin[parent.prmu[k]];
-        if (depth + 1 != jobs && lb < best0) {
-          PFSPNode child = parent;
-          child.depth = static_cast<int8_t>(depth + 1);
-          child.limit1 = static_cast<int8_t>(parent.limit1 + 1);
-          child.prmu[depth] = parent.prmu[k];
-          child.prmu[k] = parent.prmu[depth];
-          copy_node(&pool[slot++], &child);
-        }
+    const unsigned long long slot = wave_reserve(has_child, &ctl->size);
+    if (has_child) {
+      if (slot >= capacity) {
+        ctl->overflow = 1;
+      } else {
+        emit_pfsp_child(pool, slot, snodes[lp], depth, limit1, k);
       }
     }
+    wave_count(has_child, &ctl->tree);
   }
-  if (lane == 0 && wave_total > 0)
-    atomicAdd(&ctl->tree, static_cast<unsigned long long>(wave_total));
   if (n_sols > 0) atomicAdd(&ctl->sol, static_cast<unsigned long long>(n_sols));
 }
 
@@ -493,16 +574,14 @@ void launch_begin(DevCtl* ctl, unsigned long long m, unsigned long long M, hipSt
 
 void launch_copy_parents_nq(const DevCtl* ctl, const NQNode* pool, NQNode* parents,
                             unsigned long long maxChunk, hipStream_t s) {
-  const unsigned long long words = maxChunk * 3;
-  int g = grid_for(words);
+  int g = grid_for(maxChunk * 3);
   if (g > 1024) g = 1024;
   hipLaunchKernelGGL(k_copy_parents<NQNode>, dim3(g), dim3(BLOCK), 0, s, ctl, pool, parents);
 }
 
 void launch_copy_parents_pfsp(const DevCtl* ctl, const PFSPNode* pool, PFSPNode* parents,
                               unsigned long long maxChunk, hipStream_t s) {
-  const unsigned long long words = maxChunk * 3;
-  int g = grid_for(words);
+  int g = grid_for(maxChunk * 3);
   if (g > 1024) g = 1024;
   hipLaunchKernelGGL(k_copy_parents<PFSPNode>, dim3(g), dim3(BLOCK), 0, s, ctl, pool,
                      parents);
@@ -526,14 +605,14 @@ static void launch_pfsp_eval_mm(const PFSPNode* parents, int n, int jobs, int lb
                                 const PfspDevTables& tb, int best, int32_t* bounds,
                                 hipStream_t s) {
   if (lbk == 0) {  // lb1_d: thread per parent
-    hipLaunchKernelGGL((k_pfsp_eval<MM, 0>), dim3(grid_for(n)), dim3(BLOCK), 0, s, parents,
-                       n, jobs, tb, best, bounds);
+    hipLaunchKernelGGL((k_pfsp_eval_lb1d<MM>), dim3(grid_for(n)), dim3(BLOCK), 0, s, parents,
+                       n, jobs, tb, bounds);
   } else if (lbk == 1) {
-    hipLaunchKernelGGL((k_pfsp_eval<MM, 1>),
+    hipLaunchKernelGGL((k_pfsp_eval_lb1<MM>),
                        dim3(grid_for(static_cast<unsigned long long>(n) * jobs)), dim3(BLOCK),
-                       0, s, parents, n, jobs, tb, best, bounds);
+                       0, s, parents, n, jobs, tb, bounds);
   } else {
-    hipLaunchKernelGGL((k_pfsp_eval<MM, 2>),
+    hipLaunchKernelGGL((k_pfsp_eval_lb2<MM>),
                        dim3(grid_for(static_cast<unsigned long long>(n) * jobs)), dim3(BLOCK),
                        0, s, parents, n, jobs, tb, best, bounds);
   }
