@@ -131,6 +131,42 @@ int lbk_of(LbKind lb) {
 
 }  // namespace
 
+
+// Shared devpool driver: capture BATCH iterations into a hipGraph once, then
+// replay + poll the 64 B control block until the pool drops below m
+// (graph replay ~10-16 us vs ~3.5 us host cost PER LAUNCH eager — the hot
+// loop is launch-bound at chunk sizes this small).
+template <class EnqueueIter>
+static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long m,
+                               int kernels_per_iter, EnqueueIter&& enqueue_iter, Result& r) {
+  PinnedGuard<DevCtl> ctl_h(1);
+  const int BATCH = 16;
+  hipGraph_t graph = nullptr;
+  hipGraphExec_t exec = nullptr;
+  HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
+  for (int b = 0; b < BATCH; b++) enqueue_iter();
+  HIP_CHECK(hipStreamEndCapture(s, &graph));
+  HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
+  bool overflow = false;
+  while (true) {
+    HIP_CHECK(hipGraphLaunch(exec, s));
+    HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d, sizeof(DevCtl), hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    r.kernel_launch += static_cast<uint64_t>(kernels_per_iter) * BATCH;
+    r.d2h++;
+    r.d2h_bytes += sizeof(DevCtl);
+    if (ctl_h.p->overflow) {
+      overflow = true;
+      break;
+    }
+    if (ctl_h.p->size < m) break;
+  }
+  (void)hipGraphExecDestroy(exec);
+  (void)hipGraphDestroy(graph);
+  if (overflow) throw std::runtime_error("device pool overflow; raise capacity");
+  return *ctl_h.p;
+}
+
 // ---------------------------------------------------------------------------
 // N-Queens
 // ---------------------------------------------------------------------------
@@ -173,6 +209,11 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     DevGuard<NQNode> pool_d(capacity);
     DevGuard<NQNode> parents_d(M);
     DevGuard<DevCtl> ctl_d(1);
+    const int G = static_cast<int>((static_cast<unsigned long long>(M) * N +
+                                    DEV_EMIT_TILE - 1) / DEV_EMIT_TILE);
+    DevGuard<uint8_t> labels_d(static_cast<size_t>(M) * N);
+    DevGuard<uint32_t> bc_d(G), bs_d(G);
+    DevGuard<unsigned long long> boff_d(G);
     const size_t init = pool.size();
     if (init > capacity) throw std::runtime_error("devpool capacity too small");
     HIP_CHECK(hipMemcpy(pool_d.p, pool.data(), init * sizeof(NQNode), hipMemcpyHostToDevice));
@@ -184,29 +225,19 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     r.h2d += 2;
     r.h2d_bytes += init * sizeof(NQNode) + sizeof(DevCtl);
 
-    PinnedGuard<DevCtl> ctl_h(1);
-    const int BATCH = 8;
-    while (true) {
-      for (int b = 0; b < BATCH; b++) {
-        launch_begin(ctl_d.p, m, M, stream.s);
-        launch_copy_parents_nq(ctl_d.p, pool_d.p, parents_d.p, M, stream.s);
-        launch_nq_expand(ctl_d.p, parents_d.p, pool_d.p, capacity, M, N, g, stream.s);
-        r.kernel_launch += 3;
-      }
-      HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d.p, sizeof(DevCtl), hipMemcpyDeviceToHost,
-                               stream.s));
-      HIP_CHECK(hipStreamSynchronize(stream.s));
-      r.d2h++;
-      r.d2h_bytes += sizeof(DevCtl);
-      if (ctl_h.p->overflow) throw std::runtime_error("device pool overflow; raise capacity");
-      if (ctl_h.p->size < static_cast<unsigned long long>(m)) break;
-    }
-    tree = ctl_h.p->tree;
-    sol = ctl_h.p->sol;
-    r.gpu_iters = ctl_h.p->iters;
-    // bring the leftover (< m nodes) back for phase 3
+    auto iter = [&] {
+      launch_begin(ctl_d.p, m, M, stream.s);
+      launch_copy_parents_nq(ctl_d.p, pool_d.p, parents_d.p, M, stream.s);
+      launch_nq_eval3(ctl_d.p, parents_d.p, N, g, labels_d.p, bc_d.p, bs_d.p, M, stream.s);
+      launch_scan(ctl_d.p, bc_d.p, bs_d.p, boff_d.p, G, capacity, stream.s);
+      launch_emit_nq(ctl_d.p, parents_d.p, pool_d.p, labels_d.p, N, boff_d.p, M, stream.s);
+    };
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 5, iter, r);
+    tree = fin.tree;
+    sol = fin.sol;
+    r.gpu_iters = fin.iters;
     pool.clear();
-    const size_t left = ctl_h.p->size;
+    const size_t left = fin.size;
     if (left > 0) {
       std::vector<NQNode> tmp(left);
       HIP_CHECK(hipMemcpy(tmp.data(), pool_d.p, left * sizeof(NQNode), hipMemcpyDeviceToHost));
@@ -300,6 +331,11 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     DevGuard<PFSPNode> pool_d(capacity);
     DevGuard<PFSPNode> parents_d(M);
     DevGuard<DevCtl> ctl_d(1);
+    const int G = static_cast<int>((static_cast<unsigned long long>(M) * jobs +
+                                    DEV_EMIT_TILE - 1) / DEV_EMIT_TILE);
+    DevGuard<uint8_t> labels_d(static_cast<size_t>(M) * jobs);
+    DevGuard<uint32_t> bc_d(G), bs_d(G);
+    DevGuard<unsigned long long> boff_d(G);
     const size_t init = pool.size();
     if (init > capacity) throw std::runtime_error("devpool capacity too small");
     HIP_CHECK(
@@ -313,30 +349,24 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     r.h2d += 2;
     r.h2d_bytes += init * sizeof(PFSPNode) + sizeof(DevCtl);
 
-    PinnedGuard<DevCtl> ctl_h(1);
-    const int BATCH = 8;
-    while (true) {
-      for (int b = 0; b < BATCH; b++) {
-        launch_begin(ctl_d.p, m, M, stream.s);
-        launch_copy_parents_pfsp(ctl_d.p, pool_d.p, parents_d.p, M, stream.s);
-        launch_pfsp_expand(ctl_d.p, parents_d.p, pool_d.p, capacity, M, jobs, machines, lbk,
-                           tables.tb, stream.s);
-        r.kernel_launch += 3;
-      }
-      HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d.p, sizeof(DevCtl), hipMemcpyDeviceToHost,
-                               stream.s));
-      HIP_CHECK(hipStreamSynchronize(stream.s));
-      r.d2h++;
-      r.d2h_bytes += sizeof(DevCtl);
-      if (ctl_h.p->overflow) throw std::runtime_error("device pool overflow; raise capacity");
-      if (ctl_h.p->size < static_cast<unsigned long long>(m)) break;
-    }
-    tree = ctl_h.p->tree;
-    sol = ctl_h.p->sol;
-    best = ctl_h.p->best;
-    r.gpu_iters = ctl_h.p->iters;
+    auto iter = [&] {
+      launch_begin(ctl_d.p, m, M, stream.s);
+      launch_copy_parents_pfsp(ctl_d.p, pool_d.p, parents_d.p, M, stream.s);
+      launch_pfsp_eval3(ctl_d.p, parents_d.p, jobs, machines, lbk, tables.tb, labels_d.p,
+                        bc_d.p, bs_d.p, M, stream.s);
+      if (lbk == 0)  // lb1_d evaluated per-parent: counts need their own pass
+        launch_count(ctl_d.p, labels_d.p, jobs, bc_d.p, bs_d.p, M, stream.s);
+      launch_scan(ctl_d.p, bc_d.p, bs_d.p, boff_d.p, G, capacity, stream.s);
+      launch_emit_pfsp(ctl_d.p, parents_d.p, pool_d.p, labels_d.p, jobs, boff_d.p, M,
+                       stream.s);
+    };
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, lbk == 0 ? 6 : 5, iter, r);
+    tree = fin.tree;
+    sol = fin.sol;
+    best = fin.best;
+    r.gpu_iters = fin.iters;
     pool.clear();
-    const size_t left = ctl_h.p->size;
+    const size_t left = fin.size;
     if (left > 0) {
       std::vector<PFSPNode> tmp(left);
       HIP_CHECK(

@@ -43,14 +43,29 @@ void launch_copy_parents_pfsp(const DevCtl* ctl, const PFSPNode* pool, PFSPNode*
                               unsigned long long maxChunk, hipStream_t s);
 void launch_nq_eval(const NQNode* parents, int n, int N, int g, uint8_t* labels,
                     hipStream_t s);
-void launch_nq_expand(DevCtl* ctl, const NQNode* parents, NQNode* pool,
-                      unsigned long long capacity, unsigned long long maxChunk, int N, int g,
-                      hipStream_t s);
 // lbk: 0 = lb1_d, 1 = lb1, 2 = lb2
 void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, int lbk,
                       const PfspDevTables& tb, int best, int32_t* bounds, hipStream_t s);
-void launch_pfsp_expand(DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
-                        unsigned long long capacity, unsigned long long maxChunk, int jobs,
-                        int machines, int lbk, const PfspDevTables& tb, hipStream_t s);
+
+// devpool scan pipeline (see kernels.hip): eval3 -> [count] -> scan -> emit
+constexpr int DEV_EMIT_TILE = 1024;
+void launch_nq_eval3(const DevCtl* ctl, const NQNode* parents, int N, int g, uint8_t* labels,
+                     uint32_t* blockCounts, uint32_t* blockSols, unsigned long long maxChunk,
+                     hipStream_t s);
+void launch_pfsp_eval3(DevCtl* ctl, const PFSPNode* parents, int jobs, int machines, int lbk,
+                       const PfspDevTables& tb, uint8_t* labels, uint32_t* blockCounts,
+                       uint32_t* blockSols, unsigned long long maxChunk, hipStream_t s);
+void launch_count(const DevCtl* ctl, const uint8_t* labels, int per, uint32_t* blockCounts,
+                  uint32_t* blockSols, unsigned long long maxChunk, hipStream_t s);
+void launch_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
+                 unsigned long long* blockOffsets, int G, unsigned long long capacity,
+                 hipStream_t s);
+void launch_emit_nq(const DevCtl* ctl, const NQNode* parents, NQNode* pool,
+                    const uint8_t* labels, int N, const unsigned long long* blockOffsets,
+                    unsigned long long maxChunk, hipStream_t s);
+void launch_emit_pfsp(const DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
+                      const uint8_t* labels, int jobs,
+                      const unsigned long long* blockOffsets, unsigned long long maxChunk,
+                      hipStream_t s);
 
 }  // namespace gats

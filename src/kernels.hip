@@ -184,40 +184,6 @@ __global__ void k_nq_eval(const NQNode* parents, int n, int N, int g, uint8_t* l
     labels[t] = nq_safe(parent.board, depth, parent.board[k], g);
 }
 
-// devpool mode: evaluate + prune + append children on-device.
-__global__ void k_nq_expand(DevCtl* ctl, const NQNode* parents, NQNode* pool,
-                            unsigned long long capacity, int N, int g) {
-  __shared__ NQNode snodes[BLOCK + 2];
-  const unsigned long long total = ctl->chunk * N;
-  const unsigned long long t =
-      static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
-  const int lp = stage_parents(parents, total, N, snodes, t);
-  __syncthreads();
-
-  bool is_sol = false, has_child = false;
-  int depth = 0, k = 0;
-  if (lp >= 0) {
-    const NQNode& parent = snodes[lp];
-    k = static_cast<int>(t % static_cast<unsigned long long>(N));
-    depth = parent.depth;
-    if (depth == N) {
-      is_sol = (k == 0);  // leaf parent: counted once (nqueens_chpl.chpl:78-80)
-    } else if (k >= depth) {
-      has_child = nq_safe(parent.board, depth, parent.board[k], g) != 0;
-    }
-  }
-  wave_count(is_sol, &ctl->sol);
-  const unsigned long long slot = wave_reserve(has_child, &ctl->size);
-  if (has_child) {
-    if (slot >= capacity) {
-      ctl->overflow = 1;
-    } else {
-      emit_nq_child(pool, slot, snodes[lp], depth, k);
-    }
-  }
-  wave_count(has_child, &ctl->tree);
-}
-
 // ---------------------------------------------------------------------------
 // PFSP device bound math (templated on machine count MM for full unrolling;
 // every per-thread array index is compile-time so nothing spills to scratch).
@@ -451,113 +417,299 @@ __global__ void k_pfsp_eval_lb2(const PFSPNode* parents, int n, int jobs, PfspDe
     bounds[t] = lb2_child_bound<MM>(lds, parent.prmu, parent.depth, k, jobs, best, front);
 }
 
+
 // ---------------------------------------------------------------------------
-// PFSP devpool expand kernels
+// Devpool scan/compact pipeline (atomic-free)
+//
+// Per iteration: begin -> copy_parents -> eval3 (labels + per-block counts)
+// [-> count (lb1_d only)] -> scan (single block: offsets + counter update)
+// -> emit (ranked child writes). No same-cacheline global atomics: the
+// first devpool design used one atomicAdd per wave on DevCtl and saturated
+// the L2 atomic unit (~12 ns per op on one line => ~300 us per iteration at
+// chunk 50k). Labels: 0 = pruned/invalid, 1 = push child, 2 = leaf solution.
 // ---------------------------------------------------------------------------
 
+constexpr int EMIT_TILE = 1024;  // children per block in eval3/count/emit (4/thread)
+
+// stage parents covering child range [c0, c1) into LDS; returns the first pid.
+template <class NodeT, int MAXN>
+__device__ inline unsigned int stage_range(const NodeT* parents, unsigned long long c0,
+                                           unsigned long long c1, int per,
+                                           NodeT (&s)[MAXN]) {
+  const unsigned int first = static_cast<unsigned int>(c0 / per);
+  const unsigned int last = static_cast<unsigned int>((c1 - 1) / per);
+  const int words = static_cast<int>(last - first + 1) * static_cast<int>(sizeof(NodeT) / 4);
+  const uint32_t* src = reinterpret_cast<const uint32_t*>(parents + first);
+  uint32_t* dst = reinterpret_cast<uint32_t*>(&s[0]);
+  for (int i = threadIdx.x; i < words; i += blockDim.x) dst[i] = src[i];
+  return first;
+}
+
+// Exclusive scan of v over the 256-thread block; total broadcast to all lanes.
+__device__ inline uint32_t block_excl_scan(uint32_t v, uint32_t& total) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  uint32_t x = v;
+#pragma unroll
+  for (int d = 1; d < 64; d <<= 1) {
+    const uint32_t y = __shfl_up(x, d);
+    if (lane >= d) x += y;
+  }
+  __shared__ uint32_t wtot[4];
+  __syncthreads();
+  if (lane == 63) wtot[wid] = x;
+  __syncthreads();
+  uint32_t wbase = 0;
+#pragma unroll
+  for (int i = 0; i < 4; i++)
+    if (i < wid) wbase += wtot[i];
+  total = wtot[0] + wtot[1] + wtot[2] + wtot[3];
+  return wbase + x - v;
+}
+
+// N-Queens eval: labels + per-block child/solution counts.
+__global__ void k_nq_eval3(const DevCtl* ctl, const NQNode* parents, int N, int g,
+                           uint8_t* labels, uint32_t* blockCounts, uint32_t* blockSols) {
+  __shared__ NQNode s[EMIT_TILE + 2];  // N >= 1
+  const unsigned long long total = ctl->chunk * N;
+  const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
+  uint32_t cnt = 0, sols = 0;
+  unsigned int first = 0;
+  if (c0 < total) {
+    unsigned long long c1 = c0 + EMIT_TILE;
+    if (c1 > total) c1 = total;
+    first = stage_range(parents, c0, c1, N, s);
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
+      const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
+      if (t < total) {
+        const unsigned int pid = static_cast<unsigned int>(t / N);
+        const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * N);
+        const NQNode& p = s[pid - first];
+        const int depth = p.depth;
+        uint8_t lab = 0;
+        if (depth == N) {
+          lab = (k == 0) ? 2 : 0;  // leaf parent counted once (nqueens_chpl.chpl:78-80)
+        } else if (k >= depth && nq_safe(p.board, depth, p.board[k], g)) {
+          lab = 1;
+        }
+        labels[t] = lab;
+        cnt += (lab == 1);
+        sols += (lab == 2);
+      }
+    }
+  }
+  uint32_t totC, totS;
+  block_excl_scan(cnt, totC);
+  block_excl_scan(sols, totS);
+  if (threadIdx.x == 0) {
+    blockCounts[blockIdx.x] = totC;
+    blockSols[blockIdx.x] = totS;
+  }
+}
+
+// PFSP eval (lb1 / lb2): labels + per-block counts; leaves update ctl->best
+// directly (rare: only parents one level above the leaves).
 template <int MM, int LB>
-__global__ void k_pfsp_expand(DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
-                              unsigned long long capacity, int jobs, PfspDevTables tb) {
+__global__ void k_pfsp_eval3(DevCtl* ctl, const PFSPNode* parents, int jobs,
+                             PfspDevTables tb, uint8_t* labels, uint32_t* blockCounts,
+                             uint32_t* blockSols) {
   using LDS = typename std::conditional<LB == 2, LdsLb2<MM>, LdsLb1<MM>>::type;
   __shared__ LDS lds;
-  __shared__ PFSPNode snodes[BLOCK / 5 + 2];
+  __shared__ PFSPNode s[EMIT_TILE / 5 + 2];  // jobs >= 5
   if constexpr (LB == 2)
     stage_lb2_tables<MM>(lds, tb, jobs);
   else
     stage_lb1_tables<MM>(lds, tb, jobs);
-
   const unsigned long long total = ctl->chunk * jobs;
-  const unsigned long long t =
-      static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
-  const int lp = stage_parents(parents, total, jobs, snodes, t);
+  const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
+  const int best = ctl->best;
+  uint32_t cnt = 0, sols = 0;
+  unsigned int first = 0;
+  if (c0 < total) {
+    unsigned long long c1 = c0 + EMIT_TILE;
+    if (c1 > total) c1 = total;
+    first = stage_range(parents, c0, c1, jobs, s);
+  }
   __syncthreads();
-  const int best = ctl->best;  // fresher than the reference's per-launch scalar; still a
-                               // valid incumbent, so pruning stays correct
-
-  bool is_sol = false, has_child = false;
-  int depth = 0, limit1 = 0, k = 0;
-  if (lp >= 0) {
-    const PFSPNode& parent = snodes[lp];
-    k = static_cast<int>(t % static_cast<unsigned long long>(jobs));
-    depth = parent.depth;
-    limit1 = parent.limit1;
-    if (k >= limit1 + 1) {
-      int lb;
-      if constexpr (LB == 2) {
-        int* front = &lds.front[threadIdx.x * (MM + 1)];
-        lb = lb2_child_bound<MM>(lds, parent.prmu, depth, k, jobs, best, front);
-      } else {
-        lb = lb1_child_bound<MM>(lds, parent.prmu, depth, k, jobs);
-      }
-      if (depth + 1 == jobs) {
-        is_sol = true;
-        if (lb < best) atomicMin(&ctl->best, lb);
-      } else if (lb < best) {
-        has_child = true;
+  if (c0 < total) {
+    for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
+      const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
+      if (t < total) {
+        const unsigned int pid = static_cast<unsigned int>(t / jobs);
+        const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * jobs);
+        const PFSPNode& p = s[pid - first];
+        const int depth = p.depth;
+        uint8_t lab = 0;
+        if (k >= p.limit1 + 1) {
+          int lb;
+          if constexpr (LB == 2) {
+            int* front = &lds.front[threadIdx.x * (MM + 1)];
+            lb = lb2_child_bound<MM>(lds, p.prmu, depth, k, jobs, best, front);
+          } else {
+            lb = lb1_child_bound<MM>(lds, p.prmu, depth, k, jobs);
+          }
+          if (depth + 1 == jobs) {
+            lab = 2;
+            if (lb < best) atomicMin(&ctl->best, lb);
+          } else if (lb < best) {
+            lab = 1;
+          }
+        }
+        labels[t] = lab;
+        cnt += (lab == 1);
+        sols += (lab == 2);
       }
     }
   }
-  wave_count(is_sol, &ctl->sol);
-  const unsigned long long slot = wave_reserve(has_child, &ctl->size);
-  if (has_child) {
-    if (slot >= capacity) {
-      ctl->overflow = 1;
-    } else {
-      emit_pfsp_child(pool, slot, snodes[lp], depth, limit1, k);
-    }
+  uint32_t totC, totS;
+  block_excl_scan(cnt, totC);
+  block_excl_scan(sols, totS);
+  if (threadIdx.x == 0) {
+    blockCounts[blockIdx.x] = totC;
+    blockSols[blockIdx.x] = totS;
   }
-  wave_count(has_child, &ctl->tree);
 }
 
-// lb1_d devpool: one thread per parent; the child loop is wave-uniform
-// (fixed bounds 0..jobs) so wave_reserve stays collective.
+// PFSP lb1_d eval: one thread per parent (O(mn) setup amortized over all its
+// children), labels only — counts come from k_count.
 template <int MM>
-__global__ void k_pfsp_expand_lb1d(DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
-                                   unsigned long long capacity, int jobs, PfspDevTables tb) {
+__global__ void k_pfsp_eval3_lb1d(DevCtl* ctl, const PFSPNode* parents, int jobs,
+                                  PfspDevTables tb, uint8_t* labels) {
   __shared__ LdsLb1<MM> lds;
-  __shared__ PFSPNode snodes[BLOCK];
+  __shared__ PFSPNode s[BLOCK];
   stage_lb1_tables<MM>(lds, tb, jobs);
-
   const unsigned long long c = ctl->chunk;
   const unsigned long long t =
       static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
-  const int lp = stage_parents(parents, c, 1, snodes, t);
+  const int lp = stage_parents(parents, c, 1, s, t);
   __syncthreads();
-  const int best0 = ctl->best;
-  const bool active = lp >= 0;
-
+  if (lp < 0) return;
+  const int best = ctl->best;
+  const PFSPNode& p = s[lp];
+  const int depth = p.depth;
+  const int limit1 = p.limit1;
   int front[MM], remain[MM];
-  int n_sols = 0;
-  int depth = 0, limit1 = 0;
-  if (active) {
-    const PFSPNode& parent = snodes[lp];
-    depth = parent.depth;
-    limit1 = parent.limit1;
-    lb1d_setup<MM>(lds, parent.prmu, limit1, jobs, front, remain);
-  }
-
-  for (int k = 0; k < jobs; k++) {  // uniform length across the wave
-    bool has_child = false;
-    if (active && k >= limit1 + 1) {
-      const int lb = lb1d_child_bound<MM>(lds, front, remain, snodes[lp].prmu[k], jobs);
+  lb1d_setup<MM>(lds, p.prmu, limit1, jobs, front, remain);
+  for (int k = 0; k < jobs; k++) {
+    uint8_t lab = 0;
+    if (k >= limit1 + 1) {
+      const int lb = lb1d_child_bound<MM>(lds, front, remain, p.prmu[k], jobs);
       if (depth + 1 == jobs) {
-        n_sols++;
-        if (lb < best0) atomicMin(&ctl->best, lb);
-      } else if (lb < best0) {
-        has_child = true;
+        lab = 2;
+        if (lb < best) atomicMin(&ctl->best, lb);
+      } else if (lb < best) {
+        lab = 1;
       }
     }
-    const unsigned long long slot = wave_reserve(has_child, &ctl->size);
-    if (has_child) {
-      if (slot >= capacity) {
-        ctl->overflow = 1;
-      } else {
-        emit_pfsp_child(pool, slot, snodes[lp], depth, limit1, k);
-      }
-    }
-    wave_count(has_child, &ctl->tree);
+    labels[t * jobs + k] = lab;
   }
-  if (n_sols > 0) atomicAdd(&ctl->sol, static_cast<unsigned long long>(n_sols));
+}
+
+// Per-emit-block label counts (only needed when eval used a different mapping,
+// i.e. lb1_d).
+__global__ void k_count(const DevCtl* ctl, const uint8_t* labels, int per,
+                        uint32_t* blockCounts, uint32_t* blockSols) {
+  const unsigned long long total = ctl->chunk * per;
+  const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
+  uint32_t cnt = 0, sols = 0;
+#pragma unroll
+  for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
+    const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
+    if (t < total) {
+      const uint8_t v = labels[t];
+      cnt += (v == 1);
+      sols += (v == 2);
+    }
+  }
+  uint32_t totC, totS;
+  block_excl_scan(cnt, totC);
+  block_excl_scan(sols, totS);
+  if (threadIdx.x == 0) {
+    blockCounts[blockIdx.x] = totC;
+    blockSols[blockIdx.x] = totS;
+  }
+}
+
+// Single-block scan over the G per-block counts: absolute pool offsets per
+// emit block + the only writer of size/tree/sol.
+__global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
+                       unsigned long long* blockOffsets, int G, unsigned long long capacity) {
+  __shared__ unsigned long long sh_base;
+  if (threadIdx.x == 0) sh_base = ctl->size;
+  __syncthreads();
+  const unsigned long long base = sh_base;
+  unsigned long long running = 0;
+  uint32_t my_sols = 0;
+  for (int g0 = 0; g0 < G; g0 += BLOCK) {
+    const int i = g0 + threadIdx.x;
+    const uint32_t c = (i < G) ? blockCounts[i] : 0;
+    uint32_t tot;
+    const uint32_t pre = block_excl_scan(c, tot);
+    if (i < G) {
+      blockOffsets[i] = base + running + pre;
+      my_sols += blockSols[i];
+    }
+    running += tot;
+  }
+  uint32_t sol_tot;
+  block_excl_scan(my_sols, sol_tot);
+  if (threadIdx.x == 0) {
+    if (ctl->overflow) return;
+    if (base + running > capacity) {
+      ctl->overflow = 1;
+      return;
+    }
+    ctl->size = base + running;
+    ctl->tree += running;
+    ctl->sol += sol_tot;
+  }
+}
+
+// Ranked child emission: block-local scan of labels + the block's absolute
+// offset; children go straight to their pool slots.
+template <class NodeT, bool IS_NQ, int MAXN>
+__global__ void k_emit(const DevCtl* ctl, const NodeT* parents, NodeT* pool,
+                       const uint8_t* labels, int per,
+                       const unsigned long long* blockOffsets) {
+  __shared__ NodeT s[MAXN];
+  if (ctl->overflow) return;
+  const unsigned long long total = ctl->chunk * per;
+  const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
+  unsigned int first = 0;
+  if (c0 < total) {
+    unsigned long long c1 = c0 + EMIT_TILE;
+    if (c1 > total) c1 = total;
+    first = stage_range(parents, c0, c1, per, s);
+  }
+  __syncthreads();
+  uint8_t lab[EMIT_TILE / BLOCK];
+  uint32_t cnt = 0;
+#pragma unroll
+  for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
+    const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
+    lab[j] = (t < total) ? labels[t] : 0;
+    cnt += (lab[j] == 1);
+  }
+  uint32_t tot;
+  const uint32_t pre = block_excl_scan(cnt, tot);
+  if (cnt == 0) return;
+  unsigned long long slot = blockOffsets[blockIdx.x] + pre;
+#pragma unroll
+  for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
+    if (lab[j] == 1) {
+      const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
+      const unsigned int pid = static_cast<unsigned int>(t / per);
+      const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * per);
+      const NodeT& p = s[pid - first];
+      if constexpr (IS_NQ)
+        emit_nq_child(pool, slot, p, p.depth, k);
+      else
+        emit_pfsp_child(pool, slot, p, p.depth, p.limit1, k);
+      slot++;
+    }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -593,13 +745,6 @@ void launch_nq_eval(const NQNode* parents, int n, int N, int g, uint8_t* labels,
                      dim3(BLOCK), 0, s, parents, n, N, g, labels);
 }
 
-void launch_nq_expand(DevCtl* ctl, const NQNode* parents, NQNode* pool,
-                      unsigned long long capacity, unsigned long long maxChunk, int N, int g,
-                      hipStream_t s) {
-  hipLaunchKernelGGL(k_nq_expand, dim3(grid_for(maxChunk * N)), dim3(BLOCK), 0, s, ctl,
-                     parents, pool, capacity, N, g);
-}
-
 template <int MM>
 static void launch_pfsp_eval_mm(const PFSPNode* parents, int n, int jobs, int lbk,
                                 const PfspDevTables& tb, int best, int32_t* bounds,
@@ -628,31 +773,77 @@ void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, in
     launch_pfsp_eval_mm<20>(parents, n, jobs, lbk, tb, best, bounds, s);
 }
 
+// ---- devpool scan-pipeline launchers ----
+
+static inline int emit_grid(unsigned long long maxChunk, int per) {
+  return static_cast<int>((maxChunk * per + EMIT_TILE - 1) / EMIT_TILE);
+}
+
+void launch_nq_eval3(const DevCtl* ctl, const NQNode* parents, int N, int g, uint8_t* labels,
+                     uint32_t* blockCounts, uint32_t* blockSols, unsigned long long maxChunk,
+                     hipStream_t s) {
+  hipLaunchKernelGGL(k_nq_eval3, dim3(emit_grid(maxChunk, N)), dim3(BLOCK), 0, s, ctl,
+                     parents, N, g, labels, blockCounts, blockSols);
+}
+
 template <int MM>
-static void launch_pfsp_expand_mm(DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
-                                  unsigned long long capacity, unsigned long long maxChunk,
-                                  int jobs, int lbk, const PfspDevTables& tb, hipStream_t s) {
+static void launch_pfsp_eval3_mm(DevCtl* ctl, const PFSPNode* parents, int jobs, int lbk,
+                                 const PfspDevTables& tb, uint8_t* labels,
+                                 uint32_t* blockCounts, uint32_t* blockSols,
+                                 unsigned long long maxChunk, hipStream_t s) {
   if (lbk == 0) {
-    hipLaunchKernelGGL((k_pfsp_expand_lb1d<MM>), dim3(grid_for(maxChunk)), dim3(BLOCK), 0, s,
-                       ctl, parents, pool, capacity, jobs, tb);
+    hipLaunchKernelGGL((k_pfsp_eval3_lb1d<MM>), dim3(grid_for(maxChunk)), dim3(BLOCK), 0, s,
+                       ctl, parents, jobs, tb, labels);
   } else if (lbk == 1) {
-    hipLaunchKernelGGL((k_pfsp_expand<MM, 1>), dim3(grid_for(maxChunk * jobs)), dim3(BLOCK),
-                       0, s, ctl, parents, pool, capacity, jobs, tb);
+    hipLaunchKernelGGL((k_pfsp_eval3<MM, 1>), dim3(emit_grid(maxChunk, jobs)), dim3(BLOCK), 0,
+                       s, ctl, parents, jobs, tb, labels, blockCounts, blockSols);
   } else {
-    hipLaunchKernelGGL((k_pfsp_expand<MM, 2>), dim3(grid_for(maxChunk * jobs)), dim3(BLOCK),
-                       0, s, ctl, parents, pool, capacity, jobs, tb);
+    hipLaunchKernelGGL((k_pfsp_eval3<MM, 2>), dim3(emit_grid(maxChunk, jobs)), dim3(BLOCK), 0,
+                       s, ctl, parents, jobs, tb, labels, blockCounts, blockSols);
   }
 }
 
-void launch_pfsp_expand(DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
-                        unsigned long long capacity, unsigned long long maxChunk, int jobs,
-                        int machines, int lbk, const PfspDevTables& tb, hipStream_t s) {
+void launch_pfsp_eval3(DevCtl* ctl, const PFSPNode* parents, int jobs, int machines, int lbk,
+                       const PfspDevTables& tb, uint8_t* labels, uint32_t* blockCounts,
+                       uint32_t* blockSols, unsigned long long maxChunk, hipStream_t s) {
   if (machines == 5)
-    launch_pfsp_expand_mm<5>(ctl, parents, pool, capacity, maxChunk, jobs, lbk, tb, s);
+    launch_pfsp_eval3_mm<5>(ctl, parents, jobs, lbk, tb, labels, blockCounts, blockSols,
+                            maxChunk, s);
   else if (machines == 10)
-    launch_pfsp_expand_mm<10>(ctl, parents, pool, capacity, maxChunk, jobs, lbk, tb, s);
+    launch_pfsp_eval3_mm<10>(ctl, parents, jobs, lbk, tb, labels, blockCounts, blockSols,
+                             maxChunk, s);
   else
-    launch_pfsp_expand_mm<20>(ctl, parents, pool, capacity, maxChunk, jobs, lbk, tb, s);
+    launch_pfsp_eval3_mm<20>(ctl, parents, jobs, lbk, tb, labels, blockCounts, blockSols,
+                             maxChunk, s);
+}
+
+void launch_count(const DevCtl* ctl, const uint8_t* labels, int per, uint32_t* blockCounts,
+                  uint32_t* blockSols, unsigned long long maxChunk, hipStream_t s) {
+  hipLaunchKernelGGL(k_count, dim3(emit_grid(maxChunk, per)), dim3(BLOCK), 0, s, ctl, labels,
+                     per, blockCounts, blockSols);
+}
+
+void launch_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
+                 unsigned long long* blockOffsets, int G, unsigned long long capacity,
+                 hipStream_t s) {
+  hipLaunchKernelGGL(k_scan, dim3(1), dim3(BLOCK), 0, s, ctl, blockCounts, blockSols,
+                     blockOffsets, G, capacity);
+}
+
+void launch_emit_nq(const DevCtl* ctl, const NQNode* parents, NQNode* pool,
+                    const uint8_t* labels, int N, const unsigned long long* blockOffsets,
+                    unsigned long long maxChunk, hipStream_t s) {
+  hipLaunchKernelGGL((k_emit<NQNode, true, EMIT_TILE + 2>), dim3(emit_grid(maxChunk, N)),
+                     dim3(BLOCK), 0, s, ctl, parents, pool, labels, N, blockOffsets);
+}
+
+void launch_emit_pfsp(const DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
+                      const uint8_t* labels, int jobs,
+                      const unsigned long long* blockOffsets, unsigned long long maxChunk,
+                      hipStream_t s) {
+  hipLaunchKernelGGL((k_emit<PFSPNode, false, EMIT_TILE / 5 + 2>),
+                     dim3(emit_grid(maxChunk, jobs)), dim3(BLOCK), 0, s, ctl, parents, pool,
+                     labels, jobs, blockOffsets);
 }
 
 }  // namespace gats
