@@ -207,11 +207,10 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     }
   } else if (mode == "devpool") {
     DevGuard<NQNode> pool_d(capacity);
-    DevGuard<NQNode> parents_d(M);
     DevGuard<DevCtl> ctl_d(1);
-    const int G = static_cast<int>((static_cast<unsigned long long>(M) * N +
-                                    DEV_EMIT_TILE - 1) / DEV_EMIT_TILE);
-    DevGuard<uint8_t> labels_d(static_cast<size_t>(M) * N);
+    const int G = devpool_grid(M, N, 1);
+    const int stride = devpool_stride(1);
+    DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
     DevGuard<uint32_t> bc_d(G), bs_d(G);
     DevGuard<unsigned long long> boff_d(G);
     const size_t init = pool.size();
@@ -226,13 +225,12 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     r.h2d_bytes += init * sizeof(NQNode) + sizeof(DevCtl);
 
     auto iter = [&] {
-      launch_begin(ctl_d.p, m, M, stream.s);
-      launch_copy_parents_nq(ctl_d.p, pool_d.p, parents_d.p, M, stream.s);
-      launch_nq_eval3(ctl_d.p, parents_d.p, N, g, labels_d.p, bc_d.p, bs_d.p, M, stream.s);
-      launch_scan(ctl_d.p, bc_d.p, bs_d.p, boff_d.p, G, capacity, stream.s);
-      launch_emit_nq(ctl_d.p, parents_d.p, pool_d.p, labels_d.p, N, boff_d.p, M, stream.s);
+      launch_nq_x(ctl_d.p, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, N, g, m, M, stream.s);
+      launch_scan(ctl_d.p, bc_d.p, bs_d.p, boff_d.p, G, m, M, capacity, stream.s);
+      launch_gather_nq(ctl_d.p, bc_d.p, boff_d.p, childbuf_d.p, pool_d.p, stride, G,
+                       stream.s);
     };
-    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 5, iter, r);
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 3, iter, r);
     tree = fin.tree;
     sol = fin.sol;
     r.gpu_iters = fin.iters;
@@ -329,11 +327,10 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     }
   } else if (mode == "devpool") {
     DevGuard<PFSPNode> pool_d(capacity);
-    DevGuard<PFSPNode> parents_d(M);
     DevGuard<DevCtl> ctl_d(1);
-    const int G = static_cast<int>((static_cast<unsigned long long>(M) * jobs +
-                                    DEV_EMIT_TILE - 1) / DEV_EMIT_TILE);
-    DevGuard<uint8_t> labels_d(static_cast<size_t>(M) * jobs);
+    const int G = devpool_grid(M, jobs, lbk);
+    const int stride = devpool_stride(lbk);
+    DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
     DevGuard<uint32_t> bc_d(G), bs_d(G);
     DevGuard<unsigned long long> boff_d(G);
     const size_t init = pool.size();
@@ -350,17 +347,13 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     r.h2d_bytes += init * sizeof(PFSPNode) + sizeof(DevCtl);
 
     auto iter = [&] {
-      launch_begin(ctl_d.p, m, M, stream.s);
-      launch_copy_parents_pfsp(ctl_d.p, pool_d.p, parents_d.p, M, stream.s);
-      launch_pfsp_eval3(ctl_d.p, parents_d.p, jobs, machines, lbk, tables.tb, labels_d.p,
-                        bc_d.p, bs_d.p, M, stream.s);
-      if (lbk == 0)  // lb1_d evaluated per-parent: counts need their own pass
-        launch_count(ctl_d.p, labels_d.p, jobs, bc_d.p, bs_d.p, M, stream.s);
-      launch_scan(ctl_d.p, bc_d.p, bs_d.p, boff_d.p, G, capacity, stream.s);
-      launch_emit_pfsp(ctl_d.p, parents_d.p, pool_d.p, labels_d.p, jobs, boff_d.p, M,
-                       stream.s);
+      launch_pfsp_x(ctl_d.p, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk,
+                    tables.tb, m, M, stream.s);
+      launch_scan(ctl_d.p, bc_d.p, bs_d.p, boff_d.p, G, m, M, capacity, stream.s);
+      launch_gather_pfsp(ctl_d.p, bc_d.p, boff_d.p, childbuf_d.p, pool_d.p, stride, G,
+                         stream.s);
     };
-    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, lbk == 0 ? 6 : 5, iter, r);
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 3, iter, r);
     tree = fin.tree;
     sol = fin.sol;
     best = fin.best;

@@ -36,36 +36,30 @@ struct PfspDevTables {
 };
 
 // Launchers implemented in kernels.hip.
-void launch_begin(DevCtl* ctl, unsigned long long m, unsigned long long M, hipStream_t s);
-void launch_copy_parents_nq(const DevCtl* ctl, const NQNode* pool, NQNode* parents,
-                            unsigned long long maxChunk, hipStream_t s);
-void launch_copy_parents_pfsp(const DevCtl* ctl, const PFSPNode* pool, PFSPNode* parents,
-                              unsigned long long maxChunk, hipStream_t s);
+// hostpool eval (labels/bounds out):
 void launch_nq_eval(const NQNode* parents, int n, int N, int g, uint8_t* labels,
                     hipStream_t s);
 // lbk: 0 = lb1_d, 1 = lb1, 2 = lb2
 void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, int lbk,
                       const PfspDevTables& tb, int best, int32_t* bounds, hipStream_t s);
 
-// devpool scan pipeline (see kernels.hip): eval3 -> [count] -> scan -> emit
-constexpr int DEV_EMIT_TILE = 1024;
-void launch_nq_eval3(const DevCtl* ctl, const NQNode* parents, int N, int g, uint8_t* labels,
-                     uint32_t* blockCounts, uint32_t* blockSols, unsigned long long maxChunk,
-                     hipStream_t s);
-void launch_pfsp_eval3(DevCtl* ctl, const PFSPNode* parents, int jobs, int machines, int lbk,
-                       const PfspDevTables& tb, uint8_t* labels, uint32_t* blockCounts,
-                       uint32_t* blockSols, unsigned long long maxChunk, hipStream_t s);
-void launch_count(const DevCtl* ctl, const uint8_t* labels, int per, uint32_t* blockCounts,
-                  uint32_t* blockSols, unsigned long long maxChunk, hipStream_t s);
+// devpool v3 (see kernels.hip): expand -> scan -> gather, 3 kernels/iteration.
+int devpool_grid(unsigned long long M, int per, int lbk);
+int devpool_stride(int lbk);
+void launch_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
+                 uint32_t* blockCounts, uint32_t* blockSols, int N, int g,
+                 unsigned long long m, unsigned long long M, hipStream_t s);
+void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32_t* bc,
+                   uint32_t* bs, int jobs, int machines, int lbk, const PfspDevTables& tb,
+                   unsigned long long m, unsigned long long M, hipStream_t s);
 void launch_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
-                 unsigned long long* blockOffsets, int G, unsigned long long capacity,
-                 hipStream_t s);
-void launch_emit_nq(const DevCtl* ctl, const NQNode* parents, NQNode* pool,
-                    const uint8_t* labels, int N, const unsigned long long* blockOffsets,
-                    unsigned long long maxChunk, hipStream_t s);
-void launch_emit_pfsp(const DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
-                      const uint8_t* labels, int jobs,
-                      const unsigned long long* blockOffsets, unsigned long long maxChunk,
-                      hipStream_t s);
+                 unsigned long long* blockOffsets, int G, unsigned long long m,
+                 unsigned long long M, unsigned long long capacity, hipStream_t s);
+void launch_gather_nq(const DevCtl* ctl, const uint32_t* bc,
+                      const unsigned long long* boff, const NQNode* childbuf, NQNode* pool,
+                      int strideNodes, int G, hipStream_t s);
+void launch_gather_pfsp(const DevCtl* ctl, const uint32_t* bc,
+                        const unsigned long long* boff, const PFSPNode* childbuf,
+                        PFSPNode* pool, int strideNodes, int G, hipStream_t s);
 
 }  // namespace gats

@@ -36,38 +36,6 @@ namespace gats {
 // Wave64 helpers
 // ---------------------------------------------------------------------------
 
-// Every lane with pred==true gets a unique slot from ONE atomicAdd per wave.
-// Must be executed by all lanes of the wave (uniform control flow).
-__device__ inline unsigned long long wave_reserve(bool pred, unsigned long long* counter) {
-  const unsigned long long mask = __ballot(pred);
-  const int lane = threadIdx.x & 63;
-  const int total = __popcll(mask);
-  unsigned long long base = 0;
-  if (total > 0) {
-    const int leader = __ffsll(static_cast<unsigned long long>(mask)) - 1;
-    if (lane == leader) base = atomicAdd(counter, static_cast<unsigned long long>(total));
-    base = __shfl(base, leader);
-  }
-  const int before = __popcll(mask & ((1ull << lane) - 1ull));
-  return base + static_cast<unsigned long long>(before);
-}
-
-__device__ inline void wave_count(bool pred, unsigned long long* counter) {
-  const unsigned long long mask = __ballot(pred);
-  const int lane = threadIdx.x & 63;
-  if (mask != 0 && lane == __ffsll(static_cast<unsigned long long>(mask)) - 1)
-    atomicAdd(counter, static_cast<unsigned long long>(__popcll(mask)));
-}
-
-// 24-byte node copy as three 8-byte moves (nodes are 8-byte aligned).
-__device__ inline void copy_node(void* dst, const void* src) {
-  const unsigned long long* s = reinterpret_cast<const unsigned long long*>(src);
-  unsigned long long* d = reinterpret_cast<unsigned long long*>(dst);
-  d[0] = s[0];
-  d[1] = s[1];
-  d[2] = s[2];
-}
-
 // Block-cooperative staging of the parents this block touches into LDS.
 // One shared copy per parent (the reference's per-thread `var parent =
 // parents_d[parentId]` would be a 96 B/thread scratch/LDS spill on CDNA4
@@ -122,36 +90,6 @@ __device__ inline void emit_pfsp_child(PFSPNode* pool, unsigned long long slot,
   uint8_t* db = reinterpret_cast<uint8_t*>(d);
   db[2 + depth] = parent.prmu[k];
   db[2 + k] = parent.prmu[depth];
-}
-
-// ---------------------------------------------------------------------------
-// Devpool control kernels (shared by both problems)
-// ---------------------------------------------------------------------------
-
-// begin: decide this iteration's chunk (popBackBulk semantics, Pool.chpl:50-60):
-// pop min(size, M) from the back iff size >= m.
-__global__ void k_begin(DevCtl* ctl, unsigned long long m, unsigned long long M) {
-  if (ctl->overflow) {  // freeze the pool; host will abort at next readback
-    ctl->chunk = 0;
-    return;
-  }
-  unsigned long long size = ctl->size;
-  unsigned long long c = (size >= m) ? (size < M ? size : M) : 0;
-  ctl->chunk = c;
-  ctl->size = size - c;  // parents live at [size-c, size); children overwrite them
-  ctl->iters += (c > 0);
-}
-
-// copy the popped parents out of the pool so expand can append over them.
-template <typename NodeT>
-__global__ void k_copy_parents(const DevCtl* ctl, const NodeT* pool, NodeT* parents) {
-  const unsigned long long words = ctl->chunk * (sizeof(NodeT) / 8);
-  const unsigned long long* src =
-      reinterpret_cast<const unsigned long long*>(pool + ctl->size);
-  unsigned long long* dst = reinterpret_cast<unsigned long long*>(parents);
-  for (unsigned long long i = blockIdx.x * blockDim.x + threadIdx.x; i < words;
-       i += static_cast<unsigned long long>(gridDim.x) * blockDim.x)
-    dst[i] = src[i];
 }
 
 // ---------------------------------------------------------------------------
@@ -419,17 +357,31 @@ __global__ void k_pfsp_eval_lb2(const PFSPNode* parents, int n, int jobs, PfspDe
 
 
 // ---------------------------------------------------------------------------
-// Devpool scan/compact pipeline (atomic-free)
+// Devpool pipeline v3: three kernels per iteration, zero global atomics on
+// the hot path.
 //
-// Per iteration: begin -> copy_parents -> eval3 (labels + per-block counts)
-// [-> count (lb1_d only)] -> scan (single block: offsets + counter update)
-// -> emit (ranked child writes). No same-cacheline global atomics: the
-// first devpool design used one atomicAdd per wave on DevCtl and saturated
-// the L2 atomic unit (~12 ns per op on one line => ~300 us per iteration at
-// chunk 50k). Labels: 0 = pruned/invalid, 1 = push child, 2 = leaf solution.
+//   K1 expand:  every block derives this iteration's chunk c = popBackBulk
+//               semantics (a pure function of ctl->size), reads its parents
+//               straight from the pool tail (LDS-staged), evaluates children
+//               and writes the survivors COMPACTED into its private childbuf
+//               region (block-local exclusive scan for ranks). Per-block
+//               child/solution counts out; no control-block writes.
+//   K2 scan:    single block: exclusive scan of the G per-block counts into
+//               absolute pool offsets; sole writer of size/tree/sol/iters;
+//               overflow guard. (Replaces the separate `begin` kernel: the
+//               pop is re-derived here identically.)
+//   K3 gather:  each block memcpys its children from childbuf into the pool
+//               at its scan offset. Runs after K1 read the parents, so
+//               overwriting the popped tail is safe.
+//
+// History: v1 used one atomicAdd per wave on the DevCtl line and saturated
+// the L2 atomic unit (~12 ns/op on one line -> ~300 us/iter at chunk 50k,
+// 163 Mnodes/s end to end on N=17); v2 (labels + count + scan + emit, 5-6
+// kernels) reached 1992 Mnodes/s; v3 cuts the per-iteration kernel count to
+// 3 since the hot loop is boundary/launch-bound at M = 50000.
 // ---------------------------------------------------------------------------
 
-constexpr int EMIT_TILE = 1024;  // children per block in eval3/count/emit (4/thread)
+constexpr int EMIT_TILE = 1024;  // children per block in per-child kernels (4/thread)
 
 // stage parents covering child range [c0, c1) into LDS; returns the first pid.
 template <class NodeT, int MAXN>
@@ -467,14 +419,28 @@ __device__ inline uint32_t block_excl_scan(uint32_t v, uint32_t& total) {
   return wbase + x - v;
 }
 
-// N-Queens eval: labels + per-block child/solution counts.
-__global__ void k_nq_eval3(const DevCtl* ctl, const NQNode* parents, int N, int g,
-                           uint8_t* labels, uint32_t* blockCounts, uint32_t* blockSols) {
-  __shared__ NQNode s[EMIT_TILE + 2];  // N >= 1
-  const unsigned long long total = ctl->chunk * N;
+// popBackBulk chunk, re-derived identically by every kernel of an iteration
+// (pure function of ctl->size, which only K2 of the previous iteration wrote).
+__device__ inline unsigned long long derive_chunk(const DevCtl* ctl, unsigned long long m,
+                                                  unsigned long long M) {
+  if (ctl->overflow) return 0;
+  const unsigned long long size = ctl->size;
+  if (size < m) return 0;
+  return size < M ? size : M;
+}
+
+// K1 for N-Queens: evaluate + compact children into the block's childbuf slab.
+__global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
+                       uint32_t* blockCounts, uint32_t* blockSols, int N, int g,
+                       unsigned long long m, unsigned long long M) {
+  __shared__ NQNode s[EMIT_TILE + 2];
+  const unsigned long long c = derive_chunk(ctl, m, M);
+  const unsigned long long total = c * N;
+  const NQNode* parents = pool + (ctl->size - c);
   const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
   uint32_t cnt = 0, sols = 0;
   unsigned int first = 0;
+  uint8_t lab[EMIT_TILE / BLOCK] = {0, 0, 0, 0};
   if (c0 < total) {
     unsigned long long c1 = c0 + EMIT_TILE;
     if (c1 > total) c1 = total;
@@ -488,33 +454,43 @@ __global__ void k_nq_eval3(const DevCtl* ctl, const NQNode* parents, int N, int 
         const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * N);
         const NQNode& p = s[pid - first];
         const int depth = p.depth;
-        uint8_t lab = 0;
         if (depth == N) {
-          lab = (k == 0) ? 2 : 0;  // leaf parent counted once (nqueens_chpl.chpl:78-80)
+          lab[j] = (k == 0) ? 2 : 0;  // leaf parent counted once (nqueens_chpl.chpl:78-80)
         } else if (k >= depth && nq_safe(p.board, depth, p.board[k], g)) {
-          lab = 1;
+          lab[j] = 1;
         }
-        labels[t] = lab;
-        cnt += (lab == 1);
-        sols += (lab == 2);
+        cnt += (lab[j] == 1);
+        sols += (lab[j] == 2);
       }
     }
   }
   uint32_t totC, totS;
-  block_excl_scan(cnt, totC);
+  const uint32_t pre = block_excl_scan(cnt, totC);
   block_excl_scan(sols, totS);
   if (threadIdx.x == 0) {
     blockCounts[blockIdx.x] = totC;
     blockSols[blockIdx.x] = totS;
   }
+  if (cnt > 0) {
+    unsigned long long slot = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE + pre;
+#pragma unroll
+    for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
+      if (lab[j] == 1) {
+        const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
+        const unsigned int pid = static_cast<unsigned int>(t / N);
+        const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * N);
+        const NQNode& p = s[pid - first];
+        emit_nq_child(childbuf, slot++, p, p.depth, k);
+      }
+    }
+  }
 }
 
-// PFSP eval (lb1 / lb2): labels + per-block counts; leaves update ctl->best
-// directly (rare: only parents one level above the leaves).
+// K1 for PFSP lb1 / lb2 (one child per thread-slot).
 template <int MM, int LB>
-__global__ void k_pfsp_eval3(DevCtl* ctl, const PFSPNode* parents, int jobs,
-                             PfspDevTables tb, uint8_t* labels, uint32_t* blockCounts,
-                             uint32_t* blockSols) {
+__global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
+                         uint32_t* blockCounts, uint32_t* blockSols, int jobs,
+                         PfspDevTables tb, unsigned long long m, unsigned long long M) {
   using LDS = typename std::conditional<LB == 2, LdsLb2<MM>, LdsLb1<MM>>::type;
   __shared__ LDS lds;
   __shared__ PFSPNode s[EMIT_TILE / 5 + 2];  // jobs >= 5
@@ -522,11 +498,14 @@ __global__ void k_pfsp_eval3(DevCtl* ctl, const PFSPNode* parents, int jobs,
     stage_lb2_tables<MM>(lds, tb, jobs);
   else
     stage_lb1_tables<MM>(lds, tb, jobs);
-  const unsigned long long total = ctl->chunk * jobs;
+  const unsigned long long c = derive_chunk(ctl, m, M);
+  const unsigned long long total = c * jobs;
+  const PFSPNode* parents = pool + (ctl->size - c);
   const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
   const int best = ctl->best;
   uint32_t cnt = 0, sols = 0;
   unsigned int first = 0;
+  uint8_t lab[EMIT_TILE / BLOCK] = {0, 0, 0, 0};
   if (c0 < total) {
     unsigned long long c1 = c0 + EMIT_TILE;
     if (c1 > total) c1 = total;
@@ -541,7 +520,6 @@ __global__ void k_pfsp_eval3(DevCtl* ctl, const PFSPNode* parents, int jobs,
         const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * jobs);
         const PFSPNode& p = s[pid - first];
         const int depth = p.depth;
-        uint8_t lab = 0;
         if (k >= p.limit1 + 1) {
           int lb;
           if constexpr (LB == 2) {
@@ -551,102 +529,107 @@ __global__ void k_pfsp_eval3(DevCtl* ctl, const PFSPNode* parents, int jobs,
             lb = lb1_child_bound<MM>(lds, p.prmu, depth, k, jobs);
           }
           if (depth + 1 == jobs) {
-            lab = 2;
+            lab[j] = 2;
             if (lb < best) atomicMin(&ctl->best, lb);
           } else if (lb < best) {
-            lab = 1;
+            lab[j] = 1;
           }
         }
-        labels[t] = lab;
-        cnt += (lab == 1);
-        sols += (lab == 2);
+        cnt += (lab[j] == 1);
+        sols += (lab[j] == 2);
       }
     }
   }
   uint32_t totC, totS;
-  block_excl_scan(cnt, totC);
+  const uint32_t pre = block_excl_scan(cnt, totC);
   block_excl_scan(sols, totS);
   if (threadIdx.x == 0) {
     blockCounts[blockIdx.x] = totC;
     blockSols[blockIdx.x] = totS;
   }
+  if (cnt > 0) {
+    unsigned long long slot = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE + pre;
+#pragma unroll
+    for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
+      if (lab[j] == 1) {
+        const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
+        const unsigned int pid = static_cast<unsigned int>(t / jobs);
+        const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * jobs);
+        const PFSPNode& p = s[pid - first];
+        emit_pfsp_child(childbuf, slot++, p, p.depth, p.limit1, k);
+      }
+    }
+  }
 }
 
-// PFSP lb1_d eval: one thread per parent (O(mn) setup amortized over all its
-// children), labels only — counts come from k_count.
+// K1 for PFSP lb1_d: one thread per parent (O(mn) setup amortized over all
+// children); bounds are O(m) so the write pass just recomputes them instead
+// of storing a job-indexed local array (which would spill to scratch).
 template <int MM>
-__global__ void k_pfsp_eval3_lb1d(DevCtl* ctl, const PFSPNode* parents, int jobs,
-                                  PfspDevTables tb, uint8_t* labels) {
+__global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
+                              uint32_t* blockCounts, uint32_t* blockSols, int jobs,
+                              PfspDevTables tb, unsigned long long m,
+                              unsigned long long M) {
   __shared__ LdsLb1<MM> lds;
   __shared__ PFSPNode s[BLOCK];
   stage_lb1_tables<MM>(lds, tb, jobs);
-  const unsigned long long c = ctl->chunk;
+  const unsigned long long c = derive_chunk(ctl, m, M);
+  const PFSPNode* parents = pool + (ctl->size - c);
   const unsigned long long t =
       static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
   const int lp = stage_parents(parents, c, 1, s, t);
   __syncthreads();
-  if (lp < 0) return;
   const int best = ctl->best;
-  const PFSPNode& p = s[lp];
-  const int depth = p.depth;
-  const int limit1 = p.limit1;
+
   int front[MM], remain[MM];
-  lb1d_setup<MM>(lds, p.prmu, limit1, jobs, front, remain);
-  for (int k = 0; k < jobs; k++) {
-    uint8_t lab = 0;
-    if (k >= limit1 + 1) {
-      const int lb = lb1d_child_bound<MM>(lds, front, remain, p.prmu[k], jobs);
+  uint32_t cnt = 0, sols = 0;
+  int depth = 0, limit1 = 0;
+  if (lp >= 0) {
+    const PFSPNode& p = s[lp];
+    depth = p.depth;
+    limit1 = p.limit1;
+    lb1d_setup<MM>(lds, p.prmu, limit1, jobs, front, remain);
+    for (int k = limit1 + 1; k < jobs; k++) {
+      const int lb = lb1d_child_bound<MM>(lds, front, remain, s[lp].prmu[k], jobs);
       if (depth + 1 == jobs) {
-        lab = 2;
+        sols++;
         if (lb < best) atomicMin(&ctl->best, lb);
       } else if (lb < best) {
-        lab = 1;
+        cnt++;
       }
-    }
-    labels[t * jobs + k] = lab;
-  }
-}
-
-// Per-emit-block label counts (only needed when eval used a different mapping,
-// i.e. lb1_d).
-__global__ void k_count(const DevCtl* ctl, const uint8_t* labels, int per,
-                        uint32_t* blockCounts, uint32_t* blockSols) {
-  const unsigned long long total = ctl->chunk * per;
-  const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
-  uint32_t cnt = 0, sols = 0;
-#pragma unroll
-  for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
-    const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
-    if (t < total) {
-      const uint8_t v = labels[t];
-      cnt += (v == 1);
-      sols += (v == 2);
     }
   }
   uint32_t totC, totS;
-  block_excl_scan(cnt, totC);
+  const uint32_t pre = block_excl_scan(cnt, totC);
   block_excl_scan(sols, totS);
   if (threadIdx.x == 0) {
     blockCounts[blockIdx.x] = totC;
     blockSols[blockIdx.x] = totS;
   }
+  if (cnt > 0) {
+    unsigned long long slot =
+        static_cast<unsigned long long>(blockIdx.x) * (BLOCK * MAX_JOBS) + pre;
+    for (int k = limit1 + 1; k < jobs; k++) {
+      const int lb = lb1d_child_bound<MM>(lds, front, remain, s[lp].prmu[k], jobs);
+      if (depth + 1 != jobs && lb < best)
+        emit_pfsp_child(childbuf, slot++, s[lp], depth, limit1, k);
+    }
+  }
 }
 
-// Single-block scan over the G per-block counts: absolute pool offsets per
-// emit block + the only writer of size/tree/sol.
+// K2: single-block scan over the G per-block counts; sole DevCtl writer.
 __global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
-                       unsigned long long* blockOffsets, int G, unsigned long long capacity) {
-  __shared__ unsigned long long sh_base;
-  if (threadIdx.x == 0) sh_base = ctl->size;
-  __syncthreads();
-  const unsigned long long base = sh_base;
+                       unsigned long long* blockOffsets, int G, unsigned long long m,
+                       unsigned long long M, unsigned long long capacity) {
+  const unsigned long long c = derive_chunk(ctl, m, M);
+  const unsigned long long base = ctl->size - c;
   unsigned long long running = 0;
   uint32_t my_sols = 0;
   for (int g0 = 0; g0 < G; g0 += BLOCK) {
     const int i = g0 + threadIdx.x;
-    const uint32_t c = (i < G) ? blockCounts[i] : 0;
+    const uint32_t v = (i < G) ? blockCounts[i] : 0;
     uint32_t tot;
-    const uint32_t pre = block_excl_scan(c, tot);
+    const uint32_t pre = block_excl_scan(v, tot);
     if (i < G) {
       blockOffsets[i] = base + running + pre;
       my_sols += blockSols[i];
@@ -661,55 +644,28 @@ __global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t*
       ctl->overflow = 1;
       return;
     }
+    ctl->chunk = c;
+    ctl->iters += (c > 0);
     ctl->size = base + running;
     ctl->tree += running;
     ctl->sol += sol_tot;
   }
 }
 
-// Ranked child emission: block-local scan of labels + the block's absolute
-// offset; children go straight to their pool slots.
-template <class NodeT, bool IS_NQ, int MAXN>
-__global__ void k_emit(const DevCtl* ctl, const NodeT* parents, NodeT* pool,
-                       const uint8_t* labels, int per,
-                       const unsigned long long* blockOffsets) {
-  __shared__ NodeT s[MAXN];
+// K3: block-strided copy of each block's compacted children into the pool.
+template <class NodeT>
+__global__ void k_gather(const DevCtl* ctl, const uint32_t* blockCounts,
+                         const unsigned long long* blockOffsets, const NodeT* childbuf,
+                         NodeT* pool, int strideNodes) {
   if (ctl->overflow) return;
-  const unsigned long long total = ctl->chunk * per;
-  const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
-  unsigned int first = 0;
-  if (c0 < total) {
-    unsigned long long c1 = c0 + EMIT_TILE;
-    if (c1 > total) c1 = total;
-    first = stage_range(parents, c0, c1, per, s);
-  }
-  __syncthreads();
-  uint8_t lab[EMIT_TILE / BLOCK];
-  uint32_t cnt = 0;
-#pragma unroll
-  for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
-    const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
-    lab[j] = (t < total) ? labels[t] : 0;
-    cnt += (lab[j] == 1);
-  }
-  uint32_t tot;
-  const uint32_t pre = block_excl_scan(cnt, tot);
+  const uint32_t cnt = blockCounts[blockIdx.x];
   if (cnt == 0) return;
-  unsigned long long slot = blockOffsets[blockIdx.x] + pre;
-#pragma unroll
-  for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
-    if (lab[j] == 1) {
-      const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
-      const unsigned int pid = static_cast<unsigned int>(t / per);
-      const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * per);
-      const NodeT& p = s[pid - first];
-      if constexpr (IS_NQ)
-        emit_nq_child(pool, slot, p, p.depth, k);
-      else
-        emit_pfsp_child(pool, slot, p, p.depth, p.limit1, k);
-      slot++;
-    }
-  }
+  const unsigned long long* src = reinterpret_cast<const unsigned long long*>(
+      childbuf + static_cast<unsigned long long>(blockIdx.x) * strideNodes);
+  unsigned long long* dst =
+      reinterpret_cast<unsigned long long*>(pool + blockOffsets[blockIdx.x]);
+  const int words = static_cast<int>(cnt) * static_cast<int>(sizeof(NodeT) / 8);
+  for (int i = threadIdx.x; i < words; i += blockDim.x) dst[i] = src[i];
 }
 
 // ---------------------------------------------------------------------------
@@ -718,25 +674,6 @@ __global__ void k_emit(const DevCtl* ctl, const NodeT* parents, NodeT* pool,
 
 static inline int grid_for(unsigned long long threads) {
   return static_cast<int>((threads + BLOCK - 1) / BLOCK);
-}
-
-void launch_begin(DevCtl* ctl, unsigned long long m, unsigned long long M, hipStream_t s) {
-  hipLaunchKernelGGL(k_begin, dim3(1), dim3(1), 0, s, ctl, m, M);
-}
-
-void launch_copy_parents_nq(const DevCtl* ctl, const NQNode* pool, NQNode* parents,
-                            unsigned long long maxChunk, hipStream_t s) {
-  int g = grid_for(maxChunk * 3);
-  if (g > 1024) g = 1024;
-  hipLaunchKernelGGL(k_copy_parents<NQNode>, dim3(g), dim3(BLOCK), 0, s, ctl, pool, parents);
-}
-
-void launch_copy_parents_pfsp(const DevCtl* ctl, const PFSPNode* pool, PFSPNode* parents,
-                              unsigned long long maxChunk, hipStream_t s) {
-  int g = grid_for(maxChunk * 3);
-  if (g > 1024) g = 1024;
-  hipLaunchKernelGGL(k_copy_parents<PFSPNode>, dim3(g), dim3(BLOCK), 0, s, ctl, pool,
-                     parents);
 }
 
 void launch_nq_eval(const NQNode* parents, int n, int N, int g, uint8_t* labels,
@@ -773,77 +710,70 @@ void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, in
     launch_pfsp_eval_mm<20>(parents, n, jobs, lbk, tb, best, bounds, s);
 }
 
-// ---- devpool scan-pipeline launchers ----
+// ---- devpool v3 launchers ----
 
-static inline int emit_grid(unsigned long long maxChunk, int per) {
-  return static_cast<int>((maxChunk * per + EMIT_TILE - 1) / EMIT_TILE);
+int devpool_grid(unsigned long long M, int per, int lbk) {
+  if (lbk == 0)  // lb1_d: one thread per parent
+    return static_cast<int>((M + BLOCK - 1) / BLOCK);
+  return static_cast<int>((M * per + EMIT_TILE - 1) / EMIT_TILE);
 }
 
-void launch_nq_eval3(const DevCtl* ctl, const NQNode* parents, int N, int g, uint8_t* labels,
-                     uint32_t* blockCounts, uint32_t* blockSols, unsigned long long maxChunk,
-                     hipStream_t s) {
-  hipLaunchKernelGGL(k_nq_eval3, dim3(emit_grid(maxChunk, N)), dim3(BLOCK), 0, s, ctl,
-                     parents, N, g, labels, blockCounts, blockSols);
+int devpool_stride(int lbk) { return lbk == 0 ? BLOCK * MAX_JOBS : EMIT_TILE; }
+
+void launch_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
+                 uint32_t* blockCounts, uint32_t* blockSols, int N, int g,
+                 unsigned long long m, unsigned long long M, hipStream_t s) {
+  hipLaunchKernelGGL(k_nq_x, dim3(devpool_grid(M, N, 1)), dim3(BLOCK), 0, s, ctl, pool,
+                     childbuf, blockCounts, blockSols, N, g, m, M);
 }
 
 template <int MM>
-static void launch_pfsp_eval3_mm(DevCtl* ctl, const PFSPNode* parents, int jobs, int lbk,
-                                 const PfspDevTables& tb, uint8_t* labels,
-                                 uint32_t* blockCounts, uint32_t* blockSols,
-                                 unsigned long long maxChunk, hipStream_t s) {
+static void launch_pfsp_x_mm(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
+                             uint32_t* bc, uint32_t* bs, int jobs, int lbk,
+                             const PfspDevTables& tb, unsigned long long m,
+                             unsigned long long M, hipStream_t s) {
   if (lbk == 0) {
-    hipLaunchKernelGGL((k_pfsp_eval3_lb1d<MM>), dim3(grid_for(maxChunk)), dim3(BLOCK), 0, s,
-                       ctl, parents, jobs, tb, labels);
+    hipLaunchKernelGGL((k_pfsp_x_lb1d<MM>), dim3(devpool_grid(M, jobs, 0)), dim3(BLOCK), 0, s,
+                       ctl, pool, childbuf, bc, bs, jobs, tb, m, M);
   } else if (lbk == 1) {
-    hipLaunchKernelGGL((k_pfsp_eval3<MM, 1>), dim3(emit_grid(maxChunk, jobs)), dim3(BLOCK), 0,
-                       s, ctl, parents, jobs, tb, labels, blockCounts, blockSols);
+    hipLaunchKernelGGL((k_pfsp_x<MM, 1>), dim3(devpool_grid(M, jobs, 1)), dim3(BLOCK), 0, s,
+                       ctl, pool, childbuf, bc, bs, jobs, tb, m, M);
   } else {
-    hipLaunchKernelGGL((k_pfsp_eval3<MM, 2>), dim3(emit_grid(maxChunk, jobs)), dim3(BLOCK), 0,
-                       s, ctl, parents, jobs, tb, labels, blockCounts, blockSols);
+    hipLaunchKernelGGL((k_pfsp_x<MM, 2>), dim3(devpool_grid(M, jobs, 2)), dim3(BLOCK), 0, s,
+                       ctl, pool, childbuf, bc, bs, jobs, tb, m, M);
   }
 }
 
-void launch_pfsp_eval3(DevCtl* ctl, const PFSPNode* parents, int jobs, int machines, int lbk,
-                       const PfspDevTables& tb, uint8_t* labels, uint32_t* blockCounts,
-                       uint32_t* blockSols, unsigned long long maxChunk, hipStream_t s) {
+void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32_t* bc,
+                   uint32_t* bs, int jobs, int machines, int lbk, const PfspDevTables& tb,
+                   unsigned long long m, unsigned long long M, hipStream_t s) {
   if (machines == 5)
-    launch_pfsp_eval3_mm<5>(ctl, parents, jobs, lbk, tb, labels, blockCounts, blockSols,
-                            maxChunk, s);
+    launch_pfsp_x_mm<5>(ctl, pool, childbuf, bc, bs, jobs, lbk, tb, m, M, s);
   else if (machines == 10)
-    launch_pfsp_eval3_mm<10>(ctl, parents, jobs, lbk, tb, labels, blockCounts, blockSols,
-                             maxChunk, s);
+    launch_pfsp_x_mm<10>(ctl, pool, childbuf, bc, bs, jobs, lbk, tb, m, M, s);
   else
-    launch_pfsp_eval3_mm<20>(ctl, parents, jobs, lbk, tb, labels, blockCounts, blockSols,
-                             maxChunk, s);
-}
-
-void launch_count(const DevCtl* ctl, const uint8_t* labels, int per, uint32_t* blockCounts,
-                  uint32_t* blockSols, unsigned long long maxChunk, hipStream_t s) {
-  hipLaunchKernelGGL(k_count, dim3(emit_grid(maxChunk, per)), dim3(BLOCK), 0, s, ctl, labels,
-                     per, blockCounts, blockSols);
+    launch_pfsp_x_mm<20>(ctl, pool, childbuf, bc, bs, jobs, lbk, tb, m, M, s);
 }
 
 void launch_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
-                 unsigned long long* blockOffsets, int G, unsigned long long capacity,
-                 hipStream_t s) {
+                 unsigned long long* blockOffsets, int G, unsigned long long m,
+                 unsigned long long M, unsigned long long capacity, hipStream_t s) {
   hipLaunchKernelGGL(k_scan, dim3(1), dim3(BLOCK), 0, s, ctl, blockCounts, blockSols,
-                     blockOffsets, G, capacity);
+                     blockOffsets, G, m, M, capacity);
 }
 
-void launch_emit_nq(const DevCtl* ctl, const NQNode* parents, NQNode* pool,
-                    const uint8_t* labels, int N, const unsigned long long* blockOffsets,
-                    unsigned long long maxChunk, hipStream_t s) {
-  hipLaunchKernelGGL((k_emit<NQNode, true, EMIT_TILE + 2>), dim3(emit_grid(maxChunk, N)),
-                     dim3(BLOCK), 0, s, ctl, parents, pool, labels, N, blockOffsets);
+void launch_gather_nq(const DevCtl* ctl, const uint32_t* bc,
+                      const unsigned long long* boff, const NQNode* childbuf, NQNode* pool,
+                      int strideNodes, int G, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather<NQNode>, dim3(G), dim3(BLOCK), 0, s, ctl, bc, boff, childbuf,
+                     pool, strideNodes);
 }
 
-void launch_emit_pfsp(const DevCtl* ctl, const PFSPNode* parents, PFSPNode* pool,
-                      const uint8_t* labels, int jobs,
-                      const unsigned long long* blockOffsets, unsigned long long maxChunk,
-                      hipStream_t s) {
-  hipLaunchKernelGGL((k_emit<PFSPNode, false, EMIT_TILE / 5 + 2>),
-                     dim3(emit_grid(maxChunk, jobs)), dim3(BLOCK), 0, s, ctl, parents, pool,
-                     labels, jobs, blockOffsets);
+void launch_gather_pfsp(const DevCtl* ctl, const uint32_t* bc,
+                        const unsigned long long* boff, const PFSPNode* childbuf,
+                        PFSPNode* pool, int strideNodes, int G, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather<PFSPNode>, dim3(G), dim3(BLOCK), 0, s, ctl, bc, boff, childbuf,
+                     pool, strideNodes);
 }
 
 }  // namespace gats
