@@ -14,6 +14,7 @@
 //       and appends; the host polls a 64 B control block every few iterations.
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
 #include <cstring>
 #include <stdexcept>
 #include <string>
@@ -141,15 +142,24 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
                                int kernels_per_iter, EnqueueIter&& enqueue_iter, Result& r) {
   PinnedGuard<DevCtl> ctl_h(1);
   const int BATCH = 16;
+  // GATS_NO_GRAPH=1 falls back to eager launches (rocprofv3 crashes tracing
+  // hipGraph replays on ROCm 7.2; eager mode gives identical results).
+  const bool use_graph = std::getenv("GATS_NO_GRAPH") == nullptr;
   hipGraph_t graph = nullptr;
   hipGraphExec_t exec = nullptr;
-  HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
-  for (int b = 0; b < BATCH; b++) enqueue_iter();
-  HIP_CHECK(hipStreamEndCapture(s, &graph));
-  HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
+  if (use_graph) {
+    HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
+    for (int b = 0; b < BATCH; b++) enqueue_iter();
+    HIP_CHECK(hipStreamEndCapture(s, &graph));
+    HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
+  }
   bool overflow = false;
   while (true) {
-    HIP_CHECK(hipGraphLaunch(exec, s));
+    if (use_graph) {
+      HIP_CHECK(hipGraphLaunch(exec, s));
+    } else {
+      for (int b = 0; b < BATCH; b++) enqueue_iter();
+    }
     HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d, sizeof(DevCtl), hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     r.kernel_launch += static_cast<uint64_t>(kernels_per_iter) * BATCH;
@@ -161,8 +171,10 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
     }
     if (ctl_h.p->size < m) break;
   }
-  (void)hipGraphExecDestroy(exec);
-  (void)hipGraphDestroy(graph);
+  if (use_graph) {
+    (void)hipGraphExecDestroy(exec);
+    (void)hipGraphDestroy(graph);
+  }
   if (overflow) throw std::runtime_error("device pool overflow; raise capacity");
   return *ctl_h.p;
 }

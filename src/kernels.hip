@@ -618,6 +618,10 @@ __global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
 }
 
 // K2: single-block scan over the G per-block counts; sole DevCtl writer.
+// Each thread owns SCAN_PER consecutive counts so one block-scan round covers
+// G <= BLOCK*SCAN_PER blocks (the scan is on the critical path every
+// iteration; multi-round versions were ~2x slower at G ~ 1000).
+constexpr int SCAN_PER = 8;
 __global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
                        unsigned long long* blockOffsets, int G, unsigned long long m,
                        unsigned long long M, unsigned long long capacity) {
@@ -625,14 +629,23 @@ __global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t*
   const unsigned long long base = ctl->size - c;
   unsigned long long running = 0;
   uint32_t my_sols = 0;
-  for (int g0 = 0; g0 < G; g0 += BLOCK) {
-    const int i = g0 + threadIdx.x;
-    const uint32_t v = (i < G) ? blockCounts[i] : 0;
+  for (int g0 = 0; g0 < G; g0 += BLOCK * SCAN_PER) {
+    uint32_t v[SCAN_PER];
+    uint32_t mine = 0;
+#pragma unroll
+    for (int j = 0; j < SCAN_PER; j++) {
+      const int i = g0 + threadIdx.x * SCAN_PER + j;
+      v[j] = (i < G) ? blockCounts[i] : 0;
+      mine += v[j];
+      if (i < G) my_sols += blockSols[i];
+    }
     uint32_t tot;
-    const uint32_t pre = block_excl_scan(v, tot);
-    if (i < G) {
-      blockOffsets[i] = base + running + pre;
-      my_sols += blockSols[i];
+    uint32_t pre = block_excl_scan(mine, tot);
+#pragma unroll
+    for (int j = 0; j < SCAN_PER; j++) {
+      const int i = g0 + threadIdx.x * SCAN_PER + j;
+      if (i < G) blockOffsets[i] = base + running + pre;
+      pre += v[j];
     }
     running += tot;
   }
