@@ -218,6 +218,8 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
       nq_generate_children(parents.p, n, N, labels.p, tree, sol, pool);
     }
   } else if (mode == "devpool") {
+    if (static_cast<unsigned long long>(M) * N > (1ull << 31))
+      throw std::invalid_argument("devpool requires M * N <= 2^31");
     DevGuard<NQNode> pool_d(capacity);
     DevGuard<DevCtl> ctl_d(1);
     const int G = devpool_grid(M, N, 1);
@@ -338,6 +340,8 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       pfsp_generate_children(I, parents.p, n, bounds.p, tree, sol, best, pool);
     }
   } else if (mode == "devpool") {
+    if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
+      throw std::invalid_argument("devpool requires M * jobs <= 2^31");
     DevGuard<PFSPNode> pool_d(capacity);
     DevGuard<DevCtl> ctl_d(1);
     const int G = devpool_grid(M, jobs, lbk);

@@ -435,23 +435,28 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                        unsigned long long m, unsigned long long M) {
   __shared__ NQNode s[EMIT_TILE + 2];
   const unsigned long long c = derive_chunk(ctl, m, M);
-  const unsigned long long total = c * N;
+  // child indices fit u32: the engine enforces M * branching <= 2^31
+  const uint32_t total = static_cast<uint32_t>(c * N);
   const NQNode* parents = pool + (ctl->size - c);
-  const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
+  const uint32_t c0 = static_cast<uint32_t>(blockIdx.x) * EMIT_TILE;
   uint32_t cnt = 0, sols = 0;
   unsigned int first = 0;
   uint8_t lab[EMIT_TILE / BLOCK] = {0, 0, 0, 0};
+  uint16_t lpid[EMIT_TILE / BLOCK];
+  uint8_t lk[EMIT_TILE / BLOCK];
   if (c0 < total) {
-    unsigned long long c1 = c0 + EMIT_TILE;
+    uint32_t c1 = c0 + EMIT_TILE;
     if (c1 > total) c1 = total;
     first = stage_range(parents, c0, c1, N, s);
     __syncthreads();
 #pragma unroll
     for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
-      const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
+      const uint32_t t = c0 + j * BLOCK + threadIdx.x;
       if (t < total) {
-        const unsigned int pid = static_cast<unsigned int>(t / N);
-        const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * N);
+        const uint32_t pid = t / static_cast<uint32_t>(N);
+        const int k = static_cast<int>(t - pid * N);
+        lpid[j] = static_cast<uint16_t>(pid - first);
+        lk[j] = static_cast<uint8_t>(k);
         const NQNode& p = s[pid - first];
         const int depth = p.depth;
         if (depth == N) {
@@ -476,11 +481,8 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
 #pragma unroll
     for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
       if (lab[j] == 1) {
-        const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
-        const unsigned int pid = static_cast<unsigned int>(t / N);
-        const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * N);
-        const NQNode& p = s[pid - first];
-        emit_nq_child(childbuf, slot++, p, p.depth, k);
+        const NQNode& p = s[lpid[j]];
+        emit_nq_child(childbuf, slot++, p, p.depth, lk[j]);
       }
     }
   }
@@ -499,25 +501,29 @@ __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
   else
     stage_lb1_tables<MM>(lds, tb, jobs);
   const unsigned long long c = derive_chunk(ctl, m, M);
-  const unsigned long long total = c * jobs;
+  const uint32_t total = static_cast<uint32_t>(c * jobs);
   const PFSPNode* parents = pool + (ctl->size - c);
-  const unsigned long long c0 = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE;
+  const uint32_t c0 = static_cast<uint32_t>(blockIdx.x) * EMIT_TILE;
   const int best = ctl->best;
   uint32_t cnt = 0, sols = 0;
   unsigned int first = 0;
   uint8_t lab[EMIT_TILE / BLOCK] = {0, 0, 0, 0};
+  uint16_t lpid[EMIT_TILE / BLOCK];
+  uint8_t lk[EMIT_TILE / BLOCK];
   if (c0 < total) {
-    unsigned long long c1 = c0 + EMIT_TILE;
+    uint32_t c1 = c0 + EMIT_TILE;
     if (c1 > total) c1 = total;
     first = stage_range(parents, c0, c1, jobs, s);
   }
   __syncthreads();
   if (c0 < total) {
     for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
-      const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
+      const uint32_t t = c0 + j * BLOCK + threadIdx.x;
       if (t < total) {
-        const unsigned int pid = static_cast<unsigned int>(t / jobs);
-        const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * jobs);
+        const uint32_t pid = t / static_cast<uint32_t>(jobs);
+        const int k = static_cast<int>(t - pid * jobs);
+        lpid[j] = static_cast<uint16_t>(pid - first);
+        lk[j] = static_cast<uint8_t>(k);
         const PFSPNode& p = s[pid - first];
         const int depth = p.depth;
         if (k >= p.limit1 + 1) {
@@ -552,11 +558,8 @@ __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
 #pragma unroll
     for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
       if (lab[j] == 1) {
-        const unsigned long long t = c0 + j * BLOCK + threadIdx.x;
-        const unsigned int pid = static_cast<unsigned int>(t / jobs);
-        const int k = static_cast<int>(t - static_cast<unsigned long long>(pid) * jobs);
-        const PFSPNode& p = s[pid - first];
-        emit_pfsp_child(childbuf, slot++, p, p.depth, p.limit1, k);
+        const PFSPNode& p = s[lpid[j]];
+        emit_pfsp_child(childbuf, slot++, p, p.depth, p.limit1, lk[j]);
       }
     }
   }
