@@ -223,9 +223,12 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     DevGuard<NQNode> pool_d(capacity);
     DevGuard<DevCtl> ctl_d(1);
     const int G = devpool_grid(M, N, 1);
-    const int stride = devpool_stride(1);
+    // two-level expand (default): each child slot can spawn up to N-1 pushes
+    const bool two_level = std::getenv("GATS_NQ_1LEVEL") == nullptr;
+    const int stride =
+        two_level ? devpool_stride(1) * (MAX_JOBS - 1) : devpool_stride(1);
     DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
-    DevGuard<uint32_t> bc_d(G), bs_d(G);
+    DevGuard<uint32_t> bc_d(G), bs_d(G), be_d(G);
     DevGuard<unsigned long long> boff_d(G);
     const size_t init = pool.size();
     if (init > capacity) throw std::runtime_error("devpool capacity too small");
@@ -239,8 +242,13 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     r.h2d_bytes += init * sizeof(NQNode) + sizeof(DevCtl);
 
     auto iter = [&] {
-      launch_nq_x(ctl_d.p, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, N, g, m, M, stream.s);
-      launch_scan(ctl_d.p, bc_d.p, bs_d.p, boff_d.p, G, m, M, capacity, stream.s);
+      if (two_level)
+        launch_nq_x2(ctl_d.p, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, m, M,
+                     stream.s);
+      else
+        launch_nq_x(ctl_d.p, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, N, g, m, M, stream.s);
+      launch_scan(ctl_d.p, bc_d.p, bs_d.p, two_level ? be_d.p : nullptr, boff_d.p, G, m, M,
+                  capacity, stream.s);
       launch_gather_nq(ctl_d.p, bc_d.p, boff_d.p, childbuf_d.p, pool_d.p, stride, G,
                        stream.s);
     };
@@ -365,7 +373,7 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     auto iter = [&] {
       launch_pfsp_x(ctl_d.p, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk,
                     tables.tb, m, M, stream.s);
-      launch_scan(ctl_d.p, bc_d.p, bs_d.p, boff_d.p, G, m, M, capacity, stream.s);
+      launch_scan(ctl_d.p, bc_d.p, bs_d.p, nullptr, boff_d.p, G, m, M, capacity, stream.s);
       launch_gather_pfsp(ctl_d.p, bc_d.p, boff_d.p, childbuf_d.p, pool_d.p, stride, G,
                          stream.s);
     };

@@ -488,6 +488,118 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
   }
 }
 
+
+// Two-level N-Queens expand: one kernel evaluates a popped parent's children
+// AND grandchildren, pushing only the grandchildren. Counts are unchanged
+// versus one level at a time: a safe child contributes tree+1 (blockExtra); a
+// depth-N child contributes tree+1 and sol+1 immediately (the reference
+// pushes it and counts sol at its pop, nqueens_chpl.chpl:78-80 — same total);
+// pushed grandchildren contribute tree+1 at the scan like any push. Halves
+// the number of sequential iterations of the (latency-bound) hot loop.
+__global__ void k_nq_x2(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
+                        uint32_t* blockCounts, uint32_t* blockSols, uint32_t* blockExtra,
+                        int N, int g, unsigned long long m, unsigned long long M) {
+  __shared__ NQNode s[EMIT_TILE + 2];
+  const unsigned long long c = derive_chunk(ctl, m, M);
+  const uint32_t total = static_cast<uint32_t>(c * N);
+  const NQNode* parents = pool + (ctl->size - c);
+  const uint32_t c0 = static_cast<uint32_t>(blockIdx.x) * EMIT_TILE;
+  uint32_t cnt = 0, sols = 0, extra = 0;
+  unsigned int first = 0;
+  uint32_t gmask[EMIT_TILE / BLOCK] = {0, 0, 0, 0};  // bit k2 = push grandchild
+  uint16_t lpid[EMIT_TILE / BLOCK];
+  uint8_t lk[EMIT_TILE / BLOCK];
+  if (c0 < total) {
+    uint32_t c1 = c0 + EMIT_TILE;
+    if (c1 > total) c1 = total;
+    first = stage_range(parents, c0, c1, N, s);
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
+      const uint32_t t = c0 + j * BLOCK + threadIdx.x;
+      if (t < total) {
+        const uint32_t pid = t / static_cast<uint32_t>(N);
+        const int k = static_cast<int>(t - pid * N);
+        lpid[j] = static_cast<uint16_t>(pid - first);
+        lk[j] = static_cast<uint8_t>(k);
+        const NQNode& p = s[pid - first];
+        const int depth = p.depth;
+        if (depth == N) {
+          sols += (k == 0);
+        } else if (k >= depth && nq_safe(p.board, depth, p.board[k], g)) {
+          extra += 1;  // the child itself (counted, not pushed)
+          if (depth + 1 == N) {
+            sols += 1;  // depth-N child: tree+sol immediately
+          } else {
+            // grandchildren: child board = parent board with (depth,k) swapped
+            for (int k2 = depth + 1; k2 < N; k2++) {
+              const int q = (k2 == k) ? p.board[depth] : p.board[k2];
+              // safety vs child columns [0, depth+1): i < depth are parent
+              // columns (board[i], i != k since k >= depth+1 when distinct...
+              // k == depth means identity swap), i == depth is board[k]
+              uint8_t safe = 1;
+              for (int i = 0; i < depth; i++) {
+                const int o = p.board[i];
+                for (int r = 0; r < g; r++)
+                  safe &= (o != q - (depth + 1 - i)) & (o != q + (depth + 1 - i));
+              }
+              {
+                const int o = p.board[k];  // child column at position depth
+                for (int r = 0; r < g; r++) safe &= (o != q - 1) & (o != q + 1);
+              }
+              if (safe) gmask[j] |= 1u << k2;
+            }
+            cnt += __popc(gmask[j]);
+          }
+        }
+      }
+    }
+  }
+  uint32_t totC, totS, totE;
+  const uint32_t pre = block_excl_scan(cnt, totC);
+  block_excl_scan(sols, totS);
+  block_excl_scan(extra, totE);
+  if (threadIdx.x == 0) {
+    blockCounts[blockIdx.x] = totC;
+    blockSols[blockIdx.x] = totS;
+    blockExtra[blockIdx.x] = totE;
+  }
+  if (cnt > 0) {
+    unsigned long long slot =
+        static_cast<unsigned long long>(blockIdx.x) * (EMIT_TILE * (MAX_JOBS - 1)) + pre;
+#pragma unroll
+    for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
+      uint32_t mask = gmask[j];
+      if (mask == 0) continue;
+      const NQNode& p = s[lpid[j]];
+      const int depth = p.depth;
+      const int k = lk[j];
+      while (mask) {
+        const int k2 = __ffs(mask) - 1;
+        mask &= mask - 1;
+        // grandchild = parent with swaps (depth,k) then (depth+1,k2) applied;
+        // write parent qwords, patch depth byte + the affected positions with
+        // map1(map2(pos)) values (idempotent per position)
+        const unsigned long long* sp = reinterpret_cast<const unsigned long long*>(&p);
+        unsigned long long* d = reinterpret_cast<unsigned long long*>(&childbuf[slot]);
+        d[0] = (sp[0] & ~0xFFull) | static_cast<unsigned long long>(depth + 2);
+        d[1] = sp[1];
+        d[2] = sp[2];
+        uint8_t* db = reinterpret_cast<uint8_t*>(d);
+        auto map1 = [&](int pos) { return pos == depth ? k : (pos == k ? depth : pos); };
+        auto map2 = [&](int pos) {
+          return pos == depth + 1 ? k2 : (pos == k2 ? depth + 1 : pos);
+        };
+        db[1 + depth] = p.board[map1(map2(depth))];
+        db[1 + k] = p.board[map1(map2(k))];
+        db[1 + depth + 1] = p.board[map1(map2(depth + 1))];
+        db[1 + k2] = p.board[map1(map2(k2))];
+        slot++;
+      }
+    }
+  }
+}
+
 // K1 for PFSP lb1 / lb2 (one child per thread-slot).
 template <int MM, int LB>
 __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
@@ -626,12 +738,13 @@ __global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
 // iteration; multi-round versions were ~2x slower at G ~ 1000).
 constexpr int SCAN_PER = 8;
 __global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
-                       unsigned long long* blockOffsets, int G, unsigned long long m,
-                       unsigned long long M, unsigned long long capacity) {
+                       const uint32_t* blockExtra, unsigned long long* blockOffsets, int G,
+                       unsigned long long m, unsigned long long M,
+                       unsigned long long capacity) {
   const unsigned long long c = derive_chunk(ctl, m, M);
   const unsigned long long base = ctl->size - c;
   unsigned long long running = 0;
-  uint32_t my_sols = 0;
+  uint32_t my_sols = 0, my_extra = 0;
   for (int g0 = 0; g0 < G; g0 += BLOCK * SCAN_PER) {
     uint32_t v[SCAN_PER];
     uint32_t mine = 0;
@@ -640,7 +753,10 @@ __global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t*
       const int i = g0 + threadIdx.x * SCAN_PER + j;
       v[j] = (i < G) ? blockCounts[i] : 0;
       mine += v[j];
-      if (i < G) my_sols += blockSols[i];
+      if (i < G) {
+        my_sols += blockSols[i];
+        if (blockExtra) my_extra += blockExtra[i];
+      }
     }
     uint32_t tot;
     uint32_t pre = block_excl_scan(mine, tot);
@@ -652,8 +768,9 @@ __global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t*
     }
     running += tot;
   }
-  uint32_t sol_tot;
+  uint32_t sol_tot, extra_tot;
   block_excl_scan(my_sols, sol_tot);
+  block_excl_scan(my_extra, extra_tot);
   if (threadIdx.x == 0) {
     if (ctl->overflow) return;
     if (base + running > capacity) {
@@ -663,7 +780,7 @@ __global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t*
     ctl->chunk = c;
     ctl->iters += (c > 0);
     ctl->size = base + running;
-    ctl->tree += running;
+    ctl->tree += running + extra_tot;
     ctl->sol += sol_tot;
   }
 }
@@ -743,6 +860,13 @@ void launch_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                      childbuf, blockCounts, blockSols, N, g, m, M);
 }
 
+void launch_nq_x2(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
+                  uint32_t* blockCounts, uint32_t* blockSols, uint32_t* blockExtra, int N,
+                  int g, unsigned long long m, unsigned long long M, hipStream_t s) {
+  hipLaunchKernelGGL(k_nq_x2, dim3(devpool_grid(M, N, 1)), dim3(BLOCK), 0, s, ctl, pool,
+                     childbuf, blockCounts, blockSols, blockExtra, N, g, m, M);
+}
+
 template <int MM>
 static void launch_pfsp_x_mm(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
                              uint32_t* bc, uint32_t* bs, int jobs, int lbk,
@@ -772,10 +896,11 @@ void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32
 }
 
 void launch_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
-                 unsigned long long* blockOffsets, int G, unsigned long long m,
-                 unsigned long long M, unsigned long long capacity, hipStream_t s) {
+                 const uint32_t* blockExtra, unsigned long long* blockOffsets, int G,
+                 unsigned long long m, unsigned long long M, unsigned long long capacity,
+                 hipStream_t s) {
   hipLaunchKernelGGL(k_scan, dim3(1), dim3(BLOCK), 0, s, ctl, blockCounts, blockSols,
-                     blockOffsets, G, m, M, capacity);
+                     blockExtra, blockOffsets, G, m, M, capacity);
 }
 
 void launch_gather_nq(const DevCtl* ctl, const uint32_t* bc,
