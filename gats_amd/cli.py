@@ -45,6 +45,26 @@ def _results(r, optimum=None, init_ub=None):
     print("=================================================\n")
 
 
+def _workload(r):
+    # workload-share print parity (nqueens_multigpu_chpl.chpl:337)
+    w = r.get("per_worker_tree")
+    if not w:
+        return
+    total = sum(w) or 1
+    shares = ", ".join(f"{100.0 * t / total:.2f}" for t in w)
+    print("workload per GPU [%]: ", shares)
+
+
+def _stats_line(path, args, r):
+    """Append-only results ledger, reference line format parity
+    (pfsp_gpu_cuda.c:140-148: `ta%d lb%d S-GPU %.4f %llu %llu %d`)."""
+    tier_tag = {"seq": "SEQ", "gpu": "S-GPU", "multigpu": "M-GPU", "dist": "D-GPU"}[args.tier]
+    lb_num = {"lb1": 1, "lb1_d": 0, "lb2": 2}[args.lb]
+    with open(path, "a") as f:
+        f.write(f"ta{args.inst} lb{lb_num} {tier_tag} {r['time']:.4f} "
+                f"{r['tree']} {r['sol']} {r['optimum']}\n")
+
+
 def _diag(r):
     d = r.get("diag")
     if not d or d["kernel_launch"] == 0:
@@ -66,6 +86,8 @@ def add_common(p):
     p.add_argument("--mode", default="devpool", choices=["devpool", "hostpool"])
     p.add_argument("--capacity", type=int, default=1 << 27,
                    help="devpool device-pool capacity in nodes (288 GB HBM3E)")
+    p.add_argument("--stats-file", default=None,
+                   help="append a result line (reference stats_*.dat parity)")
 
 
 def main(argv=None):
@@ -113,6 +135,7 @@ def main(argv=None):
             if r is None:
                 return 0
         _results(r)
+        _workload(r)
         _diag(r)
     else:
         inst = args.inst
@@ -144,7 +167,10 @@ def main(argv=None):
             if r is None:
                 return 0
         _results(r, r["optimum"], init_ub)
+        _workload(r)
         _diag(r)
+        if args.stats_file:
+            _stats_line(args.stats_file, args, r)
 
     print("\nExploration terminated.")
     return 0

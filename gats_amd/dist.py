@@ -100,12 +100,14 @@ def run_nqueens(N, g=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
     return _reduce_stats(r, phase1, local_dev, world)
 
 
-def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
+def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 24,
              frontier_target=None, engine="gpu"):
     c = gats_amd.core()
     rank, world = init_dist()
     if frontier_target is None:
-        frontier_target = max(65536, 8192 * world)
+        # PFSP 20-job trees are small (ta014 lb1 ~2.6M nodes); a deep frontier
+        # would move a large share of the search onto the single-threaded CPU
+        frontier_target = max(8192, 2048 * world)
     nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
     my = slice_frontier(nodes, rank, world)
     phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}

@@ -41,6 +41,11 @@ py::dict result_to_dict(const Result& r) {
   diag["gpu_iters"] = r.gpu_iters;
   diag["gpu_time"] = r.gpu_time;
   d["diag"] = diag;
+  if (!r.per_worker.empty()) {
+    py::list w;
+    for (uint64_t v : r.per_worker) w.append(v);
+    d["per_worker_tree"] = w;
+  }
   return d;
 }
 

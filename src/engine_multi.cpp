@@ -301,6 +301,7 @@ Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& ev
   for (auto& dg : diags) {
     tree2 += dg.tree;
     sol2 += dg.sol;
+    r.per_worker.push_back(dg.tree);
     r.kernel_launch += dg.kernel_launch;
     r.h2d += dg.h2d;
     r.d2h += dg.d2h;
@@ -455,6 +456,7 @@ Result pfsp_multigpu(int inst, const std::string& lb_str, int ub, int m, int M, 
   for (auto& dg : diags) {
     tree2 += dg.tree;
     sol2 += dg.sol;
+    r.per_worker.push_back(dg.tree);
     if (dg.best < best) best = dg.best;  // min-reduce (pfsp_multigpu_chpl.chpl:520)
     r.kernel_launch += dg.kernel_launch;
     r.h2d += dg.h2d;

@@ -30,6 +30,9 @@ struct Result {
   uint64_t h2d_bytes = 0, d2h_bytes = 0;
   uint64_t gpu_iters = 0;
   double gpu_time = 0.0;  // cumulative device-loop wall time
+  std::vector<uint64_t> per_worker;  // multi-GPU tier: explored tree per worker
+                                     // (workload-share print parity,
+                                     // nqueens_multigpu_chpl.chpl:337)
 };
 
 enum class LbKind { LB1, LB1_D, LB2 };
