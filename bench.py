@@ -44,12 +44,12 @@ def parse_args():
 def main():
     args = parse_args()
     c = gats_amd.core()
-    rank, world = gdist.init_dist()
     ndev = c.gpu_device_count()
     if ndev == 0:
         raise RuntimeError("bench.py needs an MI355X (no HIP device visible)")
-    local = int(os.environ.get("LOCAL_RANK", rank)) % ndev
-    torch.cuda.set_device(local)
+    local = int(os.environ.get("LOCAL_RANK", "0")) % ndev
+    torch.cuda.set_device(local)  # before NCCL/RCCL process-group init
+    rank, world = gdist.init_dist()
 
     def step():
         if args.problem == "nqueens":

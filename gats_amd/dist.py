@@ -126,6 +126,51 @@ def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 
     return _reduce_stats(r, phase1, local_dev, world, best=True)
 
 
+def run_pfsp_shared_ub(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
+                       frontier_target=None, poll_s=0.02):
+    """PFSP distributed search with incumbent-UB exchange DURING the search:
+    every rank runs the devpool engine on a background thread and joins a
+    fixed-cadence all_reduce(MIN on best, SUM on still-running) loop — every
+    rank executes the same number of collectives, so no rank can deadlock.
+    The reference only min-reduces at the end (pfsp_dist_multigpu_cuda.c:694);
+    exchanging earlier strictly tightens pruning and stays correct (any
+    incumbent >= optimum is a valid UB)."""
+    import time
+
+    c = gats_amd.core()
+    rank, world = init_dist()
+    if frontier_target is None:
+        frontier_target = max(8192, 2048 * world)
+    nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
+    my = slice_frontier(nodes, rank, world)
+    local = rank % max(1, c.gpu_device_count())
+    t0 = time.perf_counter()
+    eng = c.PfspAsyncEngine(my, inst, lb, ub, best, m, M, local, capacity)
+    dev = torch.device(f"cuda:{local}") if torch.cuda.is_available() else torch.device("cpu")
+    if world > 1:
+        while True:
+            t = torch.tensor([eng.best(), 0 if eng.done() else 1], dtype=torch.int64,
+                             device=dev)
+            tb = t[:1].clone()
+            td.all_reduce(tb, op=td.ReduceOp.MIN)
+            tr = t[1:].clone()
+            td.all_reduce(tr, op=td.ReduceOp.SUM)
+            eng.update_best(int(tb.item()))
+            if int(tr.item()) == 0:
+                break
+            time.sleep(poll_s)
+    r = eng.join()
+    elapsed = time.perf_counter() - t0
+    r = dict(r)
+    r["time"] = elapsed
+    if world == 1:
+        r["tree"] += tree1
+        r["sol"] += sol1
+        return r
+    phase1 = {"tree": tree1, "sol": sol1, "time": 0.0}
+    return _reduce_stats(r, phase1, f"cuda:{local}", world, best=True)
+
+
 def run_from_cli(args):
     """Entry for `gats-amd ... --tier dist` under torchrun; rank 0 returns the
     combined stats dict, other ranks return None."""

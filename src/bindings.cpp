@@ -336,4 +336,27 @@ PYBIND11_MODULE(_core, mod) {
           py::arg("device") = 0);
 
   mod.def("pool_selftest", &pool_selftest);
+
+  py::class_<PfspAsyncEngine>(mod, "PfspAsyncEngine",
+                              "Background PFSP devpool search with a shared incumbent "
+                              "(mid-search RCCL UB exchange hook)")
+      .def(py::init([](const py::bytes& nodes, int inst, const std::string& lb, int ub,
+                       int best0, int m, int M, int device, unsigned long long capacity) {
+             return new PfspAsyncEngine(nodes_from_bytes<PFSPNode>(nodes), inst, lb, ub,
+                                        best0, m, M, device, capacity);
+           }),
+           py::arg("nodes"), py::arg("inst"), py::arg("lb") = "lb1", py::arg("ub") = 1,
+           py::arg("best0") = 0, py::arg("m") = 25, py::arg("M") = 50000,
+           py::arg("device") = 0, py::arg("capacity") = (1ull << 24))
+      .def("best", &PfspAsyncEngine::best)
+      .def("update_best", &PfspAsyncEngine::update_best)
+      .def("done", &PfspAsyncEngine::done)
+      .def("join", [](PfspAsyncEngine& e) {
+        Result r;
+        {
+          py::gil_scoped_release rel;
+          r = e.join();
+        }
+        return result_to_dict(r);
+      });
 }

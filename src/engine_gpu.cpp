@@ -14,12 +14,14 @@
 //       and appends; the host polls a 64 B control block every few iterations.
 #include <hip/hip_runtime.h>
 
+#include <atomic>
 #include <cstdlib>
 #include <cstring>
 #include <stdexcept>
 #include <string>
 #include <vector>
 
+#include "engine_gpu.hpp"
 #include "gpu_api.hpp"
 #include "search_host.hpp"
 
@@ -137,9 +139,14 @@ int lbk_of(LbKind lb) {
 // replay + poll the 64 B control block until the pool drops below m
 // (graph replay ~10-16 us vs ~3.5 us host cost PER LAUNCH eager — the hot
 // loop is launch-bound at chunk sizes this small).
+// `shared_best` (optional): a cross-thread incumbent. Each readback publishes
+// the engine's best into it (atomic min) and adopts a lower value published by
+// other ranks/threads by writing ctl->best on the (synchronized) stream —
+// the RCCL incumbent-UB exchange of the distributed tier plugs in here.
 template <class EnqueueIter>
 static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long m,
-                               int kernels_per_iter, EnqueueIter&& enqueue_iter, Result& r) {
+                               int kernels_per_iter, EnqueueIter&& enqueue_iter, Result& r,
+                               std::atomic<int>* shared_best = nullptr) {
   PinnedGuard<DevCtl> ctl_h(1);
   const int BATCH = 16;
   // GATS_NO_GRAPH=1 falls back to eager launches (rocprofv3 crashes tracing
@@ -168,6 +175,20 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
     if (ctl_h.p->overflow) {
       overflow = true;
       break;
+    }
+    if (shared_best) {
+      int mine = ctl_h.p->best;
+      int cur = shared_best->load(std::memory_order_relaxed);
+      while (mine < cur && !shared_best->compare_exchange_weak(cur, mine,
+                                                               std::memory_order_relaxed)) {
+      }
+      cur = shared_best->load(std::memory_order_relaxed);
+      if (cur < mine) {
+        // adopt the lower incumbent; the stream is idle (synchronized above),
+        // so no device atomicMin can race this write
+        HIP_CHECK(hipMemcpyAsync(&ctl_d->best, &cur, sizeof(int), hipMemcpyHostToDevice, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+      }
     }
     if (ctl_h.p->size < m) break;
   }
@@ -311,7 +332,8 @@ Result nqueens_gpu_from_pool(const std::vector<NQNode>& nodes, int N, int g, int
 
 Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int m, int M,
                     int device, const std::string& mode, uint64_t tree0, uint64_t sol0,
-                    int best0, double phase1_time, unsigned long long capacity) {
+                    int best0, double phase1_time, unsigned long long capacity,
+                    std::atomic<int>* shared_best = nullptr) {
   Result r;
   r.phases.push_back({tree0, sol0, phase1_time});
   uint64_t tree = tree0, sol = sol0;
@@ -377,7 +399,7 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       launch_gather_pfsp(ctl_d.p, bc_d.p, boff_d.p, childbuf_d.p, pool_d.p, stride, G,
                          stream.s);
     };
-    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 3, iter, r);
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 3, iter, r, shared_best);
     tree = fin.tree;
     sol = fin.sol;
     best = fin.best;
@@ -474,6 +496,59 @@ std::vector<int32_t> pfsp_gpu_bounds(int inst, const std::string& lb_str,
   HIP_CHECK(hipMemcpy(bounds.data(), bounds_d.p, n * I.jobs * sizeof(int32_t),
                       hipMemcpyDeviceToHost));
   return bounds;
+}
+
+
+// ---------------------------------------------------------------------------
+// Asynchronous PFSP engine: runs the devpool search on a background thread
+// and exposes a shared incumbent, so the Python distributed tier can run a
+// fixed-cadence RCCL all_reduce(min) loop exchanging upper bounds DURING the
+// search (the reference only min-reduces at the end,
+// pfsp_dist_multigpu_cuda.c:694 — exchanging earlier tightens pruning and is
+// always sound since any incumbent >= the optimum is a valid UB).
+// ---------------------------------------------------------------------------
+
+PfspAsyncEngine::PfspAsyncEngine(std::vector<PFSPNode> nodes, int inst,
+                                 const std::string& lb_str, int ub, int best0, int m, int M,
+                                 int device, unsigned long long capacity)
+    : shared_best_(0), done_(false) {
+  const LbKind lb = lb_from_string(lb_str);
+  PfspInstance I = make_pfsp_instance(inst, ub);
+  const int b0 = (best0 > 0) ? best0 : I.init_ub;
+  shared_best_.store(b0);
+  th_ = std::thread([this, nodes = std::move(nodes), I = std::move(I), lb, m, M, device,
+                     capacity, b0]() mutable {
+    try {
+      Pool<PFSPNode> pool;
+      pool.pushBackBulk(nodes.data(), nodes.size());
+      result_ = pfsp_gpu_run(I, lb, pool, m, M, device, "devpool", 0, 0, b0, 0.0, capacity,
+                             &shared_best_);
+    } catch (...) {
+      err_ = std::current_exception();
+    }
+    done_.store(true, std::memory_order_release);
+  });
+}
+
+PfspAsyncEngine::~PfspAsyncEngine() {
+  if (th_.joinable()) th_.join();
+}
+
+int PfspAsyncEngine::best() const { return shared_best_.load(std::memory_order_relaxed); }
+
+void PfspAsyncEngine::update_best(int b) {
+  int cur = shared_best_.load(std::memory_order_relaxed);
+  while (b < cur &&
+         !shared_best_.compare_exchange_weak(cur, b, std::memory_order_relaxed)) {
+  }
+}
+
+bool PfspAsyncEngine::done() const { return done_.load(std::memory_order_acquire); }
+
+Result PfspAsyncEngine::join() {
+  if (th_.joinable()) th_.join();
+  if (err_) std::rethrow_exception(err_);
+  return result_;
 }
 
 }  // namespace gats

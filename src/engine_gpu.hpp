@@ -1,6 +1,8 @@
 // Public API of the single-GPU engines (implemented in engine_gpu.cpp).
 #pragma once
+#include <atomic>
 #include <string>
+#include <thread>
 #include <vector>
 
 #include "nodes.hpp"
@@ -26,6 +28,26 @@ Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst, const st
 Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& eval);
 Result pfsp_multigpu(int inst, const std::string& lb, int ub, int m, int M, int D,
                      const std::string& eval, bool share_best);
+
+// Background-thread PFSP engine with a shared incumbent for mid-search
+// RCCL UB exchange (see engine_gpu.cpp).
+class PfspAsyncEngine {
+ public:
+  PfspAsyncEngine(std::vector<PFSPNode> nodes, int inst, const std::string& lb, int ub,
+                  int best0, int m, int M, int device, unsigned long long capacity);
+  ~PfspAsyncEngine();
+  int best() const;
+  void update_best(int b);
+  bool done() const;
+  Result join();
+
+ private:
+  std::thread th_;
+  std::atomic<int> shared_best_;
+  std::atomic<bool> done_;
+  Result result_;
+  std::exception_ptr err_;
+};
 
 std::vector<uint8_t> nq_gpu_labels(int N, int g, const std::vector<NQNode>& nodes,
                                    int device);
