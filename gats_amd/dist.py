@@ -177,6 +177,9 @@ def run_from_cli(args):
     rank, world = init_dist()
     if args.problem == "nqueens":
         r = run_nqueens(args.N, args.g, args.m, args.M, args.mode, args.capacity)
+    elif args.ub == 0:
+        # open upper bound: exchange the incumbent over RCCL during the search
+        r = run_pfsp_shared_ub(args.inst, args.lb, args.ub, args.m, args.M, args.capacity)
     else:
         r = run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode, args.capacity)
     if world > 1:

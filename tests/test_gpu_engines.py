@@ -52,3 +52,27 @@ def test_gpu_from_pool_entry(gpu):
     seq = gpu.nqueens_seq(13, 1)
     assert tree + r["tree"] == seq["tree"]
     assert sol + r["sol"] == seq["sol"]
+
+
+def test_pfsp_async_engine_matches_sync(gpu):
+    # the dist tier's mid-search UB-exchange engine, driven locally (world=1)
+    nodes, tree1, sol1, best = gpu.pfsp_bfs_frontier(14, "lb1_d", 1, 4096)
+    eng = gpu.PfspAsyncEngine(nodes, 14, "lb1_d", 1, best, 25, 50000, 0, 1 << 24)
+    import time
+    while not eng.done():
+        _ = eng.best()
+        time.sleep(0.005)
+    r = eng.join()
+    seq = gpu.pfsp_seq(14, "lb1_d", 1)
+    assert tree1 + r["tree"] == seq["tree"]
+    assert sol1 + r["sol"] == seq["sol"]
+    assert r["optimum"] == 1377
+
+
+def test_pfsp_async_engine_adopts_lower_ub(gpu):
+    # an externally-published incumbent must tighten pruning mid-run
+    nodes, tree1, sol1, best = gpu.pfsp_bfs_frontier(14, "lb1", 0, 2048)
+    eng = gpu.PfspAsyncEngine(nodes, 14, "lb1", 0, 0, 25, 50000, 0, 1 << 24)
+    eng.update_best(1377)  # the known optimum, as if another rank found it
+    r = eng.join()
+    assert r["optimum"] == 1377
