@@ -143,6 +143,9 @@ int lbk_of(LbKind lb) {
 // the engine's best into it (atomic min) and adopts a lower value published by
 // other ranks/threads by writing ctl->best on the (synchronized) stream —
 // the RCCL incumbent-UB exchange of the distributed tier plugs in here.
+// enqueue_iter(parity): parity alternates 0/1 per iteration — iteration i
+// reads ctl[parity] and gather2 writes ctl[1-parity]. BATCH is even, so the
+// host always reads (and patches the incumbent of) ctl[0].
 template <class EnqueueIter>
 static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long m,
                                int kernels_per_iter, EnqueueIter&& enqueue_iter, Result& r,
@@ -156,7 +159,7 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
   hipGraphExec_t exec = nullptr;
   if (use_graph) {
     HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
-    for (int b = 0; b < BATCH; b++) enqueue_iter();
+    for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
     HIP_CHECK(hipStreamEndCapture(s, &graph));
     HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
   }
@@ -165,7 +168,7 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
     if (use_graph) {
       HIP_CHECK(hipGraphLaunch(exec, s));
     } else {
-      for (int b = 0; b < BATCH; b++) enqueue_iter();
+      for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
     }
     HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d, sizeof(DevCtl), hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
@@ -242,7 +245,7 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     if (static_cast<unsigned long long>(M) * N > (1ull << 31))
       throw std::invalid_argument("devpool requires M * N <= 2^31");
     DevGuard<NQNode> pool_d(capacity);
-    DevGuard<DevCtl> ctl_d(1);
+    DevGuard<DevCtl> ctl_d(2);  // parity-alternating control blocks
     const int G = devpool_grid(M, N, 1);
     // two-level expand (default): each child slot can spawn up to N-1 pushes
     const bool two_level = std::getenv("GATS_NQ_1LEVEL") == nullptr;
@@ -250,7 +253,6 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
         two_level ? devpool_stride(1) * (MAX_JOBS - 1) : devpool_stride(1);
     DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
     DevGuard<uint32_t> bc_d(G), bs_d(G), be_d(G);
-    DevGuard<unsigned long long> boff_d(G);
     const size_t init = pool.size();
     if (init > capacity) throw std::runtime_error("devpool capacity too small");
     HIP_CHECK(hipMemcpy(pool_d.p, pool.data(), init * sizeof(NQNode), hipMemcpyHostToDevice));
@@ -259,21 +261,22 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     ctl.tree = tree;
     ctl.sol = sol;
     HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
     r.h2d += 2;
-    r.h2d_bytes += init * sizeof(NQNode) + sizeof(DevCtl);
+    r.h2d_bytes += init * sizeof(NQNode) + 2 * sizeof(DevCtl);
 
-    auto iter = [&] {
+    auto iter = [&](int parity) {
+      DevCtl* cur = ctl_d.p + parity;
+      DevCtl* next = ctl_d.p + (1 - parity);
       if (two_level)
-        launch_nq_x2(ctl_d.p, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, m, M,
+        launch_nq_x2(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, m, M,
                      stream.s);
       else
-        launch_nq_x(ctl_d.p, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, N, g, m, M, stream.s);
-      launch_scan(ctl_d.p, bc_d.p, bs_d.p, two_level ? be_d.p : nullptr, boff_d.p, G, m, M,
-                  capacity, stream.s);
-      launch_gather_nq(ctl_d.p, bc_d.p, boff_d.p, childbuf_d.p, pool_d.p, stride, G,
-                       stream.s);
+        launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, N, g, m, M, stream.s);
+      launch_gather2_nq(cur, next, bc_d.p, bs_d.p, two_level ? be_d.p : nullptr,
+                        childbuf_d.p, pool_d.p, stride, G, m, M, capacity, stream.s);
     };
-    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 3, iter, r);
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r);
     tree = fin.tree;
     sol = fin.sol;
     r.gpu_iters = fin.iters;
@@ -373,12 +376,11 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
       throw std::invalid_argument("devpool requires M * jobs <= 2^31");
     DevGuard<PFSPNode> pool_d(capacity);
-    DevGuard<DevCtl> ctl_d(1);
+    DevGuard<DevCtl> ctl_d(2);  // parity-alternating control blocks
     const int G = devpool_grid(M, jobs, lbk);
     const int stride = devpool_stride(lbk);
     DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
     DevGuard<uint32_t> bc_d(G), bs_d(G);
-    DevGuard<unsigned long long> boff_d(G);
     const size_t init = pool.size();
     if (init > capacity) throw std::runtime_error("devpool capacity too small");
     HIP_CHECK(
@@ -389,17 +391,19 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     ctl.sol = sol;
     ctl.best = best;
     HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
     r.h2d += 2;
-    r.h2d_bytes += init * sizeof(PFSPNode) + sizeof(DevCtl);
+    r.h2d_bytes += init * sizeof(PFSPNode) + 2 * sizeof(DevCtl);
 
-    auto iter = [&] {
-      launch_pfsp_x(ctl_d.p, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk,
+    auto iter = [&](int parity) {
+      DevCtl* cur = ctl_d.p + parity;
+      DevCtl* next = ctl_d.p + (1 - parity);
+      launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk,
                     tables.tb, m, M, stream.s);
-      launch_scan(ctl_d.p, bc_d.p, bs_d.p, nullptr, boff_d.p, G, m, M, capacity, stream.s);
-      launch_gather_pfsp(ctl_d.p, bc_d.p, boff_d.p, childbuf_d.p, pool_d.p, stride, G,
-                         stream.s);
+      launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, childbuf_d.p, pool_d.p, stride, G, m,
+                          M, capacity, stream.s);
     };
-    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 3, iter, r, shared_best);
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best);
     tree = fin.tree;
     sol = fin.sol;
     best = fin.best;

@@ -55,15 +55,12 @@ void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32
 void launch_nq_x2(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                   uint32_t* blockCounts, uint32_t* blockSols, uint32_t* blockExtra, int N,
                   int g, unsigned long long m, unsigned long long M, hipStream_t s);
-void launch_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
-                 const uint32_t* blockExtra, unsigned long long* blockOffsets, int G,
-                 unsigned long long m, unsigned long long M, unsigned long long capacity,
-                 hipStream_t s);
-void launch_gather_nq(const DevCtl* ctl, const uint32_t* bc,
-                      const unsigned long long* boff, const NQNode* childbuf, NQNode* pool,
-                      int strideNodes, int G, hipStream_t s);
-void launch_gather_pfsp(const DevCtl* ctl, const uint32_t* bc,
-                        const unsigned long long* boff, const PFSPNode* childbuf,
-                        PFSPNode* pool, int strideNodes, int G, hipStream_t s);
-
+void launch_gather2_nq(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
+                       const uint32_t* bs, const uint32_t* be, const NQNode* childbuf,
+                       NQNode* pool, int strideNodes, int G, unsigned long long m,
+                       unsigned long long M, unsigned long long capacity, hipStream_t s);
+void launch_gather2_pfsp(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
+                         const uint32_t* bs, const PFSPNode* childbuf, PFSPNode* pool,
+                         int strideNodes, int G, unsigned long long m, unsigned long long M,
+                         unsigned long long capacity, hipStream_t s);
 }  // namespace gats

@@ -732,73 +732,63 @@ __global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
   }
 }
 
-// K2: single-block scan over the G per-block counts; sole DevCtl writer.
-// Each thread owns SCAN_PER consecutive counts so one block-scan round covers
-// G <= BLOCK*SCAN_PER blocks (the scan is on the critical path every
-// iteration; multi-round versions were ~2x slower at G ~ 1000).
-constexpr int SCAN_PER = 8;
-__global__ void k_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
-                       const uint32_t* blockExtra, unsigned long long* blockOffsets, int G,
-                       unsigned long long m, unsigned long long M,
-                       unsigned long long capacity) {
-  const unsigned long long c = derive_chunk(ctl, m, M);
-  const unsigned long long base = ctl->size - c;
-  unsigned long long running = 0;
-  uint32_t my_sols = 0, my_extra = 0;
-  for (int g0 = 0; g0 < G; g0 += BLOCK * SCAN_PER) {
-    uint32_t v[SCAN_PER];
-    uint32_t mine = 0;
-#pragma unroll
-    for (int j = 0; j < SCAN_PER; j++) {
-      const int i = g0 + threadIdx.x * SCAN_PER + j;
-      v[j] = (i < G) ? blockCounts[i] : 0;
-      mine += v[j];
-      if (i < G) {
-        my_sols += blockSols[i];
-        if (blockExtra) my_extra += blockExtra[i];
-      }
+
+// K2+K3 merged ("gather2"): every block derives its own pool offset by
+// summing the counts of the blocks before it (G ~ 1000 u32 loads through L2,
+// done in parallel across blocks — cheaper than serializing on the
+// single-block scan kernel), copies its children, and the LAST block writes
+// the next iteration's control block. Control blocks alternate per iteration
+// (parity) so readers of iteration i never race the writer: expand(i) and
+// gather2(i) read ctl_cur, gather2's last block writes ctl_next, and the
+// kernel boundary publishes it for iteration i+1 (placement-independent).
+template <class NodeT>
+__global__ void k_gather2(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* blockCounts,
+                          const uint32_t* blockSols, const uint32_t* blockExtra,
+                          const NodeT* childbuf, NodeT* pool, int strideNodes, int G,
+                          unsigned long long m, unsigned long long M,
+                          unsigned long long capacity) {
+  const unsigned long long c = derive_chunk(ctl_cur, m, M);
+  const unsigned long long base = ctl_cur->size - c;
+  const int b = blockIdx.x;
+  const bool last = (b == G - 1);
+
+  // prefix over blocks [0, b) — plus full sol/extra sums for the last block
+  uint32_t my_pre = 0, my_sols = 0, my_extra = 0;
+  for (int i = threadIdx.x; i < G; i += BLOCK) {
+    if (i < b) my_pre += blockCounts[i];
+    if (last) {
+      my_sols += blockSols[i];
+      if (blockExtra) my_extra += blockExtra[i];
     }
-    uint32_t tot;
-    uint32_t pre = block_excl_scan(mine, tot);
-#pragma unroll
-    for (int j = 0; j < SCAN_PER; j++) {
-      const int i = g0 + threadIdx.x * SCAN_PER + j;
-      if (i < G) blockOffsets[i] = base + running + pre;
-      pre += v[j];
-    }
-    running += tot;
   }
-  uint32_t sol_tot, extra_tot;
+  uint32_t pre_tot, sol_tot, extra_tot;
+  block_excl_scan(my_pre, pre_tot);
   block_excl_scan(my_sols, sol_tot);
   block_excl_scan(my_extra, extra_tot);
-  if (threadIdx.x == 0) {
-    if (ctl->overflow) return;
-    if (base + running > capacity) {
-      ctl->overflow = 1;
-      return;
-    }
-    ctl->chunk = c;
-    ctl->iters += (c > 0);
-    ctl->size = base + running;
-    ctl->tree += running + extra_tot;
-    ctl->sol += sol_tot;
-  }
-}
 
-// K3: block-strided copy of each block's compacted children into the pool.
-template <class NodeT>
-__global__ void k_gather(const DevCtl* ctl, const uint32_t* blockCounts,
-                         const unsigned long long* blockOffsets, const NodeT* childbuf,
-                         NodeT* pool, int strideNodes) {
-  if (ctl->overflow) return;
-  const uint32_t cnt = blockCounts[blockIdx.x];
-  if (cnt == 0) return;
-  const unsigned long long* src = reinterpret_cast<const unsigned long long*>(
-      childbuf + static_cast<unsigned long long>(blockIdx.x) * strideNodes);
-  unsigned long long* dst =
-      reinterpret_cast<unsigned long long*>(pool + blockOffsets[blockIdx.x]);
-  const int words = static_cast<int>(cnt) * static_cast<int>(sizeof(NodeT) / 8);
-  for (int i = threadIdx.x; i < words; i += blockDim.x) dst[i] = src[i];
+  const uint32_t cnt = blockCounts[b];
+  const unsigned long long off = base + pre_tot;
+  const bool over = ctl_cur->overflow || (off + cnt > capacity);
+  if (!over && cnt > 0) {
+    const unsigned long long* src = reinterpret_cast<const unsigned long long*>(
+        childbuf + static_cast<unsigned long long>(b) * strideNodes);
+    unsigned long long* dst = reinterpret_cast<unsigned long long*>(pool + off);
+    const int words = static_cast<int>(cnt) * static_cast<int>(sizeof(NodeT) / 8);
+    for (int i = threadIdx.x; i < words; i += BLOCK) dst[i] = src[i];
+  }
+  if (over && threadIdx.x == 0) ctl_next->overflow = 1;  // sticky; host aborts
+  if (last && threadIdx.x == 0) {
+    const unsigned long long total = pre_tot + cnt;
+    if (!ctl_cur->overflow && off + cnt <= capacity) {
+      ctl_next->size = base + total;
+      ctl_next->tree = ctl_cur->tree + total + extra_tot;
+      ctl_next->sol = ctl_cur->sol + sol_tot;
+      ctl_next->iters = ctl_cur->iters + (c > 0);
+      ctl_next->chunk = c;
+      ctl_next->overflow = ctl_cur->overflow;
+    }
+    ctl_next->best = ctl_cur->best;  // carries expand's atomicMin updates
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -895,26 +885,21 @@ void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32
     launch_pfsp_x_mm<20>(ctl, pool, childbuf, bc, bs, jobs, lbk, tb, m, M, s);
 }
 
-void launch_scan(DevCtl* ctl, const uint32_t* blockCounts, const uint32_t* blockSols,
-                 const uint32_t* blockExtra, unsigned long long* blockOffsets, int G,
-                 unsigned long long m, unsigned long long M, unsigned long long capacity,
-                 hipStream_t s) {
-  hipLaunchKernelGGL(k_scan, dim3(1), dim3(BLOCK), 0, s, ctl, blockCounts, blockSols,
-                     blockExtra, blockOffsets, G, m, M, capacity);
+void launch_gather2_nq(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
+                       const uint32_t* bs, const uint32_t* be, const NQNode* childbuf,
+                       NQNode* pool, int strideNodes, int G, unsigned long long m,
+                       unsigned long long M, unsigned long long capacity, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather2<NQNode>, dim3(G), dim3(BLOCK), 0, s, ctl_cur, ctl_next, bc,
+                     bs, be, childbuf, pool, strideNodes, G, m, M, capacity);
 }
 
-void launch_gather_nq(const DevCtl* ctl, const uint32_t* bc,
-                      const unsigned long long* boff, const NQNode* childbuf, NQNode* pool,
-                      int strideNodes, int G, hipStream_t s) {
-  hipLaunchKernelGGL(k_gather<NQNode>, dim3(G), dim3(BLOCK), 0, s, ctl, bc, boff, childbuf,
-                     pool, strideNodes);
-}
-
-void launch_gather_pfsp(const DevCtl* ctl, const uint32_t* bc,
-                        const unsigned long long* boff, const PFSPNode* childbuf,
-                        PFSPNode* pool, int strideNodes, int G, hipStream_t s) {
-  hipLaunchKernelGGL(k_gather<PFSPNode>, dim3(G), dim3(BLOCK), 0, s, ctl, bc, boff, childbuf,
-                     pool, strideNodes);
+void launch_gather2_pfsp(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
+                         const uint32_t* bs, const PFSPNode* childbuf, PFSPNode* pool,
+                         int strideNodes, int G, unsigned long long m, unsigned long long M,
+                         unsigned long long capacity, hipStream_t s) {
+  hipLaunchKernelGGL(k_gather2<PFSPNode>, dim3(G), dim3(BLOCK), 0, s, ctl_cur, ctl_next, bc,
+                     bs, static_cast<const uint32_t*>(nullptr), childbuf, pool, strideNodes,
+                     G, m, M, capacity);
 }
 
 }  // namespace gats
