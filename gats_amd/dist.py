@@ -107,7 +107,7 @@ def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 
     if frontier_target is None:
         # PFSP 20-job trees are small (ta014 lb1 ~2.6M nodes); a deep frontier
         # would move a large share of the search onto the single-threaded CPU
-        frontier_target = max(8192, 2048 * world)
+        frontier_target = max(2048, 2048 * world)
     nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
     my = slice_frontier(nodes, rank, world)
     phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
@@ -140,7 +140,7 @@ def run_pfsp_shared_ub(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
     c = gats_amd.core()
     rank, world = init_dist()
     if frontier_target is None:
-        frontier_target = max(8192, 2048 * world)
+        frontier_target = max(2048, 2048 * world)
     nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
     my = slice_frontier(nodes, rank, world)
     local = rank % max(1, c.gpu_device_count())

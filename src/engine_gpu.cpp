@@ -154,22 +154,29 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
   const int BATCH = 16;
   // GATS_NO_GRAPH=1 falls back to eager launches (rocprofv3 crashes tracing
   // hipGraph replays on ROCm 7.2; eager mode gives identical results).
-  const bool use_graph = std::getenv("GATS_NO_GRAPH") == nullptr;
+  // Graph capture is LAZY: instantiation costs a few ms, which dominates
+  // small searches (PFSP ta0xx finish in <10 ms), so the first EAGER_BATCHES
+  // batches run eager and the graph is built only if the search is still
+  // going.
+  const bool graph_allowed = std::getenv("GATS_NO_GRAPH") == nullptr;
+  const int EAGER_BATCHES = 4;
+  int batches = 0;
   hipGraph_t graph = nullptr;
   hipGraphExec_t exec = nullptr;
-  if (use_graph) {
-    HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
-    for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
-    HIP_CHECK(hipStreamEndCapture(s, &graph));
-    HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
-  }
   bool overflow = false;
   while (true) {
-    if (use_graph) {
+    if (graph_allowed && batches >= EAGER_BATCHES && exec == nullptr) {
+      HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
+      for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
+      HIP_CHECK(hipStreamEndCapture(s, &graph));
+      HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
+    }
+    if (exec != nullptr) {
       HIP_CHECK(hipGraphLaunch(exec, s));
     } else {
       for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
     }
+    batches++;
     HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d, sizeof(DevCtl), hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     r.kernel_launch += static_cast<uint64_t>(kernels_per_iter) * BATCH;
@@ -195,7 +202,7 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
     }
     if (ctl_h.p->size < m) break;
   }
-  if (use_graph) {
+  if (exec != nullptr) {
     (void)hipGraphExecDestroy(exec);
     (void)hipGraphDestroy(graph);
   }
@@ -247,8 +254,10 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     DevGuard<NQNode> pool_d(capacity);
     DevGuard<DevCtl> ctl_d(2);  // parity-alternating control blocks
     const int G = devpool_grid(M, N, 1);
-    // two-level expand (default): each child slot can spawn up to N-1 pushes
-    const bool two_level = std::getenv("GATS_NQ_1LEVEL") == nullptr;
+    // one-level expand is the default: under the 2-kernel pipeline it measured
+    // 3050 vs 2941 Mnodes/s for two-level at N=17 (the grandchild loops
+    // serialize work that one-level spreads across threads)
+    const bool two_level = std::getenv("GATS_NQ_2LEVEL") != nullptr;
     const int stride =
         two_level ? devpool_stride(1) * (MAX_JOBS - 1) : devpool_stride(1);
     DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
