@@ -80,10 +80,24 @@ struct PfspTablesGuard {
       p1[i] = static_cast<uint8_t>(I.lb2.pairs1[i]);
       p2[i] = static_cast<uint8_t>(I.lb2.pairs2[i]);
     }
+    std::vector<uint64_t> jp(static_cast<size_t>(pairs) * n);
+    for (int k = 0; k < pairs; k++) {
+      const int ma0 = I.lb2.pairs1[k], ma1 = I.lb2.pairs2[k];
+      for (int j = 0; j < n; j++) {
+        const int job = I.lb2.johnson_schedules[static_cast<size_t>(k) * n + j];
+        const uint64_t ptm0 = static_cast<uint64_t>(I.lb1.p_times[ma0 * n + job]);
+        const uint64_t ptm1 = static_cast<uint64_t>(I.lb1.p_times[ma1 * n + job]);
+        const uint64_t lag =
+            static_cast<uint64_t>(I.lb2.lags[static_cast<size_t>(k) * n + job]);
+        jp[static_cast<size_t>(k) * n + j] =
+            (static_cast<uint64_t>(job) << 48) | (lag << 32) | (ptm1 << 16) | ptm0;
+      }
+    }
     tb.p_times = keep(dev_upload(p16.data(), p16.size()));
     tb.min_tails = keep(dev_upload(mt.data(), mt.size()));
     tb.lags = keep(dev_upload(lags16.data(), lags16.size()));
     tb.johnson_schedules = keep(dev_upload(js8.data(), js8.size()));
+    tb.johnson_packed = keep(dev_upload(jp.data(), jp.size()));
     tb.pairs1 = keep(dev_upload(p1.data(), p1.size()));
     tb.pairs2 = keep(dev_upload(p2.data(), p2.size()));
   }

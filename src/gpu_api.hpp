@@ -31,6 +31,9 @@ struct PfspDevTables {
   const int32_t* min_tails;          // [machines]
   const int16_t* lags;               // [pairs * jobs]
   const uint8_t* johnson_schedules;  // [pairs * jobs]
+  // packed per (pair, johnson position): job<<48 | lag<<32 | ptm1<<16 | ptm0
+  // (one ds_read_b64 per inner step instead of 4 scalar LDS reads)
+  const uint64_t* johnson_packed;    // [pairs * jobs]
   const uint8_t* pairs1;             // [pairs]
   const uint8_t* pairs2;             // [pairs]
 };

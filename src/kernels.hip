@@ -138,8 +138,7 @@ struct LdsLb2 {
   static constexpr int PAIRS = MM * (MM - 1) / 2;
   int16_t p[MM * MAX_JOBS];
   int32_t min_tails[MM];
-  int16_t lags[PAIRS * MAX_JOBS];
-  uint8_t js[PAIRS * MAX_JOBS];  // per-pair Johnson schedules
+  uint64_t jp[PAIRS * MAX_JOBS];  // packed johnson: job<<48|lag<<32|ptm1<<16|ptm0
   uint8_t pair1[PAIRS], pair2[PAIRS];
   // per-thread `front` scratch, runtime-indexed by machine-pair ids; padded
   // stride MM+1 keeps the 32-bank groups conflict-free (stride odd vs 32)
@@ -156,10 +155,8 @@ template <int MM>
 __device__ inline void stage_lb2_tables(LdsLb2<MM>& lds, const PfspDevTables& tb, int jobs) {
   stage_lb1_tables<MM>(lds, tb, jobs);
   constexpr int PAIRS = LdsLb2<MM>::PAIRS;
-  for (int i = threadIdx.x; i < PAIRS * jobs; i += blockDim.x) {
-    lds.lags[i] = tb.lags[i];
-    lds.js[i] = tb.johnson_schedules[i];
-  }
+  for (int i = threadIdx.x; i < PAIRS * jobs; i += blockDim.x)
+    lds.jp[i] = tb.johnson_packed[i];
   if (threadIdx.x < PAIRS) {
     lds.pair1[threadIdx.x] = tb.pairs1[threadIdx.x];
     lds.pair2[threadIdx.x] = tb.pairs2[threadIdx.x];
@@ -278,14 +275,14 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
     const int ma1 = lds.pair2[l];
     int tmp0 = front[ma0];
     int tmp1 = front[ma1];
-    const uint8_t* js = &lds.js[l * jobs];
-    const int16_t* lag = &lds.lags[l * jobs];
+    const uint64_t* jp = &lds.jp[l * jobs];
     for (int j = 0; j < jobs; j++) {
-      const int job = js[j];
+      const uint64_t v = jp[j];  // one ds_read_b64 replaces 4 scalar LDS reads
+      const int job = static_cast<int>(v >> 48);
       if (!(scheduled >> job & 1u)) {
-        tmp0 += lds.p[ma0 * jobs + job];
-        tmp1 = max(tmp1, tmp0 + lag[job]);
-        tmp1 += lds.p[ma1 * jobs + job];
+        tmp0 += static_cast<int>(v & 0xffff);
+        tmp1 = max(tmp1, tmp0 + static_cast<int>((v >> 32) & 0xffff));
+        tmp1 += static_cast<int>((v >> 16) & 0xffff);
       }
     }
     lb = max(lb, max(tmp1 + lds.min_tails[ma1], tmp0 + lds.min_tails[ma0]));
