@@ -38,6 +38,14 @@ struct PfspDevTables {
   const uint8_t* pairs2;             // [pairs]
 };
 
+// Same tables with machine pairs reordered strongest-first (devpool expand
+// only): the push/prune decision `lb < best` and the leaf best-update (only
+// taken on full, non-early-exited evaluations == the true max) are both
+// invariant to pair order, so the early exit fires pairs earlier. The
+// hostpool eval kernels keep the reference's identity order so their bound
+// VALUES stay bit-identical to the CPU oracle (tests/test_gpu_kernels.py).
+
+
 // Launchers implemented in kernels.hip.
 // hostpool eval (labels/bounds out):
 void launch_nq_eval(const NQNode* parents, int n, int N, int g, uint8_t* labels,
