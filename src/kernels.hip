@@ -270,22 +270,35 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
 
   constexpr int PAIRS = LdsLb2<MM>::PAIRS;
   int lb = 0;
-  for (int l = 0; l < PAIRS; l++) {
-    const int ma0 = lds.pair1[l];
-    const int ma1 = lds.pair2[l];
-    int tmp0 = front[ma0];
-    int tmp1 = front[ma1];
-    const uint64_t* jp = &lds.jp[l * jobs];
+  // two pairs in flight per step: each pair's (tmp0, tmp1) update chain is a
+  // serial LDS-load -> add -> max dependency; interleaving two independent
+  // chains doubles the ILP available to hide the ds_read_b64 latency
+  for (int l = 0; l < PAIRS; l += 2) {
+    const bool two = (l + 1) < PAIRS;
+    const int ma0a = lds.pair1[l], ma1a = lds.pair2[l];
+    const int ma0b = lds.pair1[two ? l + 1 : l], ma1b = lds.pair2[two ? l + 1 : l];
+    int t0a = front[ma0a], t1a = front[ma1a];
+    int t0b = front[ma0b], t1b = front[ma1b];
+    const uint64_t* jpa = &lds.jp[l * jobs];
+    const uint64_t* jpb = &lds.jp[(two ? l + 1 : l) * jobs];
     for (int j = 0; j < jobs; j++) {
-      const uint64_t v = jp[j];  // one ds_read_b64 replaces 4 scalar LDS reads
-      const int job = static_cast<int>(v >> 48);
-      if (!(scheduled >> job & 1u)) {
-        tmp0 += static_cast<int>(v & 0xffff);
-        tmp1 = max(tmp1, tmp0 + static_cast<int>((v >> 32) & 0xffff));
-        tmp1 += static_cast<int>((v >> 16) & 0xffff);
+      const uint64_t va = jpa[j];  // one ds_read_b64 replaces 4 scalar LDS reads
+      const uint64_t vb = jpb[j];
+      const int ja = static_cast<int>(va >> 48);
+      const int jb = static_cast<int>(vb >> 48);
+      if (!(scheduled >> ja & 1u)) {
+        t0a += static_cast<int>(va & 0xffff);
+        t1a = max(t1a, t0a + static_cast<int>((va >> 32) & 0xffff));
+        t1a += static_cast<int>((va >> 16) & 0xffff);
+      }
+      if (!(scheduled >> jb & 1u)) {
+        t0b += static_cast<int>(vb & 0xffff);
+        t1b = max(t1b, t0b + static_cast<int>((vb >> 32) & 0xffff));
+        t1b += static_cast<int>((vb >> 16) & 0xffff);
       }
     }
-    lb = max(lb, max(tmp1 + lds.min_tails[ma1], tmp0 + lds.min_tails[ma0]));
+    lb = max(lb, max(t1a + lds.min_tails[ma1a], t0a + lds.min_tails[ma0a]));
+    if (two) lb = max(lb, max(t1b + lds.min_tails[ma1b], t0b + lds.min_tails[ma0b]));
     if (lb > best) break;
   }
   return lb;
