@@ -297,9 +297,16 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
         t1b += static_cast<int>((vb >> 16) & 0xffff);
       }
     }
+    // merge pair a, check, then pair b: keeps the returned value bit-equal to
+    // the reference's per-pair early exit (c_bound_johnson.c:231-233), which
+    // the hostpool-vs-CPU-oracle tests assert; pair b's work is wasted only
+    // on the exit iteration
     lb = max(lb, max(t1a + lds.min_tails[ma1a], t0a + lds.min_tails[ma0a]));
-    if (two) lb = max(lb, max(t1b + lds.min_tails[ma1b], t0b + lds.min_tails[ma0b]));
     if (lb > best) break;
+    if (two) {
+      lb = max(lb, max(t1b + lds.min_tails[ma1b], t0b + lds.min_tails[ma0b]));
+      if (lb > best) break;
+    }
   }
   return lb;
 }
