@@ -80,7 +80,7 @@ def main():
 
     # MAX over ranks (contract): reduce elapsed
     if world > 1:
-        dev = torch.device(f"cuda:{local}")
+        dev = gdist._backend_device(f"cuda:{local}")
         e = torch.tensor([elapsed], dtype=torch.float32, device=dev)
         torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(e[0].item())

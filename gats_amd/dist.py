@@ -35,7 +35,8 @@ def init_dist():
         return td.get_rank(), td.get_world_size()
     if "RANK" not in os.environ:
         return 0, 1
-    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    backend = os.environ.get("GATS_DIST_BACKEND") or (
+        "nccl" if torch.cuda.is_available() else "gloo")
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     td.init_process_group(backend=backend)
     return td.get_rank(), td.get_world_size()
@@ -51,9 +52,15 @@ def slice_frontier(nodes: bytes, rank: int, world: int) -> bytes:
     return bytes(out)
 
 
+def _backend_device(device):
+    if td.is_initialized() and td.get_backend() == "gloo":
+        return torch.device("cpu")
+    return torch.device(device) if torch.cuda.is_available() else torch.device("cpu")
+
+
 def _reduce_stats(r, phase1, device, world, best=None):
     """Global reductions; returns combined stats dict on every rank."""
-    dev = torch.device(device) if torch.cuda.is_available() else torch.device("cpu")
+    dev = _backend_device(device)
     t = torch.tensor([r["tree"], r["sol"]], dtype=torch.int64, device=dev)
     td.all_reduce(t, op=td.ReduceOp.SUM)
     elapsed = torch.tensor([r["time"]], dtype=torch.float64 if dev.type == "cpu" else torch.float32,
@@ -146,7 +153,7 @@ def run_pfsp_shared_ub(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
     local = rank % max(1, c.gpu_device_count())
     t0 = time.perf_counter()
     eng = c.PfspAsyncEngine(my, inst, lb, ub, best, m, M, local, capacity)
-    dev = torch.device(f"cuda:{local}") if torch.cuda.is_available() else torch.device("cpu")
+    dev = _backend_device(f"cuda:{local}")
     if world > 1:
         while True:
             t = torch.tensor([eng.best(), 0 if eng.done() else 1], dtype=torch.int64,

@@ -279,7 +279,10 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
   StreamGuard stream;
   const double t2 = now_sec();
 
-  if (mode == "hostpool") {
+  std::string mode_eff = mode;
+  if (mode == "devpool" && N < 4) mode_eff = "hostpool";  // expand tiles assume N >= 4
+
+  if (mode_eff == "hostpool") {
     PinnedGuard<NQNode> parents(M);
     PinnedGuard<uint8_t> labels(static_cast<size_t>(M) * N);
     DevGuard<NQNode> parents_d(M);
@@ -300,7 +303,7 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
       r.gpu_iters++;
       nq_generate_children(parents.p, n, N, labels.p, tree, sol, pool);
     }
-  } else if (mode == "devpool") {
+  } else if (mode_eff == "devpool") {
     if (static_cast<unsigned long long>(M) * N > (1ull << 31))
       throw std::invalid_argument("devpool requires M * N <= 2^31");
     DevGuard<NQNode> pool_d(capacity);

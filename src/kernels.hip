@@ -450,7 +450,7 @@ __device__ inline unsigned long long derive_chunk(const DevCtl* ctl, unsigned lo
 __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                        uint32_t* blockCounts, uint32_t* blockSols, int N, int g,
                        unsigned long long m, unsigned long long M) {
-  __shared__ NQNode s[EMIT_TILE + 2];
+  __shared__ NQNode s[EMIT_TILE / 4 + 2];  // N >= 4 (engine falls back below)
   const unsigned long long c = derive_chunk(ctl, m, M);
   // child indices fit u32: the engine enforces M * branching <= 2^31
   const uint32_t total = static_cast<uint32_t>(c * N);
@@ -516,7 +516,7 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
 __global__ void k_nq_x2(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                         uint32_t* blockCounts, uint32_t* blockSols, uint32_t* blockExtra,
                         int N, int g, unsigned long long m, unsigned long long M) {
-  __shared__ NQNode s[EMIT_TILE + 2];
+  __shared__ NQNode s[EMIT_TILE / 4 + 2];  // N >= 4 (engine falls back below)
   const unsigned long long c = derive_chunk(ctl, m, M);
   const uint32_t total = static_cast<uint32_t>(c * N);
   const NQNode* parents = pool + (ctl->size - c);
