@@ -17,6 +17,7 @@
 #include <algorithm>
 #include <atomic>
 #include <cstdlib>
+#include <thread>
 #include <cstring>
 #include <stdexcept>
 #include <string>
@@ -262,6 +263,137 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
   return *ctl_h.p;
 }
 
+
+// ---------------------------------------------------------------------------
+// Multi-slice devpool: S independent device pools driven on S HIP streams by
+// S host threads fill the chip — ONE devpool loop is a serial expand->gather
+// dependency chain whose kernels leave the MI355X roughly half idle at
+// M = 50000 (measured: two ranks time-sharing one GPU nearly doubled
+// throughput). The frontier is round-robin split exactly like the multi-GPU
+// partition (nqueens_multigpu_chpl.chpl:221-226), just inside one device.
+// No stealing between slices; S = 4 keeps the device busy while stragglers
+// finish. Counts are slice-order independent.
+// ---------------------------------------------------------------------------
+
+static int devpool_slices() {
+  if (const char* e = std::getenv("GATS_SLICES")) {
+    int v = atoi(e);
+    return v < 1 ? 1 : v;
+  }
+  return 4;
+}
+
+struct SliceOut {
+  DevCtl fin{};
+  Result diag;
+};
+
+static SliceOut devpool_worker_nq(const std::vector<NQNode>& nodes, int N, int g, int m,
+                                  int M, int device, bool two_level,
+                                  unsigned long long capacity,
+                                  std::vector<NQNode>& leftover) {
+  HIP_CHECK(hipSetDevice(device));
+  StreamGuard stream;
+  SliceOut out;
+  Result& r = out.diag;
+  DevGuard<NQNode> pool_d(capacity);
+  DevGuard<DevCtl> ctl_d(2);  // parity-alternating control blocks
+  const int G = devpool_grid(M, N, 1);
+  const int stride = two_level ? devpool_stride(1) * (MAX_JOBS - 1) : devpool_stride(1);
+  DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
+  DevGuard<uint32_t> bc_d(G), bs_d(G), be_d(G);
+  const size_t init = nodes.size();
+  if (init > capacity) throw std::runtime_error("devpool capacity too small");
+  HIP_CHECK(hipMemcpy(pool_d.p, nodes.data(), init * sizeof(NQNode), hipMemcpyHostToDevice));
+  DevCtl ctl{};
+  ctl.size = init;
+  HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+  r.h2d += 2;
+  r.h2d_bytes += init * sizeof(NQNode) + 2 * sizeof(DevCtl);
+
+  auto iter = [&](int parity) {
+    DevCtl* cur = ctl_d.p + parity;
+    DevCtl* next = ctl_d.p + (1 - parity);
+    if (two_level)
+      launch_nq_x2(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, m, M,
+                   stream.s);
+    else
+      launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, N, g, m, M, stream.s);
+    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, two_level ? be_d.p : nullptr, childbuf_d.p,
+                      pool_d.p, stride, G, m, M, capacity, stream.s);
+  };
+  out.fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r);
+  r.gpu_iters = out.fin.iters;
+  const size_t left = out.fin.size;
+  if (left > 0) {
+    leftover.resize(left);
+    HIP_CHECK(
+        hipMemcpy(leftover.data(), pool_d.p, left * sizeof(NQNode), hipMemcpyDeviceToHost));
+    r.d2h++;
+    r.d2h_bytes += left * sizeof(NQNode);
+  }
+  return out;
+}
+
+static SliceOut devpool_worker_pfsp(const std::vector<PFSPNode>& nodes,
+                                    const PfspInstance& I, const PfspDevTables& tb_sorted,
+                                    int lbk, int best0, int m, int M, int device,
+                                    unsigned long long capacity, std::atomic<int>* shared_best,
+                                    std::vector<PFSPNode>& leftover) {
+  HIP_CHECK(hipSetDevice(device));
+  StreamGuard stream;
+  SliceOut out;
+  Result& r = out.diag;
+  const int jobs = I.jobs, machines = I.machines;
+  DevGuard<PFSPNode> pool_d(capacity);
+  DevGuard<DevCtl> ctl_d(2);
+  const int G = devpool_grid(M, jobs, lbk);
+  const int stride = devpool_stride(lbk);
+  DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
+  DevGuard<uint32_t> bc_d(G), bs_d(G);
+  const size_t init = nodes.size();
+  if (init > capacity) throw std::runtime_error("devpool capacity too small");
+  HIP_CHECK(
+      hipMemcpy(pool_d.p, nodes.data(), init * sizeof(PFSPNode), hipMemcpyHostToDevice));
+  DevCtl ctl{};
+  ctl.size = init;
+  ctl.best = best0;
+  HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+  r.h2d += 2;
+  r.h2d_bytes += init * sizeof(PFSPNode) + 2 * sizeof(DevCtl);
+
+  auto iter = [&](int parity) {
+    DevCtl* cur = ctl_d.p + parity;
+    DevCtl* next = ctl_d.p + (1 - parity);
+    launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk,
+                  tb_sorted, m, M, stream.s);
+    launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, childbuf_d.p, pool_d.p, stride, G, m, M,
+                        capacity, stream.s);
+  };
+  out.fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best);
+  r.gpu_iters = out.fin.iters;
+  const size_t left = out.fin.size;
+  if (left > 0) {
+    leftover.resize(left);
+    HIP_CHECK(
+        hipMemcpy(leftover.data(), pool_d.p, left * sizeof(PFSPNode), hipMemcpyDeviceToHost));
+    r.d2h++;
+    r.d2h_bytes += left * sizeof(PFSPNode);
+  }
+  return out;
+}
+
+static void merge_slice_diag(Result& r, const Result& d) {
+  r.kernel_launch += d.kernel_launch;
+  r.h2d += d.h2d;
+  r.d2h += d.d2h;
+  r.h2d_bytes += d.h2d_bytes;
+  r.d2h_bytes += d.d2h_bytes;
+  r.gpu_iters += d.gpu_iters;
+}
+
 // ---------------------------------------------------------------------------
 // N-Queens
 // ---------------------------------------------------------------------------
@@ -306,52 +438,39 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
   } else if (mode_eff == "devpool") {
     if (static_cast<unsigned long long>(M) * N > (1ull << 31))
       throw std::invalid_argument("devpool requires M * N <= 2^31");
-    DevGuard<NQNode> pool_d(capacity);
-    DevGuard<DevCtl> ctl_d(2);  // parity-alternating control blocks
-    const int G = devpool_grid(M, N, 1);
-    // one-level expand is the default: under the 2-kernel pipeline it measured
-    // 3050 vs 2941 Mnodes/s for two-level at N=17 (the grandchild loops
-    // serialize work that one-level spreads across threads)
     const bool two_level = std::getenv("GATS_NQ_2LEVEL") != nullptr;
-    const int stride =
-        two_level ? devpool_stride(1) * (MAX_JOBS - 1) : devpool_stride(1);
-    DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
-    DevGuard<uint32_t> bc_d(G), bs_d(G), be_d(G);
-    const size_t init = pool.size();
-    if (init > capacity) throw std::runtime_error("devpool capacity too small");
-    HIP_CHECK(hipMemcpy(pool_d.p, pool.data(), init * sizeof(NQNode), hipMemcpyHostToDevice));
-    DevCtl ctl{};
-    ctl.size = init;
-    ctl.tree = tree;
-    ctl.sol = sol;
-    HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-    HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-    r.h2d += 2;
-    r.h2d_bytes += init * sizeof(NQNode) + 2 * sizeof(DevCtl);
-
-    auto iter = [&](int parity) {
-      DevCtl* cur = ctl_d.p + parity;
-      DevCtl* next = ctl_d.p + (1 - parity);
-      if (two_level)
-        launch_nq_x2(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, m, M,
-                     stream.s);
-      else
-        launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, N, g, m, M, stream.s);
-      launch_gather2_nq(cur, next, bc_d.p, bs_d.p, two_level ? be_d.p : nullptr,
-                        childbuf_d.p, pool_d.p, stride, G, m, M, capacity, stream.s);
-    };
-    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r);
-    tree = fin.tree;
-    sol = fin.sol;
-    r.gpu_iters = fin.iters;
-    pool.clear();
-    const size_t left = fin.size;
-    if (left > 0) {
-      std::vector<NQNode> tmp(left);
-      HIP_CHECK(hipMemcpy(tmp.data(), pool_d.p, left * sizeof(NQNode), hipMemcpyDeviceToHost));
-      pool.pushBackBulk(tmp.data(), left);
-      r.d2h++;
-      r.d2h_bytes += left * sizeof(NQNode);
+    int S = devpool_slices();
+    if (pool.size() < static_cast<size_t>(S) * 64) S = 1;
+    std::vector<std::vector<NQNode>> slices(S);
+    {
+      const NQNode* src = pool.data();
+      const size_t total = pool.size();
+      for (int t = 0; t < S; t++) slices[t].reserve(total / S + 1);
+      for (size_t i = 0; i < total; i++) slices[i % S].push_back(src[i]);
+      pool.clear();
+    }
+    std::vector<SliceOut> outs(S);
+    std::vector<std::vector<NQNode>> lefts(S);
+    std::vector<std::exception_ptr> errs(S);
+    std::vector<std::thread> threads;
+    for (int t = 0; t < S; t++) {
+      threads.emplace_back([&, t] {
+        try {
+          outs[t] = devpool_worker_nq(slices[t], N, g, m, M, device, two_level, capacity,
+                                      lefts[t]);
+        } catch (...) {
+          errs[t] = std::current_exception();
+        }
+      });
+    }
+    for (auto& th : threads) th.join();
+    for (auto& e : errs)
+      if (e) std::rethrow_exception(e);
+    for (int t = 0; t < S; t++) {
+      tree += outs[t].fin.tree;
+      sol += outs[t].fin.sol;
+      merge_slice_diag(r, outs[t].diag);
+      if (!lefts[t].empty()) pool.pushBackBulk(lefts[t].data(), lefts[t].size());
     }
   } else {
     throw std::invalid_argument("mode must be hostpool or devpool");
@@ -380,7 +499,10 @@ Result nqueens_gpu(int N, int g, int m, int M, int device, const std::string& mo
   pool.pushBack(nq_root());
   uint64_t tree = 0, sol = 0;
   const double t0 = now_sec();
-  nq_bfs_until(N, g, static_cast<size_t>(m), pool, tree, sol);
+  size_t target = static_cast<size_t>(m);
+  if (mode == "devpool")  // enough frontier to split across S device slices
+    target = std::max(target, static_cast<size_t>(2048) * devpool_slices());
+  nq_bfs_until(N, g, target, pool, tree, sol);
   const double p1 = now_sec() - t0;
   return nqueens_gpu_run(pool, N, g, m, M, device, mode, tree, sol, p1, capacity);
 }
@@ -439,48 +561,47 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
   } else if (mode == "devpool") {
     if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
       throw std::invalid_argument("devpool requires M * jobs <= 2^31");
-    DevGuard<PFSPNode> pool_d(capacity);
-    DevGuard<DevCtl> ctl_d(2);  // parity-alternating control blocks
-    const int G = devpool_grid(M, jobs, lbk);
-    const int stride = devpool_stride(lbk);
-    DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
-    DevGuard<uint32_t> bc_d(G), bs_d(G);
-    const size_t init = pool.size();
-    if (init > capacity) throw std::runtime_error("devpool capacity too small");
-    HIP_CHECK(
-        hipMemcpy(pool_d.p, pool.data(), init * sizeof(PFSPNode), hipMemcpyHostToDevice));
-    DevCtl ctl{};
-    ctl.size = init;
-    ctl.tree = tree;
-    ctl.sol = sol;
-    ctl.best = best;
-    HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-    HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-    r.h2d += 2;
-    r.h2d_bytes += init * sizeof(PFSPNode) + 2 * sizeof(DevCtl);
-
-    auto iter = [&](int parity) {
-      DevCtl* cur = ctl_d.p + parity;
-      DevCtl* next = ctl_d.p + (1 - parity);
-      launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk,
-                    tables.tb_sorted, m, M, stream.s);
-      launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, childbuf_d.p, pool_d.p, stride, G, m,
-                          M, capacity, stream.s);
-    };
-    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best);
-    tree = fin.tree;
-    sol = fin.sol;
-    best = fin.best;
-    r.gpu_iters = fin.iters;
-    pool.clear();
-    const size_t left = fin.size;
-    if (left > 0) {
-      std::vector<PFSPNode> tmp(left);
-      HIP_CHECK(
-          hipMemcpy(tmp.data(), pool_d.p, left * sizeof(PFSPNode), hipMemcpyDeviceToHost));
-      pool.pushBackBulk(tmp.data(), left);
-      r.d2h++;
-      r.d2h_bytes += left * sizeof(PFSPNode);
+    int S = devpool_slices();
+    if (pool.size() < static_cast<size_t>(S) * 64) S = 1;
+    std::vector<std::vector<PFSPNode>> slices(S);
+    {
+      const PFSPNode* src = pool.data();
+      const size_t total = pool.size();
+      for (int t = 0; t < S; t++) slices[t].reserve(total / S + 1);
+      for (size_t i = 0; i < total; i++) slices[i % S].push_back(src[i]);
+      pool.clear();
+    }
+    // slices share the incumbent through this atomic even when no external
+    // one is plugged in (cross-slice pruning; identical counts at ub=1)
+    std::atomic<int> local_best{best};
+    std::atomic<int>* sb = shared_best ? shared_best : &local_best;
+    std::vector<SliceOut> outs(S);
+    std::vector<std::vector<PFSPNode>> lefts(S);
+    std::vector<std::exception_ptr> errs(S);
+    std::vector<std::thread> threads;
+    for (int t = 0; t < S; t++) {
+      threads.emplace_back([&, t] {
+        try {
+          outs[t] = devpool_worker_pfsp(slices[t], I, tables.tb_sorted, lbk, best, m, M,
+                                        device, capacity, sb, lefts[t]);
+        } catch (...) {
+          errs[t] = std::current_exception();
+        }
+      });
+    }
+    for (auto& th : threads) th.join();
+    for (auto& e : errs)
+      if (e) std::rethrow_exception(e);
+    for (int t = 0; t < S; t++) {
+      tree += outs[t].fin.tree;
+      sol += outs[t].fin.sol;
+      if (outs[t].fin.best < best) best = outs[t].fin.best;
+      merge_slice_diag(r, outs[t].diag);
+      if (!lefts[t].empty()) pool.pushBackBulk(lefts[t].data(), lefts[t].size());
+    }
+    {
+      const int sbv = sb->load(std::memory_order_relaxed);
+      if (sbv < best) best = sbv;
     }
   } else {
     throw std::invalid_argument("mode must be hostpool or devpool");
@@ -512,7 +633,10 @@ Result pfsp_gpu(int inst, const std::string& lb_str, int ub, int m, int M, int d
   uint64_t tree = 0, sol = 0;
   int best = I.init_ub;
   const double t0 = now_sec();
-  pfsp_bfs_until(I, lb, static_cast<size_t>(m), pool, tree, sol, best);
+  size_t target = static_cast<size_t>(m);
+  if (mode == "devpool")
+    target = std::max(target, static_cast<size_t>(2048) * devpool_slices());
+  pfsp_bfs_until(I, lb, target, pool, tree, sol, best);
   const double p1 = now_sec() - t0;
   return pfsp_gpu_run(I, lb, pool, m, M, device, mode, tree, sol, best, p1, capacity);
 }
