@@ -17,6 +17,7 @@
 #include <algorithm>
 #include <atomic>
 #include <cstdlib>
+#include <mutex>
 #include <thread>
 #include <cstring>
 #include <stdexcept>
@@ -219,7 +220,12 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
   bool overflow = false;
   while (true) {
     if (graph_allowed && batches >= EAGER_BATCHES && exec == nullptr) {
-      HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
+      // one capture at a time: concurrent captures from the slice threads
+      // race inside ROCm 7.2 ("previous error during capture"); relaxed mode
+      // lets the other slices keep launching meanwhile
+      static std::mutex capture_mu;
+      std::lock_guard<std::mutex> lock(capture_mu);
+      HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeRelaxed));
       for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
       HIP_CHECK(hipStreamEndCapture(s, &graph));
       HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
@@ -440,7 +446,9 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
       throw std::invalid_argument("devpool requires M * N <= 2^31");
     const bool two_level = std::getenv("GATS_NQ_2LEVEL") != nullptr;
     int S = devpool_slices();
-    if (pool.size() < static_cast<size_t>(S) * 64) S = 1;
+    while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;  // small search
+    const unsigned long long cap_slice =
+        std::max<unsigned long long>(capacity / S, 1ull << 22);
     std::vector<std::vector<NQNode>> slices(S);
     {
       const NQNode* src = pool.data();
@@ -456,7 +464,7 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     for (int t = 0; t < S; t++) {
       threads.emplace_back([&, t] {
         try {
-          outs[t] = devpool_worker_nq(slices[t], N, g, m, M, device, two_level, capacity,
+          outs[t] = devpool_worker_nq(slices[t], N, g, m, M, device, two_level, cap_slice,
                                       lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
@@ -562,7 +570,9 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
       throw std::invalid_argument("devpool requires M * jobs <= 2^31");
     int S = devpool_slices();
-    if (pool.size() < static_cast<size_t>(S) * 64) S = 1;
+    while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;
+    const unsigned long long cap_slice =
+        std::max<unsigned long long>(capacity / S, 1ull << 22);
     std::vector<std::vector<PFSPNode>> slices(S);
     {
       const PFSPNode* src = pool.data();
@@ -583,7 +593,7 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       threads.emplace_back([&, t] {
         try {
           outs[t] = devpool_worker_pfsp(slices[t], I, tables.tb_sorted, lbk, best, m, M,
-                                        device, capacity, sb, lefts[t]);
+                                        device, cap_slice, sb, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
         }
