@@ -222,13 +222,41 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
     if (graph_allowed && batches >= EAGER_BATCHES && exec == nullptr) {
       // one capture at a time: concurrent captures from the slice threads
       // race inside ROCm 7.2 ("previous error during capture"); relaxed mode
-      // lets the other slices keep launching meanwhile
+      // lets the other slices keep launching meanwhile. A failed capture is
+      // not fatal — the loop just stays eager.
       static std::mutex capture_mu;
       std::lock_guard<std::mutex> lock(capture_mu);
-      HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeRelaxed));
-      for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
-      HIP_CHECK(hipStreamEndCapture(s, &graph));
-      HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
+      hipError_t ce = hipStreamBeginCapture(s, hipStreamCaptureModeRelaxed);
+      if (ce == hipSuccess) {
+        for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
+        ce = hipStreamEndCapture(s, &graph);
+        if (ce == hipSuccess) {
+          if (hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0) != hipSuccess) {
+            (void)hipGraphDestroy(graph);
+            graph = nullptr;
+            exec = nullptr;
+          }
+        } else {
+          graph = nullptr;
+        }
+        // a successfully captured batch was not EXECUTED; replay it below or
+        // re-run eagerly — either way the work happens exactly once
+        if (exec == nullptr) {
+          (void)hipGetLastError();
+          for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
+          batches++;
+          HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d, sizeof(DevCtl), hipMemcpyDeviceToHost, s));
+          HIP_CHECK(hipStreamSynchronize(s));
+          if (ctl_h.p->overflow) {
+            overflow = true;
+            break;
+          }
+          if (ctl_h.p->size < m) break;
+          continue;
+        }
+      } else {
+        (void)hipGetLastError();
+      }
     }
     if (exec != nullptr) {
       HIP_CHECK(hipGraphLaunch(exec, s));
