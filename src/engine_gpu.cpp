@@ -203,7 +203,8 @@ int lbk_of(LbKind lb) {
 template <class EnqueueIter>
 static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long m,
                                int kernels_per_iter, EnqueueIter&& enqueue_iter, Result& r,
-                               std::atomic<int>* shared_best = nullptr) {
+                               std::atomic<int>* shared_best = nullptr,
+                               bool allow_graph = true) {
   PinnedGuard<DevCtl> ctl_h(1);
   const int BATCH = 16;
   // GATS_NO_GRAPH=1 falls back to eager launches (rocprofv3 crashes tracing
@@ -212,7 +213,11 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
   // small searches (PFSP ta0xx finish in <10 ms), so the first EAGER_BATCHES
   // batches run eager and the graph is built only if the search is still
   // going.
-  const bool graph_allowed = std::getenv("GATS_NO_GRAPH") == nullptr;
+  // multi-slice engines pass allow_graph=false: ROCm 7.2 stream capture races
+  // with concurrent async work from the other slice threads no matter the
+  // capture mode; with >1 slice in flight the host launch latency is hidden
+  // by the other slices anyway
+  const bool graph_allowed = allow_graph && std::getenv("GATS_NO_GRAPH") == nullptr;
   const int EAGER_BATCHES = 4;
   int batches = 0;
   hipGraph_t graph = nullptr;
@@ -324,7 +329,7 @@ struct SliceOut {
 
 static SliceOut devpool_worker_nq(const std::vector<NQNode>& nodes, int N, int g, int m,
                                   int M, int device, bool two_level,
-                                  unsigned long long capacity,
+                                  unsigned long long capacity, bool allow_graph,
                                   std::vector<NQNode>& leftover) {
   HIP_CHECK(hipSetDevice(device));
   StreamGuard stream;
@@ -357,7 +362,7 @@ static SliceOut devpool_worker_nq(const std::vector<NQNode>& nodes, int N, int g
     launch_gather2_nq(cur, next, bc_d.p, bs_d.p, two_level ? be_d.p : nullptr, childbuf_d.p,
                       pool_d.p, stride, G, m, M, capacity, stream.s);
   };
-  out.fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r);
+  out.fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, nullptr, allow_graph);
   r.gpu_iters = out.fin.iters;
   const size_t left = out.fin.size;
   if (left > 0) {
@@ -374,7 +379,7 @@ static SliceOut devpool_worker_pfsp(const std::vector<PFSPNode>& nodes,
                                     const PfspInstance& I, const PfspDevTables& tb_sorted,
                                     int lbk, int best0, int m, int M, int device,
                                     unsigned long long capacity, std::atomic<int>* shared_best,
-                                    std::vector<PFSPNode>& leftover) {
+                                    bool allow_graph, std::vector<PFSPNode>& leftover) {
   HIP_CHECK(hipSetDevice(device));
   StreamGuard stream;
   SliceOut out;
@@ -406,7 +411,7 @@ static SliceOut devpool_worker_pfsp(const std::vector<PFSPNode>& nodes,
     launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, childbuf_d.p, pool_d.p, stride, G, m, M,
                         capacity, stream.s);
   };
-  out.fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best);
+  out.fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best, allow_graph);
   r.gpu_iters = out.fin.iters;
   const size_t left = out.fin.size;
   if (left > 0) {
@@ -493,7 +498,7 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
       threads.emplace_back([&, t] {
         try {
           outs[t] = devpool_worker_nq(slices[t], N, g, m, M, device, two_level, cap_slice,
-                                      lefts[t]);
+                                      S == 1, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
         }
@@ -621,7 +626,7 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       threads.emplace_back([&, t] {
         try {
           outs[t] = devpool_worker_pfsp(slices[t], I, tables.tb_sorted, lbk, best, m, M,
-                                        device, cap_slice, sb, lefts[t]);
+                                        device, cap_slice, sb, S == 1, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
         }
