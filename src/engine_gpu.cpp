@@ -480,8 +480,6 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     const bool two_level = std::getenv("GATS_NQ_2LEVEL") != nullptr;
     int S = devpool_slices();
     while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;  // small search
-    const unsigned long long cap_slice =
-        std::max<unsigned long long>(capacity / S, 1ull << 22);
     std::vector<std::vector<NQNode>> slices(S);
     {
       const NQNode* src = pool.data();
@@ -497,7 +495,7 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     for (int t = 0; t < S; t++) {
       threads.emplace_back([&, t] {
         try {
-          outs[t] = devpool_worker_nq(slices[t], N, g, m, M, device, two_level, cap_slice,
+          outs[t] = devpool_worker_nq(slices[t], N, g, m, M, device, two_level, capacity,
                                       S == 1, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
@@ -604,8 +602,6 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       throw std::invalid_argument("devpool requires M * jobs <= 2^31");
     int S = devpool_slices();
     while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;
-    const unsigned long long cap_slice =
-        std::max<unsigned long long>(capacity / S, 1ull << 22);
     std::vector<std::vector<PFSPNode>> slices(S);
     {
       const PFSPNode* src = pool.data();
@@ -626,7 +622,7 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       threads.emplace_back([&, t] {
         try {
           outs[t] = devpool_worker_pfsp(slices[t], I, tables.tb_sorted, lbk, best, m, M,
-                                        device, cap_slice, sb, S == 1, lefts[t]);
+                                        device, capacity, sb, S == 1, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
         }
