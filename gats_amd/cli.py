@@ -126,7 +126,7 @@ def main(argv=None):
             _phases(r)
         elif args.tier == "multigpu":
             gats_amd.require_gpu()
-            r = c.nqueens_multigpu(args.N, args.g, args.m, args.M, args.D, "gpu")
+            r = c.nqueens_multigpu(args.N, args.g, args.m, args.M, args.D, "devpool")
             _phases(r)
         else:
             from gats_amd import dist
@@ -158,7 +158,7 @@ def main(argv=None):
             _phases(r)
         elif args.tier == "multigpu":
             gats_amd.require_gpu()
-            r = c.pfsp_multigpu(inst, args.lb, args.ub, args.m, args.M, args.D, "gpu", False)
+            r = c.pfsp_multigpu(inst, args.lb, args.ub, args.m, args.M, args.D, "devpool", False)
             _phases(r)
         else:
             from gats_amd import dist

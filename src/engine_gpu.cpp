@@ -561,7 +561,7 @@ Result nqueens_gpu_from_pool(const std::vector<NQNode>& nodes, int N, int g, int
 Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int m, int M,
                     int device, const std::string& mode, uint64_t tree0, uint64_t sol0,
                     int best0, double phase1_time, unsigned long long capacity,
-                    std::atomic<int>* shared_best = nullptr) {
+                    std::atomic<int>* shared_best /*= nullptr*/) {
   Result r;
   r.phases.push_back({tree0, sol0, phase1_time});
   uint64_t tree = tree0, sol = sol0;
@@ -677,7 +677,8 @@ Result pfsp_gpu(int inst, const std::string& lb_str, int ub, int m, int M, int d
     target = std::max(target, static_cast<size_t>(2048) * devpool_slices());
   pfsp_bfs_until(I, lb, target, pool, tree, sol, best);
   const double p1 = now_sec() - t0;
-  return pfsp_gpu_run(I, lb, pool, m, M, device, mode, tree, sol, best, p1, capacity);
+  return pfsp_gpu_run(I, lb, pool, m, M, device, mode, tree, sol, best, p1, capacity,
+                      nullptr);
 }
 
 Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst,
@@ -688,7 +689,8 @@ Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst,
   Pool<PFSPNode> pool;
   pool.pushBackBulk(nodes.data(), nodes.size());
   const int best = (best0 > 0) ? best0 : I.init_ub;
-  return pfsp_gpu_run(I, lb, pool, m, M, device, mode, 0, 0, best, 0.0, capacity);
+  return pfsp_gpu_run(I, lb, pool, m, M, device, mode, 0, 0, best, 0.0, capacity,
+                      nullptr);
 }
 
 // ---------------------------------------------------------------------------

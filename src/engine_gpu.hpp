@@ -12,6 +12,16 @@ namespace gats {
 
 int gpu_device_count();
 
+// Phase-2+3 engine cores (phase-1 pool supplied by the caller); used by the
+// CLI engines, the distributed tier and the multi-GPU tier's devpool workers.
+Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int device,
+                       const std::string& mode, uint64_t tree0, uint64_t sol0,
+                       double phase1_time, unsigned long long capacity);
+Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int m, int M,
+                    int device, const std::string& mode, uint64_t tree0, uint64_t sol0,
+                    int best0, double phase1_time, unsigned long long capacity,
+                    std::atomic<int>* shared_best);
+
 Result nqueens_gpu(int N, int g, int m, int M, int device, const std::string& mode,
                    unsigned long long capacity);
 Result nqueens_gpu_from_pool(const std::vector<NQNode>& nodes, int N, int g, int m, int M,

@@ -28,13 +28,11 @@
 #include <thread>
 #include <vector>
 
+#include "engine_gpu.hpp"
 #include "gpu_api.hpp"
 #include "search_host.hpp"
 
 namespace gats {
-
-// from engine_gpu.cpp
-int gpu_device_count();
 
 namespace {
 
@@ -234,8 +232,139 @@ struct PfspGpuCtx {
 
 }  // namespace
 
+// devpool-per-worker variant: each worker drives the full single-GPU devpool
+// engine (with its own stream slices) on its partition — the performance
+// path for the CLI multigpu tier. No inter-worker stealing (like the
+// reference's own CUDA multi-GPU baseline, nqueens_multigpu_cuda.cu:287-313);
+// the deep round-robin partition keeps workers statistically balanced, and
+// PFSP workers share the incumbent through one atomic.
+static Result nqueens_multigpu_devpool(int N, int g, int m, int M, int D,
+                                       unsigned long long capacity) {
+  Result r;
+  Pool<NQNode> pool;
+  pool.pushBack(nq_root());
+  uint64_t tree = 0, sol = 0;
+  const double t0 = now_sec();
+  nq_bfs_until(N, g, std::max<size_t>(static_cast<size_t>(D) * m, 8192 * D), pool, tree,
+               sol);
+  const double p1 = now_sec() - t0;
+  r.phases.push_back({tree, sol, p1});
+
+  const int ndev = gpu_device_count();
+  if (ndev == 0) throw std::runtime_error("no HIP device visible");
+  std::vector<Pool<NQNode>> parts(D);
+  {
+    const NQNode* src = pool.data();
+    for (size_t i = 0; i < pool.size(); i++) parts[i % D].pushBack(src[i]);
+    pool.clear();
+  }
+  std::vector<Result> outs(D);
+  std::vector<std::exception_ptr> errs(D);
+  std::vector<std::thread> threads;
+  const double t2 = now_sec();
+  for (int d = 0; d < D; d++) {
+    threads.emplace_back([&, d] {
+      try {
+        outs[d] = nqueens_gpu_run(parts[d], N, g, m, M, d % ndev, "devpool", 0, 0, 0.0,
+                                  capacity);
+      } catch (...) {
+        errs[d] = std::current_exception();
+      }
+    });
+  }
+  for (auto& t : threads) t.join();
+  for (auto& e : errs)
+    if (e) std::rethrow_exception(e);
+  uint64_t tree2 = tree, sol2 = sol;
+  for (auto& o : outs) {
+    tree2 += o.tree;
+    sol2 += o.sol;
+    r.per_worker.push_back(o.tree);
+    r.kernel_launch += o.kernel_launch;
+    r.h2d += o.h2d;
+    r.d2h += o.d2h;
+    r.h2d_bytes += o.h2d_bytes;
+    r.d2h_bytes += o.d2h_bytes;
+    r.gpu_iters += o.gpu_iters;
+  }
+  const double t3 = now_sec();
+  r.phases.push_back({tree2 - tree, sol2 - sol, t3 - t2});
+  r.phases.push_back({0, 0, 0.0});  // workers drained their own phase 3
+  r.tree = tree2;
+  r.sol = sol2;
+  r.gpu_time = t3 - t2;
+  r.time = p1 + (t3 - t2);
+  return r;
+}
+
+static Result pfsp_multigpu_devpool(int inst, const std::string& lb_str, int ub, int m,
+                                    int M, int D, unsigned long long capacity) {
+  Result r;
+  const LbKind lb = lb_from_string(lb_str);
+  PfspInstance I = make_pfsp_instance(inst, ub);
+  Pool<PFSPNode> pool;
+  pool.pushBack(pfsp_root());
+  uint64_t tree = 0, sol = 0;
+  int best = I.init_ub;
+  const double t0 = now_sec();
+  pfsp_bfs_until(I, lb, std::max<size_t>(static_cast<size_t>(D) * m, 8192 * D), pool, tree,
+                 sol, best);
+  const double p1 = now_sec() - t0;
+  r.phases.push_back({tree, sol, p1});
+
+  const int ndev = gpu_device_count();
+  if (ndev == 0) throw std::runtime_error("no HIP device visible");
+  std::vector<Pool<PFSPNode>> parts(D);
+  {
+    const PFSPNode* src = pool.data();
+    for (size_t i = 0; i < pool.size(); i++) parts[i % D].pushBack(src[i]);
+    pool.clear();
+  }
+  std::atomic<int> shared_best{best};
+  std::vector<Result> outs(D);
+  std::vector<std::exception_ptr> errs(D);
+  std::vector<std::thread> threads;
+  const double t2 = now_sec();
+  for (int d = 0; d < D; d++) {
+    threads.emplace_back([&, d] {
+      try {
+        outs[d] = pfsp_gpu_run(I, lb, parts[d], m, M, d % ndev, "devpool", 0, 0, best, 0.0,
+                               capacity, &shared_best);
+      } catch (...) {
+        errs[d] = std::current_exception();
+      }
+    });
+  }
+  for (auto& t : threads) t.join();
+  for (auto& e : errs)
+    if (e) std::rethrow_exception(e);
+  uint64_t tree2 = tree, sol2 = sol;
+  for (auto& o : outs) {
+    tree2 += o.tree;
+    sol2 += o.sol;
+    if (o.optimum < best) best = o.optimum;
+    r.per_worker.push_back(o.tree);
+    r.kernel_launch += o.kernel_launch;
+    r.h2d += o.h2d;
+    r.d2h += o.d2h;
+    r.h2d_bytes += o.h2d_bytes;
+    r.d2h_bytes += o.d2h_bytes;
+    r.gpu_iters += o.gpu_iters;
+  }
+  const double t3 = now_sec();
+  r.phases.push_back({tree2 - tree, sol2 - sol, t3 - t2});
+  r.phases.push_back({0, 0, 0.0});
+  r.tree = tree2;
+  r.sol = sol2;
+  r.optimum = best;
+  r.gpu_time = t3 - t2;
+  r.time = p1 + (t3 - t2);
+  return r;
+}
+
 Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& eval) {
   if (D < 1) throw std::invalid_argument("D must be >= 1");
+  if (eval == "devpool") return nqueens_multigpu_devpool(N, g, m, M, D, 1ull << 27);
   Result r;
   Pool<NQNode> pool;
   pool.pushBack(nq_root());
@@ -341,6 +470,7 @@ Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& ev
 Result pfsp_multigpu(int inst, const std::string& lb_str, int ub, int m, int M, int D,
                      const std::string& eval, bool share_best) {
   if (D < 1) throw std::invalid_argument("D must be >= 1");
+  if (eval == "devpool") return pfsp_multigpu_devpool(inst, lb_str, ub, m, M, D, 1ull << 27);
   const LbKind lb = lb_from_string(lb_str);
   PfspInstance I = make_pfsp_instance(inst, ub);
   Result r;

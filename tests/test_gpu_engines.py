@@ -76,3 +76,20 @@ def test_pfsp_async_engine_adopts_lower_ub(gpu):
     eng.update_best(1377)  # the known optimum, as if another rank found it
     r = eng.join()
     assert r["optimum"] == 1377
+
+
+def test_multigpu_devpool_matches_seq(gpu):
+    # devpool-per-worker multi-GPU tier (workers share one device here)
+    seq = gpu.nqueens_seq(14, 1)
+    r = gpu.nqueens_multigpu(14, 1, 25, 50000, 2, "devpool")
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    assert len(r["per_worker_tree"]) == 2
+
+
+def test_pfsp_multigpu_devpool_ub1(gpu):
+    seq = gpu.pfsp_seq(14, "lb2", 1)
+    r = gpu.pfsp_multigpu(14, "lb2", 1, 25, 50000, 2, "devpool", False)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    assert r["optimum"] == 1377
