@@ -327,10 +327,14 @@ struct SliceOut {
   Result diag;
 };
 
-static SliceOut devpool_worker_nq(const std::vector<NQNode>& nodes, int N, int g, int m,
-                                  int M, int device, bool two_level,
-                                  unsigned long long capacity, bool allow_graph,
-                                  std::vector<NQNode>& leftover) {
+// Queue-driven worker thread: allocates its device buffers once, then keeps
+// claiming frontier slices from the shared index until none remain. Slices
+// are oversubscribed (~4 per thread) so a thread whose slice finishes early
+// just pulls the next one — tail balancing without inter-thread stealing.
+static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices,
+                                  std::atomic<int>& next_slice, int N, int g, int m, int M,
+                                  int device, bool two_level, unsigned long long capacity,
+                                  bool allow_graph, std::vector<NQNode>& leftover) {
   HIP_CHECK(hipSetDevice(device));
   StreamGuard stream;
   SliceOut out;
@@ -341,15 +345,6 @@ static SliceOut devpool_worker_nq(const std::vector<NQNode>& nodes, int N, int g
   const int stride = two_level ? devpool_stride(1) * (MAX_JOBS - 1) : devpool_stride(1);
   DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G), bs_d(G), be_d(G);
-  const size_t init = nodes.size();
-  if (init > capacity) throw std::runtime_error("devpool capacity too small");
-  HIP_CHECK(hipMemcpy(pool_d.p, nodes.data(), init * sizeof(NQNode), hipMemcpyHostToDevice));
-  DevCtl ctl{};
-  ctl.size = init;
-  HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-  r.h2d += 2;
-  r.h2d_bytes += init * sizeof(NQNode) + 2 * sizeof(DevCtl);
 
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
@@ -362,27 +357,47 @@ static SliceOut devpool_worker_nq(const std::vector<NQNode>& nodes, int N, int g
     launch_gather2_nq(cur, next, bc_d.p, bs_d.p, two_level ? be_d.p : nullptr, childbuf_d.p,
                       pool_d.p, stride, G, m, M, capacity, stream.s);
   };
-  out.fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, nullptr, allow_graph);
-  r.gpu_iters = out.fin.iters;
-  const size_t left = out.fin.size;
-  if (left > 0) {
-    leftover.resize(left);
+
+  int si;
+  while ((si = next_slice.fetch_add(1)) < static_cast<int>(slices.size())) {
+    const std::vector<NQNode>& nodes = slices[si];
+    if (nodes.empty()) continue;
+    if (nodes.size() > capacity) throw std::runtime_error("devpool capacity too small");
     HIP_CHECK(
-        hipMemcpy(leftover.data(), pool_d.p, left * sizeof(NQNode), hipMemcpyDeviceToHost));
-    r.d2h++;
-    r.d2h_bytes += left * sizeof(NQNode);
+        hipMemcpy(pool_d.p, nodes.data(), nodes.size() * sizeof(NQNode), hipMemcpyHostToDevice));
+    DevCtl ctl{};
+    ctl.size = nodes.size();
+    HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    r.h2d += 2;
+    r.h2d_bytes += nodes.size() * sizeof(NQNode) + 2 * sizeof(DevCtl);
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, nullptr,
+                                        allow_graph);
+    out.fin.tree += fin.tree;
+    out.fin.sol += fin.sol;
+    r.gpu_iters += fin.iters;
+    if (fin.size > 0) {
+      const size_t base = leftover.size();
+      leftover.resize(base + fin.size);
+      HIP_CHECK(hipMemcpy(leftover.data() + base, pool_d.p, fin.size * sizeof(NQNode),
+                          hipMemcpyDeviceToHost));
+      r.d2h++;
+      r.d2h_bytes += fin.size * sizeof(NQNode);
+    }
   }
   return out;
 }
 
-static SliceOut devpool_worker_pfsp(const std::vector<PFSPNode>& nodes,
-                                    const PfspInstance& I, const PfspDevTables& tb_sorted,
-                                    int lbk, int best0, int m, int M, int device,
-                                    unsigned long long capacity, std::atomic<int>* shared_best,
-                                    bool allow_graph, std::vector<PFSPNode>& leftover) {
+static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& slices,
+                                    std::atomic<int>& next_slice, const PfspInstance& I,
+                                    const PfspDevTables& tb_sorted, int lbk, int best0,
+                                    int m, int M, int device, unsigned long long capacity,
+                                    std::atomic<int>* shared_best, bool allow_graph,
+                                    std::vector<PFSPNode>& leftover) {
   HIP_CHECK(hipSetDevice(device));
   StreamGuard stream;
   SliceOut out;
+  out.fin.best = best0;
   Result& r = out.diag;
   const int jobs = I.jobs, machines = I.machines;
   DevGuard<PFSPNode> pool_d(capacity);
@@ -391,17 +406,6 @@ static SliceOut devpool_worker_pfsp(const std::vector<PFSPNode>& nodes,
   const int stride = devpool_stride(lbk);
   DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G), bs_d(G);
-  const size_t init = nodes.size();
-  if (init > capacity) throw std::runtime_error("devpool capacity too small");
-  HIP_CHECK(
-      hipMemcpy(pool_d.p, nodes.data(), init * sizeof(PFSPNode), hipMemcpyHostToDevice));
-  DevCtl ctl{};
-  ctl.size = init;
-  ctl.best = best0;
-  HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-  r.h2d += 2;
-  r.h2d_bytes += init * sizeof(PFSPNode) + 2 * sizeof(DevCtl);
 
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
@@ -411,15 +415,36 @@ static SliceOut devpool_worker_pfsp(const std::vector<PFSPNode>& nodes,
     launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, childbuf_d.p, pool_d.p, stride, G, m, M,
                         capacity, stream.s);
   };
-  out.fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best, allow_graph);
-  r.gpu_iters = out.fin.iters;
-  const size_t left = out.fin.size;
-  if (left > 0) {
-    leftover.resize(left);
-    HIP_CHECK(
-        hipMemcpy(leftover.data(), pool_d.p, left * sizeof(PFSPNode), hipMemcpyDeviceToHost));
-    r.d2h++;
-    r.d2h_bytes += left * sizeof(PFSPNode);
+
+  int si;
+  while ((si = next_slice.fetch_add(1)) < static_cast<int>(slices.size())) {
+    const std::vector<PFSPNode>& nodes = slices[si];
+    if (nodes.empty()) continue;
+    if (nodes.size() > capacity) throw std::runtime_error("devpool capacity too small");
+    HIP_CHECK(hipMemcpy(pool_d.p, nodes.data(), nodes.size() * sizeof(PFSPNode),
+                        hipMemcpyHostToDevice));
+    DevCtl ctl{};
+    ctl.size = nodes.size();
+    // adopt the freshest incumbent before starting the slice
+    ctl.best = shared_best ? shared_best->load(std::memory_order_relaxed) : best0;
+    HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    r.h2d += 2;
+    r.h2d_bytes += nodes.size() * sizeof(PFSPNode) + 2 * sizeof(DevCtl);
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best,
+                                        allow_graph);
+    out.fin.tree += fin.tree;
+    out.fin.sol += fin.sol;
+    if (fin.best < out.fin.best) out.fin.best = fin.best;
+    r.gpu_iters += fin.iters;
+    if (fin.size > 0) {
+      const size_t base = leftover.size();
+      leftover.resize(base + fin.size);
+      HIP_CHECK(hipMemcpy(leftover.data() + base, pool_d.p, fin.size * sizeof(PFSPNode),
+                          hipMemcpyDeviceToHost));
+      r.d2h++;
+      r.d2h_bytes += fin.size * sizeof(PFSPNode);
+    }
   }
   return out;
 }
@@ -480,14 +505,16 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     const bool two_level = std::getenv("GATS_NQ_2LEVEL") != nullptr;
     int S = devpool_slices();
     while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;  // small search
-    std::vector<std::vector<NQNode>> slices(S);
+    const int NS = (S == 1) ? 1 : S * 4;  // oversubscribe: ~4 queued slices/thread
+    std::vector<std::vector<NQNode>> slices(NS);
     {
       const NQNode* src = pool.data();
       const size_t total = pool.size();
-      for (int t = 0; t < S; t++) slices[t].reserve(total / S + 1);
-      for (size_t i = 0; i < total; i++) slices[i % S].push_back(src[i]);
+      for (int t = 0; t < NS; t++) slices[t].reserve(total / NS + 1);
+      for (size_t i = 0; i < total; i++) slices[i % NS].push_back(src[i]);
       pool.clear();
     }
+    std::atomic<int> next_slice{0};
     std::vector<SliceOut> outs(S);
     std::vector<std::vector<NQNode>> lefts(S);
     std::vector<std::exception_ptr> errs(S);
@@ -495,8 +522,8 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     for (int t = 0; t < S; t++) {
       threads.emplace_back([&, t] {
         try {
-          outs[t] = devpool_worker_nq(slices[t], N, g, m, M, device, two_level, capacity,
-                                      S == 1, lefts[t]);
+          outs[t] = devpool_thread_nq(slices, next_slice, N, g, m, M, device, two_level,
+                                      capacity, S == 1, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
         }
@@ -602,18 +629,20 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       throw std::invalid_argument("devpool requires M * jobs <= 2^31");
     int S = devpool_slices();
     while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;
-    std::vector<std::vector<PFSPNode>> slices(S);
+    const int NS = (S == 1) ? 1 : S * 4;
+    std::vector<std::vector<PFSPNode>> slices(NS);
     {
       const PFSPNode* src = pool.data();
       const size_t total = pool.size();
-      for (int t = 0; t < S; t++) slices[t].reserve(total / S + 1);
-      for (size_t i = 0; i < total; i++) slices[i % S].push_back(src[i]);
+      for (int t = 0; t < NS; t++) slices[t].reserve(total / NS + 1);
+      for (size_t i = 0; i < total; i++) slices[i % NS].push_back(src[i]);
       pool.clear();
     }
     // slices share the incumbent through this atomic even when no external
     // one is plugged in (cross-slice pruning; identical counts at ub=1)
     std::atomic<int> local_best{best};
     std::atomic<int>* sb = shared_best ? shared_best : &local_best;
+    std::atomic<int> next_slice{0};
     std::vector<SliceOut> outs(S);
     std::vector<std::vector<PFSPNode>> lefts(S);
     std::vector<std::exception_ptr> errs(S);
@@ -621,8 +650,8 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     for (int t = 0; t < S; t++) {
       threads.emplace_back([&, t] {
         try {
-          outs[t] = devpool_worker_pfsp(slices[t], I, tables.tb_sorted, lbk, best, m, M,
-                                        device, capacity, sb, S == 1, lefts[t]);
+          outs[t] = devpool_thread_pfsp(slices, next_slice, I, tables.tb_sorted, lbk, best,
+                                        m, M, device, capacity, sb, S == 1, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
         }
