@@ -9,9 +9,11 @@
 //       host buffers, async prefix-only copies (the Chapel version copies the
 //       full M-element arrays every iteration, SURVEY.md §8.4 — we don't),
 //       one HIP stream, host-side generate_children.
-//   "devpool": MI355X-native fast path — the pool lives in HBM3E, each
-//       iteration is a begin/copy/expand kernel triple with on-device pruning
-//       and appends; the host polls a 64 B control block every few iterations.
+//   "devpool": MI355X-native fast path — pools live in HBM3E; each offload
+//       round is an expand-compact + gather kernel pair (kernels.hip), with
+//       ~16 frontier slices pulled off a queue by 4 worker threads on
+//       concurrent streams to fill the chip; the host polls 64 B control
+//       blocks every few iterations.
 #include <hip/hip_runtime.h>
 
 #include <algorithm>

@@ -9,12 +9,11 @@
 //   * Two modes:
 //       "hostpool"  — reference-shaped: kernel evaluates labels/bounds, host
 //                     prunes and branches (used for oracle parity tests).
-//       "devpool"   — device-resident pool: a begin/copy/expand kernel triple
-//                     per iteration keeps the entire hot loop on the GPU;
-//                     children are pruned and appended on-device with
-//                     wave-aggregated (wave64 ballot + one atomic per wave)
-//                     pool reservations; the host only polls a 64 B control
-//                     block every few iterations.
+//       "devpool"   — device-resident pool: two kernels per offload round
+//                     (expand-compact + self-prefixing gather with parity
+//                     control blocks) keep the entire hot loop on the GPU;
+//                     the host only polls a 64 B control block every few
+//                     iterations. No global atomics on the hot path.
 //   * Bound tables staged in LDS (p_times/lags int16, Johnson schedules u8):
 //     ~12 KB for 20x20 vs 160 KB per CU.
 //   * No runtime-indexed per-thread arrays (they would spill to scratch on
