@@ -268,49 +268,44 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
   scheduled |= 1u << job_k;
 
   constexpr int PAIRS = LdsLb2<MM>::PAIRS;
-  constexpr int ILP = (PAIRS >= 8) ? 4 : 2;
   int lb = 0;
-  // ILP pairs in flight per step: each pair's (tmp0, tmp1) update chain is a
-  // serial LDS-load -> add -> max dependency; interleaving independent chains
-  // hides the ds_read_b64 latency. Out-of-range chains recompute the last
-  // pair and are skipped at merge time.
-  for (int l = 0; l < PAIRS; l += ILP) {
-    int ma0[ILP], ma1[ILP], t0[ILP], t1[ILP];
-    const uint64_t* jp[ILP];
-#pragma unroll
-    for (int c = 0; c < ILP; c++) {
-      const int li = (l + c < PAIRS) ? l + c : PAIRS - 1;
-      ma0[c] = lds.pair1[li];
-      ma1[c] = lds.pair2[li];
-      t0[c] = front[ma0[c]];
-      t1[c] = front[ma1[c]];
-      jp[c] = &lds.jp[li * jobs];
-    }
+  // two pairs in flight per step: each pair's (tmp0, tmp1) update chain is a
+  // serial LDS-load -> add -> max dependency; interleaving two independent
+  // chains doubles the ILP available to hide the ds_read_b64 latency
+  for (int l = 0; l < PAIRS; l += 2) {
+    const bool two = (l + 1) < PAIRS;
+    const int ma0a = lds.pair1[l], ma1a = lds.pair2[l];
+    const int ma0b = lds.pair1[two ? l + 1 : l], ma1b = lds.pair2[two ? l + 1 : l];
+    int t0a = front[ma0a], t1a = front[ma1a];
+    int t0b = front[ma0b], t1b = front[ma1b];
+    const uint64_t* jpa = &lds.jp[l * jobs];
+    const uint64_t* jpb = &lds.jp[(two ? l + 1 : l) * jobs];
     for (int j = 0; j < jobs; j++) {
-#pragma unroll
-      for (int c = 0; c < ILP; c++) {
-        const uint64_t v = jp[c][j];  // one ds_read_b64 replaces 4 scalar LDS reads
-        const int job = static_cast<int>(v >> 48);
-        if (!(scheduled >> job & 1u)) {
-          t0[c] += static_cast<int>(v & 0xffff);
-          t1[c] = max(t1[c], t0[c] + static_cast<int>((v >> 32) & 0xffff));
-          t1[c] += static_cast<int>((v >> 16) & 0xffff);
-        }
+      const uint64_t va = jpa[j];  // one ds_read_b64 replaces 4 scalar LDS reads
+      const uint64_t vb = jpb[j];
+      const int ja = static_cast<int>(va >> 48);
+      const int jb = static_cast<int>(vb >> 48);
+      if (!(scheduled >> ja & 1u)) {
+        t0a += static_cast<int>(va & 0xffff);
+        t1a = max(t1a, t0a + static_cast<int>((va >> 32) & 0xffff));
+        t1a += static_cast<int>((va >> 16) & 0xffff);
+      }
+      if (!(scheduled >> jb & 1u)) {
+        t0b += static_cast<int>(vb & 0xffff);
+        t1b = max(t1b, t0b + static_cast<int>((vb >> 32) & 0xffff));
+        t1b += static_cast<int>((vb >> 16) & 0xffff);
       }
     }
-    // merge in pair order with a check after each: keeps the returned value
-    // bit-equal to the reference's per-pair early exit
-    // (c_bound_johnson.c:231-233), which the hostpool-vs-CPU-oracle tests
-    // assert; later chains' work is wasted only on the exit iteration
-    bool out = false;
-#pragma unroll
-    for (int c = 0; c < ILP; c++) {
-      if (!out && l + c < PAIRS) {
-        lb = max(lb, max(t1[c] + lds.min_tails[ma1[c]], t0[c] + lds.min_tails[ma0[c]]));
-        if (lb > best) out = true;
-      }
+    // merge pair a, check, then pair b: keeps the returned value bit-equal to
+    // the reference's per-pair early exit (c_bound_johnson.c:231-233), which
+    // the hostpool-vs-CPU-oracle tests assert; pair b's work is wasted only
+    // on the exit iteration
+    lb = max(lb, max(t1a + lds.min_tails[ma1a], t0a + lds.min_tails[ma0a]));
+    if (lb > best) break;
+    if (two) {
+      lb = max(lb, max(t1b + lds.min_tails[ma1b], t0b + lds.min_tails[ma0b]));
+      if (lb > best) break;
     }
-    if (out) break;
   }
   return lb;
 }
