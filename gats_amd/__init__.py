@@ -17,12 +17,15 @@ def core():
     global _core_mod
     if _core_mod is None:
         import importlib
+        import sys
 
         try:
             _core_mod = importlib.import_module("gats_amd._core")
         except ImportError:
-            import build as _build  # repo-root build driver
-
+            repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+            if repo not in sys.path:
+                sys.path.insert(0, repo)
+            _build = importlib.import_module("build")  # repo-root build driver
             _build.build()
             _core_mod = importlib.import_module("gats_amd._core")
     return _core_mod

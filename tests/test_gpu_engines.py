@@ -93,3 +93,21 @@ def test_pfsp_multigpu_devpool_ub1(gpu):
     assert r["tree"] == seq["tree"]
     assert r["sol"] == seq["sol"]
     assert r["optimum"] == 1377
+
+
+def test_nqueens_gpu_g_multiplier(gpu):
+    # the artificial work multiplier g must not change counts (GPU path)
+    seq = gpu.nqueens_seq(12, 1)
+    r = gpu.nqueens_gpu(12, 3, 25, 50000, 0, "devpool", 1 << 22)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+
+
+def test_nqueens_gpu_mM_window(gpu):
+    # non-default offload window: same counts (popBackBulk semantics)
+    seq = gpu.nqueens_seq(13, 1)
+    r = gpu.nqueens_gpu(13, 1, 1000, 2000, 0, "devpool", 1 << 24)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    r2 = gpu.nqueens_gpu(13, 1, 1000, 2000, 0, "hostpool", 1 << 24)
+    assert r2["tree"] == seq["tree"] and r2["sol"] == seq["sol"]
