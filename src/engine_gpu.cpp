@@ -335,7 +335,7 @@ struct SliceOut {
 // just pulls the next one — tail balancing without inter-thread stealing.
 static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices,
                                   std::atomic<int>& next_slice, int N, int g, int m, int M,
-                                  int device, bool two_level, unsigned long long capacity,
+                                  int device, int finish, unsigned long long capacity,
                                   bool allow_graph, std::vector<NQNode>& leftover) {
   HIP_CHECK(hipSetDevice(device));
   StreamGuard stream;
@@ -344,20 +344,18 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
   DevGuard<NQNode> pool_d(capacity);
   DevGuard<DevCtl> ctl_d(2);  // parity-alternating control blocks
   const int G = devpool_grid(M, N, 1);
-  const int stride = two_level ? devpool_stride(1) * (MAX_JOBS - 1) : devpool_stride(1);
+  const int stride = devpool_stride(1);
   DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
-  DevGuard<uint32_t> bc_d(G), bs_d(G), be_d(G);
+  DevGuard<uint32_t> bc_d(G);
+  DevGuard<unsigned long long> bs_d(G), be_d(G);
 
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
-    if (two_level)
-      launch_nq_x2(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, m, M,
-                   stream.s);
-    else
-      launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, N, g, m, M, stream.s);
-    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, two_level ? be_d.p : nullptr, childbuf_d.p,
-                      pool_d.p, stride, G, m, M, capacity, stream.s);
+    launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, m, M,
+                stream.s);
+    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, childbuf_d.p, pool_d.p, stride, G,
+                      m, M, capacity, stream.s);
   };
 
   int si;
@@ -407,7 +405,8 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   const int G = devpool_grid(M, jobs, lbk);
   const int stride = devpool_stride(lbk);
   DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
-  DevGuard<uint32_t> bc_d(G), bs_d(G);
+  DevGuard<uint32_t> bc_d(G);
+  DevGuard<unsigned long long> bs_d(G);
 
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
@@ -504,7 +503,10 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
   } else if (mode_eff == "devpool") {
     if (static_cast<unsigned long long>(M) * N > (1ull << 31))
       throw std::invalid_argument("devpool requires M * N <= 2^31");
-    const bool two_level = std::getenv("GATS_NQ_2LEVEL") != nullptr;
+    // depth of the in-thread bitmask subtree finisher (levels from the bottom)
+    int finish = 6;
+    if (const char* e = std::getenv("GATS_NQ_FINISH")) finish = atoi(e);
+    if (finish > 8) finish = 8;  // template recursion budget (NQ_FINISH_MAX)
     int S = devpool_slices();
     while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;  // small search
     const int NS = (S == 1) ? 1 : S * 4;  // oversubscribe: ~4 queued slices/thread
@@ -524,7 +526,7 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     for (int t = 0; t < S; t++) {
       threads.emplace_back([&, t] {
         try {
-          outs[t] = devpool_thread_nq(slices, next_slice, N, g, m, M, device, two_level,
+          outs[t] = devpool_thread_nq(slices, next_slice, N, g, m, M, device, finish,
                                       capacity, S == 1, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();

@@ -58,20 +58,20 @@ void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, in
 int devpool_grid(unsigned long long M, int per, int lbk);
 int devpool_stride(int lbk);
 void launch_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
-                 uint32_t* blockCounts, uint32_t* blockSols, int N, int g,
+                 uint32_t* blockCounts, unsigned long long* blockSols,
+                 unsigned long long* blockExtra, int N, int g, int finish,
                  unsigned long long m, unsigned long long M, hipStream_t s);
 void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32_t* bc,
-                   uint32_t* bs, int jobs, int machines, int lbk, const PfspDevTables& tb,
-                   unsigned long long m, unsigned long long M, hipStream_t s);
-void launch_nq_x2(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
-                  uint32_t* blockCounts, uint32_t* blockSols, uint32_t* blockExtra, int N,
-                  int g, unsigned long long m, unsigned long long M, hipStream_t s);
+                   unsigned long long* bs, int jobs, int machines, int lbk,
+                   const PfspDevTables& tb, unsigned long long m, unsigned long long M,
+                   hipStream_t s);
 void launch_gather2_nq(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
-                       const uint32_t* bs, const uint32_t* be, const NQNode* childbuf,
-                       NQNode* pool, int strideNodes, int G, unsigned long long m,
-                       unsigned long long M, unsigned long long capacity, hipStream_t s);
+                       const unsigned long long* bs, const unsigned long long* be,
+                       const NQNode* childbuf, NQNode* pool, int strideNodes, int G,
+                       unsigned long long m, unsigned long long M,
+                       unsigned long long capacity, hipStream_t s);
 void launch_gather2_pfsp(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
-                         const uint32_t* bs, const PFSPNode* childbuf, PFSPNode* pool,
-                         int strideNodes, int G, unsigned long long m, unsigned long long M,
-                         unsigned long long capacity, hipStream_t s);
+                         const unsigned long long* bs, const PFSPNode* childbuf,
+                         PFSPNode* pool, int strideNodes, int G, unsigned long long m,
+                         unsigned long long M, unsigned long long capacity, hipStream_t s);
 }  // namespace gats

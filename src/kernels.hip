@@ -435,6 +435,49 @@ __device__ inline uint32_t block_excl_scan(uint32_t v, uint32_t& total) {
   return wbase + x - v;
 }
 
+// Block-wide sum of a u64 per-thread value (for subtree-finisher counters,
+// which can exceed 32 bits per block).
+__device__ inline unsigned long long block_reduce_u64(unsigned long long v) {
+  const int wid = threadIdx.x >> 6;
+#pragma unroll
+  for (int d = 32; d >= 1; d >>= 1) v += __shfl_xor(v, d);
+  __shared__ unsigned long long wsum[4];
+  __syncthreads();
+  if ((threadIdx.x & 63) == 0) wsum[wid] = v;
+  __syncthreads();
+  return wsum[0] + wsum[1] + wsum[2] + wsum[3];
+}
+
+// Register-resident bitmask DFS over the last levels of the N-Queens tree:
+// counts every safe node (tree) and every depth-N placement (sol) below the
+// current state. Equivalent to the reference's decomposition: candidate
+// VALUES are the unused ones (cols mask == the permutation's remaining
+// values) and safety is the two diagonal masks — identical node set, and
+// N-Queens counts are traversal-order independent. Template depth budget B
+// keeps every stack frame in registers (no runtime-indexed arrays).
+template <int B>
+__device__ inline void nq_dfs(uint32_t cols, uint32_t d1, uint32_t d2, int placed, int N,
+                              unsigned long long& tree, unsigned long long& sol) {
+  if constexpr (B == 0) {
+    return;
+  } else {
+    const uint32_t msk = (1u << N) - 1u;
+    uint32_t free = ~(cols | d1 | d2) & msk;
+    while (free) {
+      const uint32_t bit = free & (0u - free);
+      free ^= bit;
+      tree++;
+      if (placed + 1 == N) {
+        sol++;
+      } else {
+        nq_dfs<B - 1>(cols | bit, ((d1 | bit) << 1) & msk, (d2 | bit) >> 1, placed + 1, N,
+                      tree, sol);
+      }
+    }
+  }
+}
+constexpr int NQ_FINISH_MAX = 8;
+
 // popBackBulk chunk, re-derived identically by every kernel of an iteration
 // (pure function of ctl->size, which only K2 of the previous iteration wrote).
 __device__ inline unsigned long long derive_chunk(const DevCtl* ctl, unsigned long long m,
@@ -446,8 +489,12 @@ __device__ inline unsigned long long derive_chunk(const DevCtl* ctl, unsigned lo
 }
 
 // K1 for N-Queens: evaluate + compact children into the block's childbuf slab.
+// Children within `finish` levels of the bottom are counted in-thread by the
+// bitmask DFS (nq_dfs) instead of being pushed: the deepest levels dominate
+// the tree, so the pool machinery only carries the shallow part.
 __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
-                       uint32_t* blockCounts, uint32_t* blockSols, int N, int g,
+                       uint32_t* blockCounts, unsigned long long* blockSols,
+                       unsigned long long* blockExtra, int N, int g, int finish,
                        unsigned long long m, unsigned long long M) {
   __shared__ NQNode s[EMIT_TILE / 4 + 2];  // N >= 4 (engine falls back below)
   const unsigned long long c = derive_chunk(ctl, m, M);
@@ -455,7 +502,8 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
   const uint32_t total = static_cast<uint32_t>(c * N);
   const NQNode* parents = pool + (ctl->size - c);
   const uint32_t c0 = static_cast<uint32_t>(blockIdx.x) * EMIT_TILE;
-  uint32_t cnt = 0, sols = 0;
+  uint32_t cnt = 0;
+  unsigned long long sols = 0, extra = 0;
   unsigned int first = 0;
   uint8_t lab[EMIT_TILE / BLOCK] = {0, 0, 0, 0};
   uint16_t lpid[EMIT_TILE / BLOCK];
@@ -476,21 +524,45 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
         const NQNode& p = s[pid - first];
         const int depth = p.depth;
         if (depth == N) {
-          lab[j] = (k == 0) ? 2 : 0;  // leaf parent counted once (nqueens_chpl.chpl:78-80)
+          sols += (k == 0);  // leaf parent counted once (nqueens_chpl.chpl:78-80)
         } else if (k >= depth && nq_safe(p.board, depth, p.board[k], g)) {
-          lab[j] = 1;
+          const int rem = N - (depth + 1);  // levels below the child
+          if (rem <= finish) {
+            // child + its whole subtree counted here, nothing pushed
+            extra += 1;
+            if (rem == 0) {
+              sols += 1;
+            } else {
+              const uint32_t msk = (1u << N) - 1u;
+              uint32_t cols = 0, d1 = 0, d2 = 0;
+              for (int i = 0; i < depth; i++) {
+                const uint32_t b = 1u << s[pid - first].board[i];
+                cols |= b;
+                d1 = ((d1 | b) << 1) & msk;
+                d2 = (d2 | b) >> 1;
+              }
+              const uint32_t b = 1u << s[pid - first].board[k];
+              cols |= b;
+              d1 = ((d1 | b) << 1) & msk;
+              d2 = (d2 | b) >> 1;
+              nq_dfs<NQ_FINISH_MAX>(cols, d1, d2, depth + 1, N, extra, sols);
+            }
+          } else {
+            lab[j] = 1;
+          }
         }
         cnt += (lab[j] == 1);
-        sols += (lab[j] == 2);
       }
     }
   }
-  uint32_t totC, totS;
+  uint32_t totC;
   const uint32_t pre = block_excl_scan(cnt, totC);
-  block_excl_scan(sols, totS);
+  const unsigned long long totS = block_reduce_u64(sols);
+  const unsigned long long totE = block_reduce_u64(extra);
   if (threadIdx.x == 0) {
     blockCounts[blockIdx.x] = totC;
     blockSols[blockIdx.x] = totS;
+    blockExtra[blockIdx.x] = totE;
   }
   if (cnt > 0) {
     unsigned long long slot = static_cast<unsigned long long>(blockIdx.x) * EMIT_TILE + pre;
@@ -505,121 +577,10 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
 }
 
 
-// Two-level N-Queens expand: one kernel evaluates a popped parent's children
-// AND grandchildren, pushing only the grandchildren. Counts are unchanged
-// versus one level at a time: a safe child contributes tree+1 (blockExtra); a
-// depth-N child contributes tree+1 and sol+1 immediately (the reference
-// pushes it and counts sol at its pop, nqueens_chpl.chpl:78-80 — same total);
-// pushed grandchildren contribute tree+1 at the scan like any push. Halves
-// the number of sequential iterations of the (latency-bound) hot loop.
-__global__ void k_nq_x2(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
-                        uint32_t* blockCounts, uint32_t* blockSols, uint32_t* blockExtra,
-                        int N, int g, unsigned long long m, unsigned long long M) {
-  __shared__ NQNode s[EMIT_TILE / 4 + 2];  // N >= 4 (engine falls back below)
-  const unsigned long long c = derive_chunk(ctl, m, M);
-  const uint32_t total = static_cast<uint32_t>(c * N);
-  const NQNode* parents = pool + (ctl->size - c);
-  const uint32_t c0 = static_cast<uint32_t>(blockIdx.x) * EMIT_TILE;
-  uint32_t cnt = 0, sols = 0, extra = 0;
-  unsigned int first = 0;
-  uint32_t gmask[EMIT_TILE / BLOCK] = {0, 0, 0, 0};  // bit k2 = push grandchild
-  uint16_t lpid[EMIT_TILE / BLOCK];
-  uint8_t lk[EMIT_TILE / BLOCK];
-  if (c0 < total) {
-    uint32_t c1 = c0 + EMIT_TILE;
-    if (c1 > total) c1 = total;
-    first = stage_range(parents, c0, c1, N, s);
-    __syncthreads();
-#pragma unroll
-    for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
-      const uint32_t t = c0 + j * BLOCK + threadIdx.x;
-      if (t < total) {
-        const uint32_t pid = t / static_cast<uint32_t>(N);
-        const int k = static_cast<int>(t - pid * N);
-        lpid[j] = static_cast<uint16_t>(pid - first);
-        lk[j] = static_cast<uint8_t>(k);
-        const NQNode& p = s[pid - first];
-        const int depth = p.depth;
-        if (depth == N) {
-          sols += (k == 0);
-        } else if (k >= depth && nq_safe(p.board, depth, p.board[k], g)) {
-          extra += 1;  // the child itself (counted, not pushed)
-          if (depth + 1 == N) {
-            sols += 1;  // depth-N child: tree+sol immediately
-          } else {
-            // grandchildren: child board = parent board with (depth,k) swapped
-            for (int k2 = depth + 1; k2 < N; k2++) {
-              const int q = (k2 == k) ? p.board[depth] : p.board[k2];
-              // safety vs child columns [0, depth+1): i < depth are parent
-              // columns (board[i], i != k since k >= depth+1 when distinct...
-              // k == depth means identity swap), i == depth is board[k]
-              uint8_t safe = 1;
-              for (int i = 0; i < depth; i++) {
-                const int o = p.board[i];
-                for (int r = 0; r < g; r++)
-                  safe &= (o != q - (depth + 1 - i)) & (o != q + (depth + 1 - i));
-              }
-              {
-                const int o = p.board[k];  // child column at position depth
-                for (int r = 0; r < g; r++) safe &= (o != q - 1) & (o != q + 1);
-              }
-              if (safe) gmask[j] |= 1u << k2;
-            }
-            cnt += __popc(gmask[j]);
-          }
-        }
-      }
-    }
-  }
-  uint32_t totC, totS, totE;
-  const uint32_t pre = block_excl_scan(cnt, totC);
-  block_excl_scan(sols, totS);
-  block_excl_scan(extra, totE);
-  if (threadIdx.x == 0) {
-    blockCounts[blockIdx.x] = totC;
-    blockSols[blockIdx.x] = totS;
-    blockExtra[blockIdx.x] = totE;
-  }
-  if (cnt > 0) {
-    unsigned long long slot =
-        static_cast<unsigned long long>(blockIdx.x) * (EMIT_TILE * (MAX_JOBS - 1)) + pre;
-#pragma unroll
-    for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
-      uint32_t mask = gmask[j];
-      if (mask == 0) continue;
-      const NQNode& p = s[lpid[j]];
-      const int depth = p.depth;
-      const int k = lk[j];
-      while (mask) {
-        const int k2 = __ffs(mask) - 1;
-        mask &= mask - 1;
-        // grandchild = parent with swaps (depth,k) then (depth+1,k2) applied;
-        // write parent qwords, patch depth byte + the affected positions with
-        // map1(map2(pos)) values (idempotent per position)
-        const unsigned long long* sp = reinterpret_cast<const unsigned long long*>(&p);
-        unsigned long long* d = reinterpret_cast<unsigned long long*>(&childbuf[slot]);
-        d[0] = (sp[0] & ~0xFFull) | static_cast<unsigned long long>(depth + 2);
-        d[1] = sp[1];
-        d[2] = sp[2];
-        uint8_t* db = reinterpret_cast<uint8_t*>(d);
-        auto map1 = [&](int pos) { return pos == depth ? k : (pos == k ? depth : pos); };
-        auto map2 = [&](int pos) {
-          return pos == depth + 1 ? k2 : (pos == k2 ? depth + 1 : pos);
-        };
-        db[1 + depth] = p.board[map1(map2(depth))];
-        db[1 + k] = p.board[map1(map2(k))];
-        db[1 + depth + 1] = p.board[map1(map2(depth + 1))];
-        db[1 + k2] = p.board[map1(map2(k2))];
-        slot++;
-      }
-    }
-  }
-}
-
 // K1 for PFSP lb1 / lb2 (one child per thread-slot).
 template <int MM, int LB>
 __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
-                         uint32_t* blockCounts, uint32_t* blockSols, int jobs,
+                         uint32_t* blockCounts, unsigned long long* blockSols, int jobs,
                          PfspDevTables tb, unsigned long long m, unsigned long long M) {
   using LDS = typename std::conditional<LB == 2, LdsLb2<MM>, LdsLb1<MM>>::type;
   __shared__ LDS lds;
@@ -633,7 +594,8 @@ __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
   const PFSPNode* parents = pool + (ctl->size - c);
   const uint32_t c0 = static_cast<uint32_t>(blockIdx.x) * EMIT_TILE;
   const int best = ctl->best;
-  uint32_t cnt = 0, sols = 0;
+  uint32_t cnt = 0;
+  unsigned long long sols = 0;
   unsigned int first = 0;
   uint8_t lab[EMIT_TILE / BLOCK] = {0, 0, 0, 0};
   uint16_t lpid[EMIT_TILE / BLOCK];
@@ -674,9 +636,9 @@ __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
       }
     }
   }
-  uint32_t totC, totS;
+  uint32_t totC;
   const uint32_t pre = block_excl_scan(cnt, totC);
-  block_excl_scan(sols, totS);
+  const unsigned long long totS = block_reduce_u64(sols);
   if (threadIdx.x == 0) {
     blockCounts[blockIdx.x] = totC;
     blockSols[blockIdx.x] = totS;
@@ -698,7 +660,7 @@ __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
 // of storing a job-indexed local array (which would spill to scratch).
 template <int MM>
 __global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
-                              uint32_t* blockCounts, uint32_t* blockSols, int jobs,
+                              uint32_t* blockCounts, unsigned long long* blockSols, int jobs,
                               PfspDevTables tb, unsigned long long m,
                               unsigned long long M) {
   __shared__ LdsLb1<MM> lds;
@@ -713,7 +675,8 @@ __global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
   const int best = ctl->best;
 
   int front[MM], remain[MM];
-  uint32_t cnt = 0, sols = 0;
+  uint32_t cnt = 0;
+  unsigned long long sols = 0;
   int depth = 0, limit1 = 0;
   if (lp >= 0) {
     const PFSPNode& p = s[lp];
@@ -730,9 +693,9 @@ __global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
       }
     }
   }
-  uint32_t totC, totS;
+  uint32_t totC;
   const uint32_t pre = block_excl_scan(cnt, totC);
-  block_excl_scan(sols, totS);
+  const unsigned long long totS = block_reduce_u64(sols);
   if (threadIdx.x == 0) {
     blockCounts[blockIdx.x] = totC;
     blockSols[blockIdx.x] = totS;
@@ -759,17 +722,18 @@ __global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
 // kernel boundary publishes it for iteration i+1 (placement-independent).
 template <class NodeT>
 __global__ void k_gather2(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* blockCounts,
-                          const uint32_t* blockSols, const uint32_t* blockExtra,
-                          const NodeT* childbuf, NodeT* pool, int strideNodes, int G,
-                          unsigned long long m, unsigned long long M,
-                          unsigned long long capacity) {
+                          const unsigned long long* blockSols,
+                          const unsigned long long* blockExtra, const NodeT* childbuf,
+                          NodeT* pool, int strideNodes, int G, unsigned long long m,
+                          unsigned long long M, unsigned long long capacity) {
   const unsigned long long c = derive_chunk(ctl_cur, m, M);
   const unsigned long long base = ctl_cur->size - c;
   const int b = blockIdx.x;
   const bool last = (b == G - 1);
 
   // prefix over blocks [0, b) — plus full sol/extra sums for the last block
-  uint32_t my_pre = 0, my_sols = 0, my_extra = 0;
+  uint32_t my_pre = 0;
+  unsigned long long my_sols = 0, my_extra = 0;
   for (int i = threadIdx.x; i < G; i += BLOCK) {
     if (i < b) my_pre += blockCounts[i];
     if (last) {
@@ -777,10 +741,10 @@ __global__ void k_gather2(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_
       if (blockExtra) my_extra += blockExtra[i];
     }
   }
-  uint32_t pre_tot, sol_tot, extra_tot;
+  uint32_t pre_tot;
   block_excl_scan(my_pre, pre_tot);
-  block_excl_scan(my_sols, sol_tot);
-  block_excl_scan(my_extra, extra_tot);
+  const unsigned long long sol_tot = block_reduce_u64(my_sols);
+  const unsigned long long extra_tot = block_reduce_u64(my_extra);
 
   const uint32_t cnt = blockCounts[b];
   const unsigned long long off = base + pre_tot;
@@ -860,22 +824,16 @@ int devpool_grid(unsigned long long M, int per, int lbk) {
 int devpool_stride(int lbk) { return lbk == 0 ? BLOCK * MAX_JOBS : EMIT_TILE; }
 
 void launch_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
-                 uint32_t* blockCounts, uint32_t* blockSols, int N, int g,
+                 uint32_t* blockCounts, unsigned long long* blockSols,
+                 unsigned long long* blockExtra, int N, int g, int finish,
                  unsigned long long m, unsigned long long M, hipStream_t s) {
   hipLaunchKernelGGL(k_nq_x, dim3(devpool_grid(M, N, 1)), dim3(BLOCK), 0, s, ctl, pool,
-                     childbuf, blockCounts, blockSols, N, g, m, M);
-}
-
-void launch_nq_x2(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
-                  uint32_t* blockCounts, uint32_t* blockSols, uint32_t* blockExtra, int N,
-                  int g, unsigned long long m, unsigned long long M, hipStream_t s) {
-  hipLaunchKernelGGL(k_nq_x2, dim3(devpool_grid(M, N, 1)), dim3(BLOCK), 0, s, ctl, pool,
-                     childbuf, blockCounts, blockSols, blockExtra, N, g, m, M);
+                     childbuf, blockCounts, blockSols, blockExtra, N, g, finish, m, M);
 }
 
 template <int MM>
 static void launch_pfsp_x_mm(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
-                             uint32_t* bc, uint32_t* bs, int jobs, int lbk,
+                             uint32_t* bc, unsigned long long* bs, int jobs, int lbk,
                              const PfspDevTables& tb, unsigned long long m,
                              unsigned long long M, hipStream_t s) {
   if (lbk == 0) {
@@ -891,7 +849,8 @@ static void launch_pfsp_x_mm(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childb
 }
 
 void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32_t* bc,
-                   uint32_t* bs, int jobs, int machines, int lbk, const PfspDevTables& tb,
+                   unsigned long long* bs, int jobs, int machines, int lbk,
+                   const PfspDevTables& tb,
                    unsigned long long m, unsigned long long M, hipStream_t s) {
   if (machines == 5)
     launch_pfsp_x_mm<5>(ctl, pool, childbuf, bc, bs, jobs, lbk, tb, m, M, s);
@@ -902,20 +861,21 @@ void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32
 }
 
 void launch_gather2_nq(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
-                       const uint32_t* bs, const uint32_t* be, const NQNode* childbuf,
-                       NQNode* pool, int strideNodes, int G, unsigned long long m,
-                       unsigned long long M, unsigned long long capacity, hipStream_t s) {
+                       const unsigned long long* bs, const unsigned long long* be,
+                       const NQNode* childbuf, NQNode* pool, int strideNodes, int G,
+                       unsigned long long m, unsigned long long M,
+                       unsigned long long capacity, hipStream_t s) {
   hipLaunchKernelGGL(k_gather2<NQNode>, dim3(G), dim3(BLOCK), 0, s, ctl_cur, ctl_next, bc,
                      bs, be, childbuf, pool, strideNodes, G, m, M, capacity);
 }
 
 void launch_gather2_pfsp(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
-                         const uint32_t* bs, const PFSPNode* childbuf, PFSPNode* pool,
-                         int strideNodes, int G, unsigned long long m, unsigned long long M,
-                         unsigned long long capacity, hipStream_t s) {
+                         const unsigned long long* bs, const PFSPNode* childbuf,
+                         PFSPNode* pool, int strideNodes, int G, unsigned long long m,
+                         unsigned long long M, unsigned long long capacity, hipStream_t s) {
   hipLaunchKernelGGL(k_gather2<PFSPNode>, dim3(G), dim3(BLOCK), 0, s, ctl_cur, ctl_next, bc,
-                     bs, static_cast<const uint32_t*>(nullptr), childbuf, pool, strideNodes,
-                     G, m, M, capacity);
+                     bs, static_cast<const unsigned long long*>(nullptr), childbuf, pool,
+                     strideNodes, G, m, M, capacity);
 }
 
 }  // namespace gats
