@@ -506,7 +506,7 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     // depth of the in-thread bitmask subtree finisher (levels from the bottom)
     int finish = 8;
     if (const char* e = std::getenv("GATS_NQ_FINISH")) finish = atoi(e);
-    if (finish > 12) finish = 12;  // template recursion budget (NQ_FINISH_MAX)
+    if (finish > 8) finish = 8;  // template recursion budget (NQ_FINISH_MAX)
     int S = devpool_slices();
     while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;  // small search
     const int NS = (S == 1) ? 1 : S * 4;  // oversubscribe: ~4 queued slices/thread

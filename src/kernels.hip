@@ -476,7 +476,7 @@ __device__ inline void nq_dfs(uint32_t cols, uint32_t d1, uint32_t d2, int place
     }
   }
 }
-constexpr int NQ_FINISH_MAX = 12;
+constexpr int NQ_FINISH_MAX = 8;
 
 // popBackBulk chunk, re-derived identically by every kernel of an iteration
 // (pure function of ctl->size, which only K2 of the previous iteration wrote).
