@@ -19,6 +19,7 @@
 #include <algorithm>
 #include <atomic>
 #include <cstdlib>
+#include <map>
 #include <mutex>
 #include <thread>
 #include <cstring>
@@ -49,11 +50,69 @@ int gpu_device_count() {
 
 namespace {
 
+// Process-level buffer cache: hipMalloc of the multi-GB slice pools costs
+// tens of ms, which dominates repeated searches (bench steps, PFSP
+// instances) once a search itself is ~100 ms. Buffers are recycled by exact
+// (device, size); sizes are deterministic per engine config so the hit rate
+// is ~100% after the first search. Freed only at process exit.
+struct BufferCache {
+  std::mutex mu;
+  std::map<std::pair<int, size_t>, std::vector<void*>> dev, pinned;
+};
+BufferCache& buffer_cache() {
+  static BufferCache c;
+  return c;
+}
+
+void* cached_dev_alloc(size_t bytes) {
+  int dev = 0;
+  (void)hipGetDevice(&dev);
+  {
+    std::lock_guard<std::mutex> l(buffer_cache().mu);
+    auto& v = buffer_cache().dev[{dev, bytes}];
+    if (!v.empty()) {
+      void* p = v.back();
+      v.pop_back();
+      return p;
+    }
+  }
+  void* p = nullptr;
+  HIP_CHECK(hipMalloc(&p, bytes));
+  return p;
+}
+
+void cached_dev_free(void* p, size_t bytes) {
+  if (!p) return;
+  int dev = 0;
+  (void)hipGetDevice(&dev);
+  std::lock_guard<std::mutex> l(buffer_cache().mu);
+  buffer_cache().dev[{dev, bytes}].push_back(p);
+}
+
+void* cached_pinned_alloc(size_t bytes) {
+  {
+    std::lock_guard<std::mutex> l(buffer_cache().mu);
+    auto& v = buffer_cache().pinned[{0, bytes}];
+    if (!v.empty()) {
+      void* p = v.back();
+      v.pop_back();
+      return p;
+    }
+  }
+  void* p = nullptr;
+  HIP_CHECK(hipHostMalloc(&p, bytes));
+  return p;
+}
+
+void cached_pinned_free(void* p, size_t bytes) {
+  if (!p) return;
+  std::lock_guard<std::mutex> l(buffer_cache().mu);
+  buffer_cache().pinned[{0, bytes}].push_back(p);
+}
+
 template <typename T>
 T* dev_alloc(size_t n) {
-  void* p = nullptr;
-  HIP_CHECK(hipMalloc(&p, n * sizeof(T)));
-  return static_cast<T*>(p);
+  return static_cast<T*>(cached_dev_alloc(n * sizeof(T)));
 }
 
 template <typename T>
@@ -62,6 +121,8 @@ T* dev_upload(const T* src, size_t n) {
   HIP_CHECK(hipMemcpy(d, src, n * sizeof(T), hipMemcpyHostToDevice));
   return d;
 }
+
+void dev_release(void* p, size_t bytes) { cached_dev_free(p, bytes); }
 
 // Owns the int16/uint8-compressed PFSP bound tables on device. `tb` keeps the
 // reference's identity pair order (oracle-comparable); `tb_sorted` reorders
@@ -101,13 +162,13 @@ struct PfspTablesGuard {
             (static_cast<uint64_t>(job) << 48) | (lag << 32) | (ptm1 << 16) | ptm0;
       }
     }
-    tb.p_times = keep(dev_upload(p16.data(), p16.size()));
-    tb.min_tails = keep(dev_upload(mt.data(), mt.size()));
-    tb.lags = keep(dev_upload(lags16.data(), lags16.size()));
-    tb.johnson_schedules = keep(dev_upload(js8.data(), js8.size()));
-    tb.johnson_packed = keep(dev_upload(jp.data(), jp.size()));
-    tb.pairs1 = keep(dev_upload(p1.data(), p1.size()));
-    tb.pairs2 = keep(dev_upload(p2.data(), p2.size()));
+    tb.p_times = keep(dev_upload(p16.data(), p16.size()), p16.size() * sizeof(p16[0]));
+    tb.min_tails = keep(dev_upload(mt.data(), mt.size()), mt.size() * sizeof(mt[0]));
+    tb.lags = keep(dev_upload(lags16.data(), lags16.size()), lags16.size() * sizeof(lags16[0]));
+    tb.johnson_schedules = keep(dev_upload(js8.data(), js8.size()), js8.size() * sizeof(js8[0]));
+    tb.johnson_packed = keep(dev_upload(jp.data(), jp.size()), jp.size() * sizeof(jp[0]));
+    tb.pairs1 = keep(dev_upload(p1.data(), p1.size()), p1.size() * sizeof(p1[0]));
+    tb.pairs2 = keep(dev_upload(p2.data(), p2.size()), p2.size() * sizeof(p2[0]));
 
     // strength = the pair's 2-machine bound on the root relaxation (strong
     // pairs stay strong deeper in the tree); sort descending
@@ -139,17 +200,19 @@ struct PfspTablesGuard {
       p2_s[r] = p2[k];
     }
     tb_sorted = tb;
-    tb_sorted.johnson_packed = keep(dev_upload(jp_s.data(), jp_s.size()));
-    tb_sorted.pairs1 = keep(dev_upload(p1_s.data(), p1_s.size()));
-    tb_sorted.pairs2 = keep(dev_upload(p2_s.data(), p2_s.size()));
+    tb_sorted.johnson_packed = keep(dev_upload(jp_s.data(), jp_s.size()), jp_s.size() * sizeof(jp_s[0]));
+    tb_sorted.pairs1 = keep(dev_upload(p1_s.data(), p1_s.size()), p1_s.size() * sizeof(p1_s[0]));
+    tb_sorted.pairs2 = keep(dev_upload(p2_s.data(), p2_s.size()), p2_s.size() * sizeof(p2_s[0]));
   }
+  std::vector<size_t> alloc_bytes;
   template <typename T>
-  T* keep(T* p) {
+  T* keep(T* p, size_t bytes) {
     allocs.push_back(p);
+    alloc_bytes.push_back(bytes);
     return p;
   }
   ~PfspTablesGuard() {
-    for (void* p : allocs) (void)hipFree(p);
+    for (size_t i = 0; i < allocs.size(); i++) cached_dev_free(allocs[i], alloc_bytes[i]);
   }
 };
 
@@ -162,19 +225,21 @@ struct StreamGuard {
 template <typename T>
 struct DevGuard {
   T* p = nullptr;
-  explicit DevGuard(size_t n) { p = dev_alloc<T>(n); }
-  ~DevGuard() { (void)hipFree(p); }
+  size_t bytes = 0;
+  explicit DevGuard(size_t n) : bytes(n * sizeof(T)) {
+    p = static_cast<T*>(cached_dev_alloc(bytes));
+  }
+  ~DevGuard() { cached_dev_free(p, bytes); }
 };
 
 template <typename T>
 struct PinnedGuard {
   T* p = nullptr;
-  explicit PinnedGuard(size_t n) {
-    void* q = nullptr;
-    HIP_CHECK(hipHostMalloc(&q, n * sizeof(T)));
-    p = static_cast<T*>(q);
+  size_t bytes = 0;
+  explicit PinnedGuard(size_t n) : bytes(n * sizeof(T)) {
+    p = static_cast<T*>(cached_pinned_alloc(bytes));
   }
-  ~PinnedGuard() { (void)hipHostFree(p); }
+  ~PinnedGuard() { cached_pinned_free(p, bytes); }
 };
 
 int lbk_of(LbKind lb) {
