@@ -27,8 +27,8 @@ from gats_amd import dist as gdist  # noqa: E402
 def parse_args():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=3)
-    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--problem", default="nqueens", choices=["nqueens", "pfsp"])
     ap.add_argument("--N", type=int, default=17)
     ap.add_argument("--g", type=int, default=1)
