@@ -77,12 +77,10 @@ __device__ inline void emit_nq_child(NQNode* pool, unsigned long long slot,
 
 __device__ inline void emit_pfsp_child(PFSPNode* pool, unsigned long long slot,
                                        const PFSPNode& parent, int depth, int limit1,
-                                       int k, uint8_t pair_hint = 0) {
+                                       int k) {
   const unsigned long long* s = reinterpret_cast<const unsigned long long*>(&parent);
   unsigned long long* d = reinterpret_cast<unsigned long long*>(&pool[slot]);
-  // byte 0 = depth, byte 1 = limit1; prmu bytes at 2..21; byte 22 = the lb2
-  // argmax-pair hint (children start their pair loop at the parent's
-  // strongest pair — rotation is prune-decision invariant)
+  // byte 0 = depth, byte 1 = limit1; prmu bytes at 2..21
   d[0] = (s[0] & ~0xFFFFull) |
          static_cast<unsigned long long>(static_cast<uint8_t>(depth + 1)) |
          (static_cast<unsigned long long>(static_cast<uint8_t>(limit1 + 1)) << 8);
@@ -91,7 +89,6 @@ __device__ inline void emit_pfsp_child(PFSPNode* pool, unsigned long long slot,
   uint8_t* db = reinterpret_cast<uint8_t*>(d);
   db[2 + depth] = parent.prmu[k];
   db[2 + k] = parent.prmu[depth];
-  db[22] = pair_hint;
 }
 
 // ---------------------------------------------------------------------------
@@ -262,8 +259,7 @@ __device__ inline int lb1d_child_bound(const LdsLb1<MM>& lds, const int* front,
 // this thread's LDS slice (runtime-indexed by pair machine ids).
 template <int MM>
 __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu, int depth,
-                                      int k, int jobs, int best, int* front, int start = 0,
-                                      uint8_t* argmax_out = nullptr) {
+                                      int k, int jobs, int best, int* front) {
   const int job_k = prmu[k];
   child_front<MM>(lds, prmu, depth, job_k, jobs, front, 1);
 
@@ -273,22 +269,17 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
 
   constexpr int PAIRS = LdsLb2<MM>::PAIRS;
   int lb = 0;
-  int argmax = start;
   // two pairs in flight per step: each pair's (tmp0, tmp1) update chain is a
   // serial LDS-load -> add -> max dependency; interleaving two independent
   // chains doubles the ILP available to hide the ds_read_b64 latency
-  for (int lr = 0; lr < PAIRS; lr += 2) {
-    const bool two = (lr + 1) < PAIRS;
-    int la = lr + start;
-    if (la >= PAIRS) la -= PAIRS;
-    int lb_i = (two ? lr + 1 : lr) + start;
-    if (lb_i >= PAIRS) lb_i -= PAIRS;
-    const int ma0a = lds.pair1[la], ma1a = lds.pair2[la];
-    const int ma0b = lds.pair1[lb_i], ma1b = lds.pair2[lb_i];
+  for (int l = 0; l < PAIRS; l += 2) {
+    const bool two = (l + 1) < PAIRS;
+    const int ma0a = lds.pair1[l], ma1a = lds.pair2[l];
+    const int ma0b = lds.pair1[two ? l + 1 : l], ma1b = lds.pair2[two ? l + 1 : l];
     int t0a = front[ma0a], t1a = front[ma1a];
     int t0b = front[ma0b], t1b = front[ma1b];
-    const uint64_t* jpa = &lds.jp[la * jobs];
-    const uint64_t* jpb = &lds.jp[lb_i * jobs];
+    const uint64_t* jpa = &lds.jp[l * jobs];
+    const uint64_t* jpb = &lds.jp[(two ? l + 1 : l) * jobs];
     for (int j = 0; j < jobs; j++) {
       const uint64_t va = jpa[j];  // one ds_read_b64 replaces 4 scalar LDS reads
       const uint64_t vb = jpb[j];
@@ -309,22 +300,13 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
     // the reference's per-pair early exit (c_bound_johnson.c:231-233), which
     // the hostpool-vs-CPU-oracle tests assert; pair b's work is wasted only
     // on the exit iteration
-    const int va = max(t1a + lds.min_tails[ma1a], t0a + lds.min_tails[ma0a]);
-    if (va > lb) {
-      lb = va;
-      argmax = la;
-    }
+    lb = max(lb, max(t1a + lds.min_tails[ma1a], t0a + lds.min_tails[ma0a]));
     if (lb > best) break;
     if (two) {
-      const int vb = max(t1b + lds.min_tails[ma1b], t0b + lds.min_tails[ma0b]);
-      if (vb > lb) {
-        lb = vb;
-        argmax = lb_i;
-      }
+      lb = max(lb, max(t1b + lds.min_tails[ma1b], t0b + lds.min_tails[ma0b]));
       if (lb > best) break;
     }
   }
-  if (argmax_out) *argmax_out = static_cast<uint8_t>(argmax);
   return lb;
 }
 
@@ -616,7 +598,6 @@ __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
   unsigned long long sols = 0;
   unsigned int first = 0;
   uint8_t lab[EMIT_TILE / BLOCK] = {0, 0, 0, 0};
-  uint8_t hint[EMIT_TILE / BLOCK] = {0, 0, 0, 0};
   uint16_t lpid[EMIT_TILE / BLOCK];
   uint8_t lk[EMIT_TILE / BLOCK];
   if (c0 < total) {
@@ -639,9 +620,7 @@ __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
           int lb;
           if constexpr (LB == 2) {
             int* front = &lds.front[threadIdx.x * (MM + 1)];
-            const int start = p.pad_[0] < LdsLb2<MM>::PAIRS ? p.pad_[0] : 0;
-            lb = lb2_child_bound<MM>(lds, p.prmu, depth, k, jobs, best, front, start,
-                                     &hint[j]);
+            lb = lb2_child_bound<MM>(lds, p.prmu, depth, k, jobs, best, front);
           } else {
             lb = lb1_child_bound<MM>(lds, p.prmu, depth, k, jobs);
           }
@@ -670,8 +649,7 @@ __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
     for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
       if (lab[j] == 1) {
         const PFSPNode& p = s[lpid[j]];
-        emit_pfsp_child(childbuf, slot++, p, p.depth, p.limit1, lk[j],
-                        LB == 2 ? hint[j] : 0);
+        emit_pfsp_child(childbuf, slot++, p, p.depth, p.limit1, lk[j]);
       }
     }
   }
