@@ -132,16 +132,31 @@ struct LdsLb1 {
   int32_t min_tails[MM];
 };
 
+// LB2_FULL's machine-pair list is the deterministic lexicographic (i < j)
+// enumeration (c_bound_johnson.c:57-70): with the pair loop fully unrolled
+// these fold to compile-time constants, so the pair-indexed `front` reads
+// become register accesses (the earlier design kept a 21.5 KB per-block LDS
+// scratch for runtime-indexed front, capping occupancy at 3 blocks/CU).
+template <int MM>
+__host__ __device__ constexpr int pair_first(int l) {
+  int i = 0, rem = l;
+  while (rem >= MM - 1 - i) {
+    rem -= MM - 1 - i;
+    i++;
+  }
+  return i;
+}
+template <int MM>
+__host__ __device__ constexpr int pair_second(int l) {
+  return pair_first<MM>(l) + 1 + (l - (pair_first<MM>(l) * (2 * MM - pair_first<MM>(l) - 1)) / 2);
+}
+
 template <int MM>
 struct LdsLb2 {
   static constexpr int PAIRS = MM * (MM - 1) / 2;
   int16_t p[MM * MAX_JOBS];
   int32_t min_tails[MM];
   uint64_t jp[PAIRS * MAX_JOBS];  // packed johnson: job<<48|lag<<32|ptm1<<16|ptm0
-  uint8_t pair1[PAIRS], pair2[PAIRS];
-  // per-thread `front` scratch, runtime-indexed by machine-pair ids; padded
-  // stride MM+1 keeps the 32-bank groups conflict-free (stride odd vs 32)
-  int front[BLOCK * (MM + 1)];
 };
 
 template <int MM, class LDS>
@@ -156,10 +171,6 @@ __device__ inline void stage_lb2_tables(LdsLb2<MM>& lds, const PfspDevTables& tb
   constexpr int PAIRS = LdsLb2<MM>::PAIRS;
   for (int i = threadIdx.x; i < PAIRS * jobs; i += blockDim.x)
     lds.jp[i] = tb.johnson_packed[i];
-  if (threadIdx.x < PAIRS) {
-    lds.pair1[threadIdx.x] = tb.pairs1[threadIdx.x];
-    lds.pair2[threadIdx.x] = tb.pairs2[threadIdx.x];
-  }
 }
 
 // front <- completion times of the child prefix (parent prefix + job k placed
@@ -259,7 +270,8 @@ __device__ inline int lb1d_child_bound(const LdsLb1<MM>& lds, const int* front,
 // this thread's LDS slice (runtime-indexed by pair machine ids).
 template <int MM>
 __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu, int depth,
-                                      int k, int jobs, int best, int* front) {
+                                      int k, int jobs, int best) {
+  int front[MM];
   const int job_k = prmu[k];
   child_front<MM>(lds, prmu, depth, job_k, jobs, front, 1);
 
@@ -269,13 +281,14 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
 
   constexpr int PAIRS = LdsLb2<MM>::PAIRS;
   int lb = 0;
-  // two pairs in flight per step: each pair's (tmp0, tmp1) update chain is a
-  // serial LDS-load -> add -> max dependency; interleaving two independent
-  // chains doubles the ILP available to hide the ds_read_b64 latency
+  // fully unrolled over the compile-time pair list; two pairs in flight per
+  // step (each pair's update chain is serial — interleaving two independent
+  // chains hides the ds_read_b64 latency)
+#pragma unroll
   for (int l = 0; l < PAIRS; l += 2) {
     const bool two = (l + 1) < PAIRS;
-    const int ma0a = lds.pair1[l], ma1a = lds.pair2[l];
-    const int ma0b = lds.pair1[two ? l + 1 : l], ma1b = lds.pair2[two ? l + 1 : l];
+    const int ma0a = pair_first<MM>(l), ma1a = pair_second<MM>(l);
+    const int ma0b = pair_first<MM>(two ? l + 1 : l), ma1b = pair_second<MM>(two ? l + 1 : l);
     int t0a = front[ma0a], t1a = front[ma1a];
     int t0b = front[ma0b], t1b = front[ma1b];
     const uint64_t* jpa = &lds.jp[l * jobs];
@@ -298,8 +311,7 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
     }
     // merge pair a, check, then pair b: keeps the returned value bit-equal to
     // the reference's per-pair early exit (c_bound_johnson.c:231-233), which
-    // the hostpool-vs-CPU-oracle tests assert; pair b's work is wasted only
-    // on the exit iteration
+    // the hostpool-vs-CPU-oracle tests assert
     lb = max(lb, max(t1a + lds.min_tails[ma1a], t0a + lds.min_tails[ma0a]));
     if (lb > best) break;
     if (two) {
@@ -366,9 +378,8 @@ __global__ void k_pfsp_eval_lb2(const PFSPNode* parents, int n, int jobs, PfspDe
   if (lp < 0) return;
   const int k = static_cast<int>(t % static_cast<unsigned long long>(jobs));
   const PFSPNode& parent = snodes[lp];
-  int* front = &lds.front[threadIdx.x * (MM + 1)];
   if (k >= parent.limit1 + 1)
-    bounds[t] = lb2_child_bound<MM>(lds, parent.prmu, parent.depth, k, jobs, best, front);
+    bounds[t] = lb2_child_bound<MM>(lds, parent.prmu, parent.depth, k, jobs, best);
 }
 
 
@@ -619,8 +630,7 @@ __global__ void k_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
         if (k >= p.limit1 + 1) {
           int lb;
           if constexpr (LB == 2) {
-            int* front = &lds.front[threadIdx.x * (MM + 1)];
-            lb = lb2_child_bound<MM>(lds, p.prmu, depth, k, jobs, best, front);
+            lb = lb2_child_bound<MM>(lds, p.prmu, depth, k, jobs, best);
           } else {
             lb = lb1_child_bound<MM>(lds, p.prmu, depth, k, jobs);
           }
