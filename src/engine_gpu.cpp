@@ -124,12 +124,10 @@ T* dev_upload(const T* src, size_t n) {
 
 void dev_release(void* p, size_t bytes) { cached_dev_free(p, bytes); }
 
-// Owns the int16/uint8-compressed PFSP bound tables on device. `tb` keeps the
-// reference's identity pair order (oracle-comparable); `tb_sorted` reorders
-// machine pairs strongest-first for the devpool expand path (see gpu_api.hpp).
+// Owns the int16/uint8-compressed PFSP bound tables on device (identity pair
+// order — the kernels' pair map is the compile-time lexicographic one).
 struct PfspTablesGuard {
   PfspDevTables tb{};
-  PfspDevTables tb_sorted{};
   std::vector<void*> allocs;
 
   PfspTablesGuard(const PfspInstance& I) {
@@ -170,39 +168,6 @@ struct PfspTablesGuard {
     tb.pairs1 = keep(dev_upload(p1.data(), p1.size()), p1.size() * sizeof(p1[0]));
     tb.pairs2 = keep(dev_upload(p2.data(), p2.size()), p2.size() * sizeof(p2[0]));
 
-    // strength = the pair's 2-machine bound on the root relaxation (strong
-    // pairs stay strong deeper in the tree); sort descending
-    std::vector<std::pair<int, int>> strength(pairs);
-    for (int k = 0; k < pairs; k++) {
-      const int ma0 = I.lb2.pairs1[k], ma1 = I.lb2.pairs2[k];
-      int tmp0 = I.lb1.min_heads[ma0], tmp1 = I.lb1.min_heads[ma1];
-      for (int j = 0; j < n; j++) {
-        const int job = I.lb2.johnson_schedules[static_cast<size_t>(k) * n + j];
-        tmp0 += I.lb1.p_times[ma0 * n + job];
-        const int t = tmp0 + I.lb2.lags[static_cast<size_t>(k) * n + job];
-        if (t > tmp1) tmp1 = t;
-        tmp1 += I.lb1.p_times[ma1 * n + job];
-      }
-      const int a = tmp1 + I.lb1.min_tails[ma1];
-      const int b = tmp0 + I.lb1.min_tails[ma0];
-      strength[k] = {a > b ? a : b, k};
-    }
-    std::sort(strength.begin(), strength.end(),
-              [](const auto& x, const auto& y) { return x.first > y.first; });
-    std::vector<uint64_t> jp_s(jp.size());
-    std::vector<uint8_t> p1_s(pairs), p2_s(pairs);
-    for (int r = 0; r < pairs; r++) {
-      const int k = strength[r].second;
-      std::copy(jp.begin() + static_cast<size_t>(k) * n,
-                jp.begin() + static_cast<size_t>(k + 1) * n,
-                jp_s.begin() + static_cast<size_t>(r) * n);
-      p1_s[r] = p1[k];
-      p2_s[r] = p2[k];
-    }
-    tb_sorted = tb;
-    tb_sorted.johnson_packed = keep(dev_upload(jp_s.data(), jp_s.size()), jp_s.size() * sizeof(jp_s[0]));
-    tb_sorted.pairs1 = keep(dev_upload(p1_s.data(), p1_s.size()), p1_s.size() * sizeof(p1_s[0]));
-    tb_sorted.pairs2 = keep(dev_upload(p2_s.data(), p2_s.size()), p2_s.size() * sizeof(p2_s[0]));
   }
   std::vector<size_t> alloc_bytes;
   template <typename T>
@@ -719,7 +684,7 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     for (int t = 0; t < S; t++) {
       threads.emplace_back([&, t] {
         try {
-          outs[t] = devpool_thread_pfsp(slices, next_slice, I, tables.tb_sorted, lbk, best,
+          outs[t] = devpool_thread_pfsp(slices, next_slice, I, tables.tb, lbk, best,
                                         m, M, device, capacity, sb, S == 1, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();

@@ -38,12 +38,6 @@ struct PfspDevTables {
   const uint8_t* pairs2;             // [pairs]
 };
 
-// Same tables with machine pairs reordered strongest-first (devpool expand
-// only): the push/prune decision `lb < best` and the leaf best-update (only
-// taken on full, non-early-exited evaluations == the true max) are both
-// invariant to pair order, so the early exit fires pairs earlier. The
-// hostpool eval kernels keep the reference's identity order so their bound
-// VALUES stay bit-identical to the CPU oracle (tests/test_gpu_kernels.py).
 
 
 // Launchers implemented in kernels.hip.
