@@ -722,6 +722,158 @@ __global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
 }
 
 
+
+// Wave-cooperative lb2 expand: phase A builds each child's front schedule and
+// scheduled-mask into LDS (one thread per child, fully parallel); phase B
+// assigns each 64-lane wave ONE child at a time — lanes evaluate machine
+// pairs in parallel and a shfl max-reduce after each 64-pair round makes the
+// early exit COLLECTIVE. The per-lane kernel's exit was wave-granular (a wave
+// ran all 190 pairs whenever any lane's child was pushed, i.e. essentially
+// always); here the serial chain per child shrinks from 190x20 to ceil(190/64)
+// rounds of 20 steps, and pruned children really do stop early.
+// Decision parity: an exit implies partial-max > best => prune (identical to
+// the full max's decision); a completed reduce IS the exact full max, so leaf
+// best-updates only use exact values (reference semantics,
+// c_bound_johnson.c:211-254).
+template <int MM>
+struct LdsLb2w {
+  static constexpr int PAIRS = MM * (MM - 1) / 2;
+  int16_t p[MM * MAX_JOBS];
+  int32_t min_tails[MM];
+  uint64_t jp[PAIRS * MAX_JOBS];
+  uint8_t pair1[PAIRS], pair2[PAIRS];
+  uint16_t fronts[BLOCK][MM + 1];  // child completion times (values <= 20*20*99)
+  uint32_t smask[BLOCK];           // scheduled-job bitmask per child
+  uint8_t lpar[BLOCK];             // child's parent index in snodes
+  int8_t ck[BLOCK];                // child's k, or -1 invalid, <=-2 leaf (-2-k)
+  PFSPNode snodes[BLOCK / 5 + 2];
+};
+
+__device__ inline int wave_max_i32(int v) {
+#pragma unroll
+  for (int d = 32; d >= 1; d >>= 1) v = max(v, __shfl_xor(v, d));
+  return v;
+}
+
+template <int MM>
+__global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
+                              uint32_t* waveCounts, unsigned long long* waveSols, int jobs,
+                              PfspDevTables tb, unsigned long long m,
+                              unsigned long long M) {
+  __shared__ LdsLb2w<MM> lds;
+  constexpr int PAIRS = LdsLb2w<MM>::PAIRS;
+  for (int i = threadIdx.x; i < MM * jobs; i += blockDim.x) lds.p[i] = tb.p_times[i];
+  if (threadIdx.x < MM) lds.min_tails[threadIdx.x] = tb.min_tails[threadIdx.x];
+  for (int i = threadIdx.x; i < PAIRS * jobs; i += blockDim.x)
+    lds.jp[i] = tb.johnson_packed[i];
+  if (threadIdx.x < PAIRS) {
+    lds.pair1[threadIdx.x] = static_cast<uint8_t>(pair_first<MM>(threadIdx.x));
+    lds.pair2[threadIdx.x] = static_cast<uint8_t>(pair_second<MM>(threadIdx.x));
+  }
+
+  const unsigned long long c = derive_chunk(ctl, m, M);
+  const uint32_t total = static_cast<uint32_t>(c * jobs);
+  const PFSPNode* parents = pool + (ctl->size - c);
+  const uint32_t c0 = static_cast<uint32_t>(blockIdx.x) * BLOCK;
+  unsigned int first = 0;
+  if (c0 < total) {
+    uint32_t c1 = c0 + BLOCK;
+    if (c1 > total) c1 = total;
+    first = stage_range(parents, c0, c1, jobs, lds.snodes);
+  }
+  __syncthreads();
+
+  // ---- phase A: one thread per child slot ----
+  {
+    const uint32_t t = c0 + threadIdx.x;
+    int8_t state = -1;
+    if (c0 < total && t < total) {
+      const uint32_t pid = t / static_cast<uint32_t>(jobs);
+      const int k = static_cast<int>(t - pid * jobs);
+      const PFSPNode& p = lds.snodes[pid - first];
+      lds.lpar[threadIdx.x] = static_cast<uint8_t>(pid - first);
+      if (k >= p.limit1 + 1) {
+        int front[MM];
+        child_front<MM>(lds, p.prmu, p.depth, p.prmu[k], jobs, front, 1);
+        uint32_t sched = 0;
+        for (int i = 0; i < p.depth; i++) sched |= 1u << p.prmu[i];
+        sched |= 1u << p.prmu[k];
+        lds.smask[threadIdx.x] = sched;
+#pragma unroll
+        for (int i = 0; i < MM; i++)
+          lds.fronts[threadIdx.x][i] = static_cast<uint16_t>(front[i]);
+        state = (p.depth + 1 == jobs) ? static_cast<int8_t>(-2 - k)
+                                      : static_cast<int8_t>(k);
+      }
+    }
+    lds.ck[threadIdx.x] = state;
+  }
+  __syncthreads();
+
+  // ---- phase B: one child per wave at a time, pairs across lanes ----
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int best = ctl->best;
+  uint32_t mycnt = 0;
+  unsigned long long mysols = 0;
+  const unsigned long long wave_slab =
+      (static_cast<unsigned long long>(blockIdx.x) * 4 + wid) * 64ull;
+
+  for (int cc = 0; cc < 64; cc++) {
+    const int ct = wid * 64 + cc;
+    const int8_t state = lds.ck[ct];  // same address across the wave: broadcast
+    if (state == -1) continue;
+    const bool leaf = state <= -2;
+    const int k = leaf ? (-2 - state) : state;
+    const uint16_t* fr = lds.fronts[ct];
+    const uint32_t sched = lds.smask[ct];
+
+    int mylb = 0;
+    bool exited = false;
+    constexpr int ROUNDS = (PAIRS + 63) / 64;
+#pragma unroll
+    for (int r = 0; r < ROUNDS; r++) {
+      if (!exited) {
+        const int pr = lane + r * 64;
+        if (pr < PAIRS) {
+          const int ma0 = lds.pair1[pr];
+          const int ma1 = lds.pair2[pr];
+          int t0 = fr[ma0];
+          int t1 = fr[ma1];
+          const uint64_t* jp = &lds.jp[pr * jobs];
+          for (int j = 0; j < jobs; j++) {
+            const uint64_t v = jp[j];
+            const int job = static_cast<int>(v >> 48);
+            if (!(sched >> job & 1u)) {
+              t0 += static_cast<int>(v & 0xffff);
+              t1 = max(t1, t0 + static_cast<int>((v >> 32) & 0xffff));
+              t1 += static_cast<int>((v >> 16) & 0xffff);
+            }
+          }
+          mylb = max(mylb, max(t1 + lds.min_tails[ma1], t0 + lds.min_tails[ma0]));
+        }
+        if (wave_max_i32(mylb) > best) exited = true;  // collective early exit
+      }
+    }
+    const int lb = wave_max_i32(mylb);
+    if (lane == 0) {
+      if (leaf) {
+        mysols++;
+        if (!exited && lb < best) atomicMin(&ctl->best, lb);
+      } else if (!exited && lb < best) {
+        const PFSPNode& p = lds.snodes[lds.lpar[ct]];
+        emit_pfsp_child(childbuf, wave_slab + mycnt, p, p.depth, p.limit1, k);
+        mycnt++;
+      }
+    }
+  }
+  if (lane == 0) {
+    const unsigned long long gw = static_cast<unsigned long long>(blockIdx.x) * 4 + wid;
+    waveCounts[gw] = mycnt;
+    waveSols[gw] = mysols;
+  }
+}
+
 // K2+K3 merged ("gather2"): every block derives its own pool offset by
 // summing the counts of the blocks before it (G ~ 1000 u32 loads through L2,
 // done in parallel across blocks — cheaper than serializing on the
@@ -828,10 +980,17 @@ void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, in
 int devpool_grid(unsigned long long M, int per, int lbk) {
   if (lbk == 0)  // lb1_d: one thread per parent
     return static_cast<int>((M + BLOCK - 1) / BLOCK);
+  if (lbk == 2)  // wave-cooperative lb2: counts/slabs are per WAVE (64 slots),
+                 // padded to whole 4-wave blocks (gw index = blockIdx*4 + wid)
+    return static_cast<int>((M * per + BLOCK - 1) / BLOCK) * 4;
   return static_cast<int>((M * per + EMIT_TILE - 1) / EMIT_TILE);
 }
 
-int devpool_stride(int lbk) { return lbk == 0 ? BLOCK * MAX_JOBS : EMIT_TILE; }
+int devpool_stride(int lbk) {
+  if (lbk == 0) return BLOCK * MAX_JOBS;
+  if (lbk == 2) return 64;  // one wave's child slab
+  return EMIT_TILE;
+}
 
 void launch_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                  uint32_t* blockCounts, unsigned long long* blockSols,
@@ -853,8 +1012,10 @@ static void launch_pfsp_x_mm(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childb
     hipLaunchKernelGGL((k_pfsp_x<MM, 1>), dim3(devpool_grid(M, jobs, 1)), dim3(BLOCK), 0, s,
                        ctl, pool, childbuf, bc, bs, jobs, tb, m, M);
   } else {
-    hipLaunchKernelGGL((k_pfsp_x<MM, 2>), dim3(devpool_grid(M, jobs, 2)), dim3(BLOCK), 0, s,
-                       ctl, pool, childbuf, bc, bs, jobs, tb, m, M);
+    // 256 child slots per block (4 waves x 64); count arrays are per wave
+    const int blocks = static_cast<int>((M * jobs + BLOCK - 1) / BLOCK);
+    hipLaunchKernelGGL((k_pfsp_x_lb2w<MM>), dim3(blocks), dim3(BLOCK), 0, s, ctl, pool,
+                       childbuf, bc, bs, jobs, tb, m, M);
   }
 }
 
