@@ -420,7 +420,7 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
 
 static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& slices,
                                     std::atomic<int>& next_slice, const PfspInstance& I,
-                                    const PfspDevTables& tb_sorted, int lbk, int best0,
+                                    const PfspDevTables& tb, int lbk, int best0,
                                     int m, int M, int device, unsigned long long capacity,
                                     std::atomic<int>* shared_best, bool allow_graph,
                                     std::vector<PFSPNode>& leftover) {
@@ -437,14 +437,19 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G);
   DevGuard<unsigned long long> bs_d(G);
+  // per-wave counts (lb2) make G large enough that gather's prefix walk needs
+  // group sums; for lb1/lb1_d the walk is short and the extra kernel is skipped
+  const bool presum = (lbk == 2);
+  DevGuard<uint32_t> gsum_d(presum ? (G + 255) / 256 : 1);
 
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
     launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk,
-                  tb_sorted, m, M, stream.s);
-    launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, childbuf_d.p, pool_d.p, stride, G, m, M,
-                        capacity, stream.s);
+                  tb, m, M, stream.s);
+    if (presum) launch_presum(bc_d.p, gsum_d.p, G, stream.s);
+    launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, presum ? gsum_d.p : nullptr,
+                        childbuf_d.p, pool_d.p, stride, G, m, M, capacity, stream.s);
   };
 
   int si;

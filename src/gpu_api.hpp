@@ -64,8 +64,10 @@ void launch_gather2_nq(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* 
                        const NQNode* childbuf, NQNode* pool, int strideNodes, int G,
                        unsigned long long m, unsigned long long M,
                        unsigned long long capacity, hipStream_t s);
+void launch_presum(const uint32_t* bc, uint32_t* groupSums, int G, hipStream_t s);
 void launch_gather2_pfsp(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
-                         const unsigned long long* bs, const PFSPNode* childbuf,
-                         PFSPNode* pool, int strideNodes, int G, unsigned long long m,
-                         unsigned long long M, unsigned long long capacity, hipStream_t s);
+                         const unsigned long long* bs, const uint32_t* groupSums,
+                         const PFSPNode* childbuf, PFSPNode* pool, int strideNodes, int G,
+                         unsigned long long m, unsigned long long M,
+                         unsigned long long capacity, hipStream_t s);
 }  // namespace gats

@@ -874,6 +874,19 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
   }
 }
 
+
+// Per-256-entry group sums of the wave counts: with per-WAVE count granularity
+// (lb2's G ~ 15k at M = 50000) the gather blocks' linear prefix walk costs
+// more than the expand itself; group sums cut each block's walk from G loads
+// to G/256 + 255.
+__global__ void k_presum(const uint32_t* blockCounts, uint32_t* groupSums, int G) {
+  const int i = blockIdx.x * BLOCK + threadIdx.x;
+  const uint32_t v = (i < G) ? blockCounts[i] : 0;
+  uint32_t tot;
+  block_excl_scan(v, tot);
+  if (threadIdx.x == 0) groupSums[blockIdx.x] = tot;
+}
+
 // K2+K3 merged ("gather2"): every block derives its own pool offset by
 // summing the counts of the blocks before it (G ~ 1000 u32 loads through L2,
 // done in parallel across blocks — cheaper than serializing on the
@@ -885,22 +898,36 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
 template <class NodeT>
 __global__ void k_gather2(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* blockCounts,
                           const unsigned long long* blockSols,
-                          const unsigned long long* blockExtra, const NodeT* childbuf,
-                          NodeT* pool, int strideNodes, int G, unsigned long long m,
-                          unsigned long long M, unsigned long long capacity) {
+                          const unsigned long long* blockExtra, const uint32_t* groupSums,
+                          const NodeT* childbuf, NodeT* pool, int strideNodes, int G,
+                          unsigned long long m, unsigned long long M,
+                          unsigned long long capacity) {
   const unsigned long long c = derive_chunk(ctl_cur, m, M);
   const unsigned long long base = ctl_cur->size - c;
   const int b = blockIdx.x;
   const bool last = (b == G - 1);
 
-  // prefix over blocks [0, b) — plus full sol/extra sums for the last block
+  // prefix over count entries [0, b) — via whole-group sums when available —
+  // plus full sol/extra sums for the last block
   uint32_t my_pre = 0;
   unsigned long long my_sols = 0, my_extra = 0;
-  for (int i = threadIdx.x; i < G; i += BLOCK) {
-    if (i < b) my_pre += blockCounts[i];
+  if (groupSums) {
+    const int gfull = b / BLOCK;  // whole groups strictly before b
+    for (int i = threadIdx.x; i < gfull; i += BLOCK) my_pre += groupSums[i];
+    for (int i = gfull * BLOCK + threadIdx.x; i < b; i += BLOCK) my_pre += blockCounts[i];
     if (last) {
-      my_sols += blockSols[i];
-      if (blockExtra) my_extra += blockExtra[i];
+      for (int i = threadIdx.x; i < G; i += BLOCK) {
+        my_sols += blockSols[i];
+        if (blockExtra) my_extra += blockExtra[i];
+      }
+    }
+  } else {
+    for (int i = threadIdx.x; i < G; i += BLOCK) {
+      if (i < b) my_pre += blockCounts[i];
+      if (last) {
+        my_sols += blockSols[i];
+        if (blockExtra) my_extra += blockExtra[i];
+      }
     }
   }
   uint32_t pre_tot;
@@ -1031,22 +1058,29 @@ void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32
     launch_pfsp_x_mm<20>(ctl, pool, childbuf, bc, bs, jobs, lbk, tb, m, M, s);
 }
 
+void launch_presum(const uint32_t* bc, uint32_t* groupSums, int G, hipStream_t s) {
+  hipLaunchKernelGGL(k_presum, dim3((G + BLOCK - 1) / BLOCK), dim3(BLOCK), 0, s, bc,
+                     groupSums, G);
+}
+
 void launch_gather2_nq(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
                        const unsigned long long* bs, const unsigned long long* be,
                        const NQNode* childbuf, NQNode* pool, int strideNodes, int G,
                        unsigned long long m, unsigned long long M,
                        unsigned long long capacity, hipStream_t s) {
   hipLaunchKernelGGL(k_gather2<NQNode>, dim3(G), dim3(BLOCK), 0, s, ctl_cur, ctl_next, bc,
-                     bs, be, childbuf, pool, strideNodes, G, m, M, capacity);
+                     bs, be, static_cast<const uint32_t*>(nullptr), childbuf, pool,
+                     strideNodes, G, m, M, capacity);
 }
 
 void launch_gather2_pfsp(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
-                         const unsigned long long* bs, const PFSPNode* childbuf,
-                         PFSPNode* pool, int strideNodes, int G, unsigned long long m,
-                         unsigned long long M, unsigned long long capacity, hipStream_t s) {
+                         const unsigned long long* bs, const uint32_t* groupSums,
+                         const PFSPNode* childbuf, PFSPNode* pool, int strideNodes, int G,
+                         unsigned long long m, unsigned long long M,
+                         unsigned long long capacity, hipStream_t s) {
   hipLaunchKernelGGL(k_gather2<PFSPNode>, dim3(G), dim3(BLOCK), 0, s, ctl_cur, ctl_next, bc,
-                     bs, static_cast<const unsigned long long*>(nullptr), childbuf, pool,
-                     strideNodes, G, m, M, capacity);
+                     bs, static_cast<const unsigned long long*>(nullptr), groupSums,
+                     childbuf, pool, strideNodes, G, m, M, capacity);
 }
 
 }  // namespace gats
