@@ -122,7 +122,6 @@ T* dev_upload(const T* src, size_t n) {
   return d;
 }
 
-void dev_release(void* p, size_t bytes) { cached_dev_free(p, bytes); }
 
 // Owns the int16/uint8-compressed PFSP bound tables on device (identity pair
 // order — the kernels' pair map is the compile-time lexicographic one).
