@@ -52,7 +52,14 @@ void nq_decompose(const NQNode& parent, int N, int g, uint64_t& tree, uint64_t& 
   }
 }
 
+static void nq_check(int N, int g) {
+  if (N < 1 || N > MAX_JOBS)
+    throw std::invalid_argument("N must be in 1..20 (MAX_QUEENS parity, NQueens_node.chpl:7)");
+  if (g < 1) throw std::invalid_argument("g must be >= 1");
+}
+
 Result nqueens_seq(int N, int g) {
+  nq_check(N, g);
   Result r;
   Pool<NQNode> pool;
   pool.pushBack(nq_root());
@@ -66,6 +73,7 @@ Result nqueens_seq(int N, int g) {
 
 void nq_bfs_until(int N, int g, size_t target, Pool<NQNode>& pool, uint64_t& tree,
                   uint64_t& sol) {
+  nq_check(N, g);
   NQNode parent;
   while (pool.size() < target) {
     if (!pool.popFront(parent)) break;

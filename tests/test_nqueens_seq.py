@@ -24,3 +24,14 @@ def test_g_multiplier_does_not_change_counts(core):
     r1 = core.nqueens_seq(9, 1)
     r3 = core.nqueens_seq(9, 3)
     assert r1["tree"] == r3["tree"] and r1["sol"] == r3["sol"]
+
+
+def test_n_bounds_validated(core):
+    import pytest as _pytest
+
+    with _pytest.raises(Exception):
+        core.nqueens_seq(21, 1)
+    with _pytest.raises(Exception):
+        core.nqueens_seq(0, 1)
+    with _pytest.raises(Exception):
+        core.nqueens_seq(10, 0)
