@@ -50,6 +50,9 @@ def main():
     local = int(os.environ.get("LOCAL_RANK", "0")) % ndev
     torch.cuda.set_device(local)  # before NCCL/RCCL process-group init
     rank, world = gdist.init_dist()
+    if world != args.gpus and rank == 0:
+        print(f"# note: --gpus {args.gpus} but world_size {world}; using world_size",
+              file=sys.stderr)
 
     def step():
         if args.problem == "nqueens":

@@ -111,3 +111,11 @@ def test_nqueens_gpu_mM_window(gpu):
     assert r["sol"] == seq["sol"]
     r2 = gpu.nqueens_gpu(13, 1, 1000, 2000, 0, "hostpool", 1 << 24)
     assert r2["tree"] == seq["tree"] and r2["sol"] == seq["sol"]
+
+
+def test_from_pool_empty_frontier(gpu):
+    # world > frontier-size in the dist tier hands some ranks an empty slice
+    r = gpu.nqueens_gpu_from_pool(b"", 17, 1, 25, 50000, 0, "devpool", 1 << 22)
+    assert r["tree"] == 0 and r["sol"] == 0
+    p = gpu.pfsp_gpu_from_pool(b"", 14, "lb1", 1, 0, 25, 50000, 0, "devpool", 1 << 22)
+    assert p["tree"] == 0 and p["sol"] == 0
