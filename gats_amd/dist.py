@@ -42,6 +42,71 @@ def init_dist():
     return td.get_rank(), td.get_world_size()
 
 
+def _shared_counter():
+    """Atomic cross-rank counter backed by the process group's store (used as
+    a dynamic slice queue). Returns a callable yielding the next index, or
+    None when unavailable (fall back to static partitioning)."""
+    try:
+        from torch.distributed import distributed_c10d as c10d
+
+        store = c10d._get_default_store()
+        key = f"gats_slice_queue_{_shared_counter._epoch}"
+        _shared_counter._epoch += 1
+
+        def next_index():
+            return store.add(key, 1) - 1
+
+        return next_index
+    except Exception:
+        return None
+
+
+_shared_counter._epoch = 0
+
+
+def _run_dynamic(nodes, world, per_rank_slices, run_slice):
+    """Dynamic work distribution across ranks: the frontier is cut into
+    world*per_rank_slices round-robin sub-slices and ranks claim them off a
+    shared atomic queue until it drains — the reference's static interleave
+    (nqueens_dist_multigpu_chpl.chpl:223-227) upgraded with cross-rank load
+    balancing (its Chapel dist driver steals between locales for the same
+    reason; the queue achieves it without pausing running engines). Falls
+    back to the static partition when no shared store is available.
+    run_slice(bytes) -> result dict; returns the rank's summed dict."""
+    nslices = world * per_rank_slices
+    counter = _shared_counter() if world > 1 else None
+    total = {"tree": 0, "sol": 0, "optimum": 0, "time": 0.0, "diag": {}}
+    import time as _t
+
+    t0 = _t.perf_counter()
+    if counter is None:
+        my = [slice_frontier(nodes, 0, 1)] if world == 1 else None
+        if my is None:
+            rank = td.get_rank()
+            my = [slice_frontier(nodes, rank * per_rank_slices + j, nslices)
+                  for j in range(per_rank_slices)]
+        for s in my:
+            r = run_slice(s)
+            total["tree"] += r["tree"]
+            total["sol"] += r["sol"]
+            if r.get("optimum"):
+                total["optimum"] = min(total["optimum"] or r["optimum"], r["optimum"])
+            total["diag"] = r.get("diag", {})
+    else:
+        while True:
+            i = counter()
+            if i >= nslices:
+                break
+            r = run_slice(slice_frontier(nodes, i, nslices))
+            total["tree"] += r["tree"]
+            total["sol"] += r["sol"]
+            if r.get("optimum"):
+                total["optimum"] = min(total["optimum"] or r["optimum"], r["optimum"])
+            total["diag"] = r.get("diag", {})
+    total["time"] = _t.perf_counter() - t0
+    return total
+
+
 def slice_frontier(nodes: bytes, rank: int, world: int) -> bytes:
     """Round-robin node slice (rank, rank+world, ...), the reference's static
     interleaved partition (nqueens_dist_multigpu_chpl.chpl:223-227)."""
@@ -90,13 +155,15 @@ def run_nqueens(N, g=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
     if frontier_target is None:
         frontier_target = max(65536, 8192 * world)
     nodes, tree1, sol1 = c.nq_bfs_frontier(N, g, frontier_target)
-    my = slice_frontier(nodes, rank, world)
     phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
-    if engine == "gpu":
-        local = rank % max(1, c.gpu_device_count())
-        r = c.nqueens_gpu_from_pool(my, N, g, m, M, local, mode, capacity)
-    else:  # CPU path for gloo CI
-        r = c.nqueens_seq_from_pool(my, N, g)
+    local = rank % max(1, c.gpu_device_count())
+
+    def run_slice(sl):
+        if engine == "gpu":
+            return c.nqueens_gpu_from_pool(sl, N, g, m, M, local, mode, capacity)
+        return c.nqueens_seq_from_pool(sl, N, g)  # CPU path for gloo CI
+
+    r = _run_dynamic(nodes, world, 4, run_slice)
     if world == 1:
         r = dict(r)
         r["tree"] += phase1["tree"]
@@ -116,13 +183,17 @@ def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 
         # would move a large share of the search onto the single-threaded CPU
         frontier_target = max(2048, 2048 * world)
     nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
-    my = slice_frontier(nodes, rank, world)
     phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
-    if engine == "gpu":
-        local = rank % max(1, c.gpu_device_count())
-        r = c.pfsp_gpu_from_pool(my, inst, lb, ub, best, m, M, local, mode, capacity)
-    else:
-        r = c.pfsp_seq_from_pool(my, inst, lb, ub, best)
+    local = rank % max(1, c.gpu_device_count())
+
+    def run_slice(sl):
+        if engine == "gpu":
+            return c.pfsp_gpu_from_pool(sl, inst, lb, ub, best, m, M, local, mode, capacity)
+        return c.pfsp_seq_from_pool(sl, inst, lb, ub, best)
+
+    r = _run_dynamic(nodes, world, 4, run_slice)
+    if r["optimum"] == 0:
+        r["optimum"] = best
     if world == 1:
         r = dict(r)
         r["tree"] += phase1["tree"]
