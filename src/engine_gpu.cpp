@@ -180,10 +180,37 @@ struct PfspTablesGuard {
   }
 };
 
+// Streams are recycled per device like buffers (creation is ~0.1-1 ms, which
+// matters for few-ms PFSP searches). A released stream is always synchronized.
+struct StreamCache {
+  std::mutex mu;
+  std::map<int, std::vector<hipStream_t>> free_;
+};
+StreamCache& stream_cache() {
+  static StreamCache c;
+  return c;
+}
+
 struct StreamGuard {
   hipStream_t s{};
-  StreamGuard() { HIP_CHECK(hipStreamCreate(&s)); }
-  ~StreamGuard() { (void)hipStreamDestroy(s); }
+  int dev = 0;
+  StreamGuard() {
+    (void)hipGetDevice(&dev);
+    {
+      std::lock_guard<std::mutex> l(stream_cache().mu);
+      auto& v = stream_cache().free_[dev];
+      if (!v.empty()) {
+        s = v.back();
+        v.pop_back();
+        return;
+      }
+    }
+    HIP_CHECK(hipStreamCreate(&s));
+  }
+  ~StreamGuard() {
+    std::lock_guard<std::mutex> l(stream_cache().mu);
+    stream_cache().free_[dev].push_back(s);
+  }
 };
 
 template <typename T>
@@ -205,6 +232,20 @@ struct PinnedGuard {
   }
   ~PinnedGuard() { cached_pinned_free(p, bytes); }
 };
+
+// PFSP device bound tables are immutable per instance: build + upload once
+// per (device, instance) and share across searches/slices (rebuilding cost a
+// visible slice of few-ms searches).
+const PfspDevTables& pfsp_tables_cached(const PfspInstance& I, int device) {
+  static std::mutex mu;
+  static std::map<std::pair<int, int>, PfspTablesGuard*> cache;
+  std::lock_guard<std::mutex> l(mu);
+  auto key = std::make_pair(device, I.inst);
+  auto it = cache.find(key);
+  if (it == cache.end())
+    it = cache.emplace(key, new PfspTablesGuard(I)).first;  // lives for the process
+  return it->second->tb;
+}
 
 int lbk_of(LbKind lb) {
   switch (lb) {
@@ -636,7 +677,7 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
 
   HIP_CHECK(hipSetDevice(device));
   StreamGuard stream;
-  PfspTablesGuard tables(I);
+  const PfspDevTables& tb_dev = pfsp_tables_cached(I, device);
   const double t2 = now_sec();
 
   if (mode == "hostpool") {
@@ -649,7 +690,7 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       if (n == 0) break;
       HIP_CHECK(hipMemcpyAsync(parents_d.p, parents.p, n * sizeof(PFSPNode),
                                hipMemcpyHostToDevice, stream.s));
-      launch_pfsp_eval(parents_d.p, static_cast<int>(n), jobs, machines, lbk, tables.tb, best,
+      launch_pfsp_eval(parents_d.p, static_cast<int>(n), jobs, machines, lbk, tb_dev, best,
                        bounds_d.p, stream.s);
       HIP_CHECK(hipMemcpyAsync(bounds.p, bounds_d.p, n * jobs * sizeof(int32_t),
                                hipMemcpyDeviceToHost, stream.s));
@@ -688,7 +729,7 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     for (int t = 0; t < S; t++) {
       threads.emplace_back([&, t] {
         try {
-          outs[t] = devpool_thread_pfsp(slices, next_slice, I, tables.tb, lbk, best,
+          outs[t] = devpool_thread_pfsp(slices, next_slice, I, tb_dev, lbk, best,
                                         m, M, device, capacity, sb, S == 1, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
@@ -730,10 +771,23 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
   return r;
 }
 
+// Host-side instance (Taillard matrix + lb1/lb2 tables incl. 190 Johnson
+// sorts) cached per (inst, ub).
+static const PfspInstance& pfsp_instance_cached(int inst, int ub) {
+  static std::mutex mu;
+  static std::map<std::pair<int, int>, PfspInstance*> cache;
+  std::lock_guard<std::mutex> l(mu);
+  auto key = std::make_pair(inst, ub);
+  auto it = cache.find(key);
+  if (it == cache.end())
+    it = cache.emplace(key, new PfspInstance(make_pfsp_instance(inst, ub))).first;
+  return *it->second;
+}
+
 Result pfsp_gpu(int inst, const std::string& lb_str, int ub, int m, int M, int device,
                 const std::string& mode, unsigned long long capacity) {
   const LbKind lb = lb_from_string(lb_str);
-  PfspInstance I = make_pfsp_instance(inst, ub);
+  const PfspInstance& I = pfsp_instance_cached(inst, ub);
   Pool<PFSPNode> pool;
   pool.pushBack(pfsp_root());
   uint64_t tree = 0, sol = 0;
@@ -752,7 +806,7 @@ Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst,
                           const std::string& lb_str, int ub, int best0, int m, int M,
                           int device, const std::string& mode, unsigned long long capacity) {
   const LbKind lb = lb_from_string(lb_str);
-  PfspInstance I = make_pfsp_instance(inst, ub);
+  const PfspInstance& I = pfsp_instance_cached(inst, ub);
   Pool<PFSPNode> pool;
   pool.pushBackBulk(nodes.data(), nodes.size());
   const int best = (best0 > 0) ? best0 : I.init_ub;
@@ -813,11 +867,11 @@ PfspAsyncEngine::PfspAsyncEngine(std::vector<PFSPNode> nodes, int inst,
                                  int device, unsigned long long capacity)
     : shared_best_(0), done_(false) {
   const LbKind lb = lb_from_string(lb_str);
-  PfspInstance I = make_pfsp_instance(inst, ub);
+  const PfspInstance& I = pfsp_instance_cached(inst, ub);
   const int b0 = (best0 > 0) ? best0 : I.init_ub;
   shared_best_.store(b0);
-  th_ = std::thread([this, nodes = std::move(nodes), I = std::move(I), lb, m, M, device,
-                     capacity, b0]() mutable {
+  th_ = std::thread([this, nodes = std::move(nodes), &I, lb, m, M, device, capacity,
+                     b0]() mutable {
     try {
       Pool<PFSPNode> pool;
       pool.pushBackBulk(nodes.data(), nodes.size());
