@@ -194,9 +194,13 @@ StreamCache& stream_cache() {
 struct StreamGuard {
   hipStream_t s{};
   int dev = 0;
+  static bool caching() {
+    static const bool on = std::getenv("GATS_NO_STREAM_CACHE") == nullptr;
+    return on;
+  }
   StreamGuard() {
     (void)hipGetDevice(&dev);
-    {
+    if (caching()) {
       std::lock_guard<std::mutex> l(stream_cache().mu);
       auto& v = stream_cache().free_[dev];
       if (!v.empty()) {
@@ -208,8 +212,12 @@ struct StreamGuard {
     HIP_CHECK(hipStreamCreate(&s));
   }
   ~StreamGuard() {
-    std::lock_guard<std::mutex> l(stream_cache().mu);
-    stream_cache().free_[dev].push_back(s);
+    if (caching()) {
+      std::lock_guard<std::mutex> l(stream_cache().mu);
+      stream_cache().free_[dev].push_back(s);
+    } else {
+      (void)hipStreamDestroy(s);
+    }
   }
 };
 
