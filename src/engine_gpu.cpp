@@ -211,7 +211,7 @@ struct StreamGuard {
         return;
       }
     }
-    HIP_CHECK(hipStreamCreate(&s));
+    HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
   }
   ~StreamGuard() {
     if (caching()) {
@@ -443,12 +443,16 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
     const std::vector<NQNode>& nodes = slices[si];
     if (nodes.empty()) continue;
     if (nodes.size() > capacity) throw std::runtime_error("devpool capacity too small");
-    HIP_CHECK(
-        hipMemcpy(pool_d.p, nodes.data(), nodes.size() * sizeof(NQNode), hipMemcpyHostToDevice));
+    // per-stream async copies: synchronous hipMemcpy runs on the NULL stream
+    // and would serialize every slice thread at each slice boundary
+    HIP_CHECK(hipMemcpyAsync(pool_d.p, nodes.data(), nodes.size() * sizeof(NQNode),
+                             hipMemcpyHostToDevice, stream.s));
     DevCtl ctl{};
     ctl.size = nodes.size();
-    HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-    HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpyAsync(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
+    HIP_CHECK(
+        hipMemcpyAsync(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
+    HIP_CHECK(hipStreamSynchronize(stream.s));  // ctl is a stack temporary
     r.h2d += 2;
     r.h2d_bytes += nodes.size() * sizeof(NQNode) + 2 * sizeof(DevCtl);
     const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, nullptr,
@@ -459,8 +463,9 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
     if (fin.size > 0) {
       const size_t base = leftover.size();
       leftover.resize(base + fin.size);
-      HIP_CHECK(hipMemcpy(leftover.data() + base, pool_d.p, fin.size * sizeof(NQNode),
-                          hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpyAsync(leftover.data() + base, pool_d.p, fin.size * sizeof(NQNode),
+                               hipMemcpyDeviceToHost, stream.s));
+      HIP_CHECK(hipStreamSynchronize(stream.s));
       r.d2h++;
       r.d2h_bytes += fin.size * sizeof(NQNode);
     }
@@ -507,14 +512,16 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
     const std::vector<PFSPNode>& nodes = slices[si];
     if (nodes.empty()) continue;
     if (nodes.size() > capacity) throw std::runtime_error("devpool capacity too small");
-    HIP_CHECK(hipMemcpy(pool_d.p, nodes.data(), nodes.size() * sizeof(PFSPNode),
-                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpyAsync(pool_d.p, nodes.data(), nodes.size() * sizeof(PFSPNode),
+                             hipMemcpyHostToDevice, stream.s));
     DevCtl ctl{};
     ctl.size = nodes.size();
     // adopt the freshest incumbent before starting the slice
     ctl.best = shared_best ? shared_best->load(std::memory_order_relaxed) : best0;
-    HIP_CHECK(hipMemcpy(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
-    HIP_CHECK(hipMemcpy(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpyAsync(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
+    HIP_CHECK(
+        hipMemcpyAsync(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
+    HIP_CHECK(hipStreamSynchronize(stream.s));
     r.h2d += 2;
     r.h2d_bytes += nodes.size() * sizeof(PFSPNode) + 2 * sizeof(DevCtl);
     const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best,
@@ -526,8 +533,10 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
     if (fin.size > 0) {
       const size_t base = leftover.size();
       leftover.resize(base + fin.size);
-      HIP_CHECK(hipMemcpy(leftover.data() + base, pool_d.p, fin.size * sizeof(PFSPNode),
-                          hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpyAsync(leftover.data() + base, pool_d.p,
+                               fin.size * sizeof(PFSPNode), hipMemcpyDeviceToHost,
+                               stream.s));
+      HIP_CHECK(hipStreamSynchronize(stream.s));
       r.d2h++;
       r.d2h_bytes += fin.size * sizeof(PFSPNode);
     }
