@@ -185,11 +185,18 @@ def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 
     nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
     phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
     local = rank % max(1, c.gpu_device_count())
+    best_so_far = [best]
 
     def run_slice(sl):
+        # carry this rank's improved incumbent into its next queue slice
         if engine == "gpu":
-            return c.pfsp_gpu_from_pool(sl, inst, lb, ub, best, m, M, local, mode, capacity)
-        return c.pfsp_seq_from_pool(sl, inst, lb, ub, best)
+            r = c.pfsp_gpu_from_pool(sl, inst, lb, ub, best_so_far[0], m, M, local, mode,
+                                     capacity)
+        else:
+            r = c.pfsp_seq_from_pool(sl, inst, lb, ub, best_so_far[0])
+        if r.get("optimum"):
+            best_so_far[0] = min(best_so_far[0], r["optimum"])
+        return r
 
     r = _run_dynamic(nodes, world, 4, run_slice)
     if r["optimum"] == 0:
