@@ -39,3 +39,17 @@ def test_pfsp_multipool_lb2_ub1_matches_seq(core):
     r = core.pfsp_multigpu(14, "lb2", 1, 25, 2000, 3, "cpu", True)
     assert r["tree"] == seq["tree"]
     assert r["sol"] == seq["sol"]
+
+
+def test_tiny_problem_more_workers_than_nodes(core):
+    # frontier smaller than D*m: workers must still terminate via all-idle
+    seq = core.nqueens_seq(6, 1)
+    r = core.nqueens_multigpu(6, 1, 25, 5000, 8, "cpu")
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+
+
+def test_pfsp_lb1_multipool_matches_seq(core):
+    seq = core.pfsp_seq(2, "lb1", 0)
+    r = core.pfsp_multigpu(2, "lb1", 0, 5, 128, 3, "cpu", True)
+    assert r["optimum"] == seq["optimum"] == 1359
