@@ -11,9 +11,11 @@ Capability parity with the reference's distributed drivers
   - end-of-search collectives: all_reduce SUM on tree/sol, MIN on the PFSP
     incumbent, MAX on elapsed time (MPI parity: MPI_Reduce calls at
     pfsp_dist_multigpu_cuda.c:681-694)
-No inter-rank work stealing yet — the reference's own MPI/CUDA baseline has
-none either; the round-robin interleave of a deep frontier
-keeps the slices statistically balanced.
+Cross-rank load balancing: ranks claim 4x-oversubscribed round-robin frontier
+sub-slices off a shared atomic queue (the process-group store) — the role the
+reference's inter-locale PGAS stealing plays
+(nqueens_dist_multigpu_chpl.chpl:332-377); its own MPI/CUDA baseline has no
+inter-rank stealing at all.
 
 The RCCL collectives are tiny (a few int64 scalars); latency-bound, far from
 any xGMI bandwidth limit, so RCCL defaults are the right choice (SURVEY.md
