@@ -503,11 +503,15 @@ __device__ inline unsigned long long derive_chunk(const DevCtl* ctl, unsigned lo
 // Children within `finish` levels of the bottom are counted in-thread by the
 // bitmask DFS (nq_dfs) instead of being pushed: the deepest levels dominate
 // the tree, so the pool machinery only carries the shallow part.
+// A phase-A pass computes each PARENT's (cols, d1, d2) diagonal masks once
+// (one thread per staged parent), so a child's safety test and the finisher's
+// starting masks are O(1) instead of an O(depth) walk per child.
 __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                        uint32_t* blockCounts, unsigned long long* blockSols,
                        unsigned long long* blockExtra, int N, int g, int finish,
                        unsigned long long m, unsigned long long M) {
   __shared__ NQNode s[EMIT_TILE / 4 + 2];  // N >= 4 (engine falls back below)
+  __shared__ uint32_t pmask[EMIT_TILE / 4 + 2][3];  // per-parent cols/d1/d2
   const unsigned long long c = derive_chunk(ctl, m, M);
   // child indices fit u32: the engine enforces M * branching <= 2^31
   const uint32_t total = static_cast<uint32_t>(c * N);
@@ -524,6 +528,26 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
     if (c1 > total) c1 = total;
     first = stage_range(parents, c0, c1, N, s);
     __syncthreads();
+    {  // phase A: one thread per staged parent
+      const uint32_t msk = (1u << N) - 1u;
+      const int nblk = static_cast<int>((c1 - 1) / N - first + 1);
+      for (int pi = threadIdx.x; pi < nblk; pi += blockDim.x) {
+        const NQNode& p = s[pi];
+        uint32_t cols = 0, d1 = 0, d2 = 0;
+        const int depth = p.depth;
+        for (int i = 0; i < depth && i < N; i++) {
+          uint32_t b = 1u << p.board[i];
+          for (int r = 1; r < g; r++) b = 1u << p.board[i];  // artificial work knob
+          cols |= b;
+          d1 = ((d1 | b) << 1) & msk;
+          d2 = (d2 | b) >> 1;
+        }
+        pmask[pi][0] = cols;
+        pmask[pi][1] = d1;
+        pmask[pi][2] = d2;
+      }
+    }
+    __syncthreads();
 #pragma unroll
     for (int j = 0; j < EMIT_TILE / BLOCK; j++) {
       const uint32_t t = c0 + j * BLOCK + threadIdx.x;
@@ -536,30 +560,26 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
         const int depth = p.depth;
         if (depth == N) {
           sols += (k == 0);  // leaf parent counted once (nqueens_chpl.chpl:78-80)
-        } else if (k >= depth && nq_safe(p.board, depth, p.board[k], g)) {
-          const int rem = N - (depth + 1);  // levels below the child
-          if (rem <= finish) {
-            // child + its whole subtree counted here, nothing pushed
-            extra += 1;
-            if (rem == 0) {
-              sols += 1;
-            } else {
-              const uint32_t msk = (1u << N) - 1u;
-              uint32_t cols = 0, d1 = 0, d2 = 0;
-              for (int i = 0; i < depth; i++) {
-                const uint32_t b = 1u << s[pid - first].board[i];
-                cols |= b;
-                d1 = ((d1 | b) << 1) & msk;
-                d2 = (d2 | b) >> 1;
+        } else if (k >= depth) {
+          const uint32_t cols = pmask[pid - first][0];
+          const uint32_t d1 = pmask[pid - first][1];
+          const uint32_t d2 = pmask[pid - first][2];
+          const uint32_t b = 1u << p.board[k];
+          if (!(b & (cols | d1 | d2))) {  // == nq_safe (diagonal masks)
+            const int rem = N - (depth + 1);  // levels below the child
+            if (rem <= finish) {
+              // child + its whole subtree counted here, nothing pushed
+              extra += 1;
+              if (rem == 0) {
+                sols += 1;
+              } else {
+                const uint32_t msk = (1u << N) - 1u;
+                nq_dfs<NQ_FINISH_MAX>(cols | b, ((d1 | b) << 1) & msk, (d2 | b) >> 1,
+                                      depth + 1, N, extra, sols);
               }
-              const uint32_t b = 1u << s[pid - first].board[k];
-              cols |= b;
-              d1 = ((d1 | b) << 1) & msk;
-              d2 = (d2 | b) >> 1;
-              nq_dfs<NQ_FINISH_MAX>(cols, d1, d2, depth + 1, N, extra, sols);
+            } else {
+              lab[j] = 1;
             }
-          } else {
-            lab[j] = 1;
           }
         }
         cnt += (lab[j] == 1);
