@@ -119,3 +119,11 @@ def test_from_pool_empty_frontier(gpu):
     assert r["tree"] == 0 and r["sol"] == 0
     p = gpu.pfsp_gpu_from_pool(b"", 14, "lb1", 1, 0, 25, 50000, 0, "devpool", 1 << 22)
     assert p["tree"] == 0 and p["sol"] == 0
+
+
+@pytest.mark.parametrize("inst", [3, 9, 16, 18])
+def test_pfsp_instance_breadth_lb2(gpu, inst):
+    # fast members of the ta001-ta020 sweep (full sweep:
+    # profiles/taillard_sweep_lb2.txt); optimum must be proven from ub=1
+    r = gpu.pfsp_gpu(inst, "lb2", 1, 25, 50000, 0, "devpool", 1 << 25)
+    assert r["optimum"] == gpu.taillard_best_ub(inst)
