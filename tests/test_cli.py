@@ -34,3 +34,25 @@ def test_pfsp_seq_output_contract(core, tmp_path):
     assert "Optimal makespan:  1359  (improved)" in out
     line = stats.read_text().strip()
     assert line.startswith("ta2 lb2 SEQ ") and line.endswith(" 33110 59 1359")
+
+
+def test_help_exits_cleanly(core):
+    import pytest as _pytest
+
+    from gats_amd import cli
+
+    with _pytest.raises(SystemExit) as e:
+        cli.main(["nqueens", "--help"])
+    assert e.value.code == 0
+    with _pytest.raises(SystemExit) as e:
+        cli.main(["pfsp", "-h"])
+    assert e.value.code == 0
+
+
+def test_invalid_lb_rejected(core):
+    import pytest as _pytest
+
+    from gats_amd import cli
+
+    with _pytest.raises(SystemExit):
+        cli.main(["pfsp", "--lb", "bogus", "--tier", "seq"])
