@@ -14,12 +14,13 @@
 //                     control blocks) keep the entire hot loop on the GPU;
 //                     the host only polls a 64 B control block every few
 //                     iterations. No global atomics on the hot path.
-//   * Bound tables staged in LDS (p_times/lags int16, Johnson schedules u8):
-//     ~12 KB for 20x20 vs 160 KB per CU.
+//   * Bound tables staged in LDS (p_times int16; Johnson entries packed into
+//     one u64 per (pair, position): job<<48|lag<<32|ptm1<<16|ptm0): ~31 KB
+//     for 20x20 vs 160 KB per CU.
 //   * No runtime-indexed per-thread arrays (they would spill to scratch on
-//     CDNA4): lb2's machine-pair-indexed `front` lives in LDS with a
-//     conflict-free padded stride; lb1_d iterates positions with uniform-
-//     length wave loops instead of a job-indexed local array.
+//     CDNA4): machine loops are fully unrolled via a machine-count template
+//     parameter; lb2's machine pairs are a compile-time lexicographic map;
+//     lb1_d iterates positions instead of a job-indexed local array.
 //   * Wavefront = 64 idioms throughout (__ballot is 64-bit).
 #include <hip/hip_runtime.h>
 
@@ -32,7 +33,7 @@ namespace gats {
 #define BLOCK 256
 
 // ---------------------------------------------------------------------------
-// Wave64 helpers
+// Shared helpers (LDS staging, child emission)
 // ---------------------------------------------------------------------------
 
 // Block-cooperative staging of the parents this block touches into LDS.
