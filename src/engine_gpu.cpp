@@ -19,6 +19,9 @@
 #include <algorithm>
 #include <atomic>
 #include <cstdlib>
+#include <chrono>
+#include <condition_variable>
+#include <functional>
 #include <map>
 #include <mutex>
 #include <thread>
@@ -282,11 +285,17 @@ int lbk_of(LbKind lb) {
 // enqueue_iter(parity): parity alternates 0/1 per iteration — iteration i
 // reads ctl[parity] and gather2 writes ctl[1-parity]. BATCH is even, so the
 // host always reads (and patches the incumbent of) ctl[0].
+// `readback_hook(host_ctl, done_soon)`: called after every batch readback with
+// the stream idle; may reduce host_ctl->size after carving the pool (the
+// donor path of SliceShare) — it must then also write the device ctl itself.
+using ReadbackHook = std::function<void(DevCtl*)>;
+
 template <class EnqueueIter>
 static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long m,
                                int kernels_per_iter, EnqueueIter&& enqueue_iter, Result& r,
                                std::atomic<int>* shared_best = nullptr,
-                               bool allow_graph = true) {
+                               bool allow_graph = true,
+                               const ReadbackHook& readback_hook = {}) {
   PinnedGuard<DevCtl> ctl_h(1);
   const int BATCH = 16;
   // GATS_NO_GRAPH=1 falls back to eager launches (rocprofv3 crashes tracing
@@ -360,6 +369,7 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
       overflow = true;
       break;
     }
+    if (readback_hook) readback_hook(ctl_h.p);
     if (shared_best) {
       int mine = ctl_h.p->best;
       int cur = shared_best->load(std::memory_order_relaxed);
@@ -409,14 +419,135 @@ struct SliceOut {
   Result diag;
 };
 
+// Work-sharing between the slice threads of ONE engine: a thread whose queue
+// drained waits here; a running thread donates the BACK half of its device
+// pool at a readback boundary (stream synced, so the handoff region is
+// quiescent until the thief's D2D copy acks). Steal-half + the >=2m donor
+// threshold mirror the reference's stealing rule (Pool_par.chpl:153-165);
+// all waits are bounded, and a waiter exits once no runner remains.
+struct SliceShare {
+  std::mutex mu;
+  std::condition_variable cv;
+  int idle = 0;     // threads waiting for donated work
+  int runners = 0;  // threads currently inside a devpool loop
+  bool offer_ready = false;
+  bool offer_claimed = false;  // a thief took the pointer and is copying
+  bool offer_taken = false;    // the thief's D2D copy completed
+  const void* offer_src = nullptr;  // device ptr to the donated node range
+  unsigned long long offer_n = 0;
+  int offer_best = 0;
+};
+
 // Queue-driven worker thread: allocates its device buffers once, then keeps
 // claiming frontier slices from the shared index until none remain. Slices
 // are oversubscribed (~4 per thread) so a thread whose slice finishes early
 // just pulls the next one — tail balancing without inter-thread stealing.
+// Generic steal-share helpers for a slice thread (node type erased to bytes).
+// Donor side: runs at readback boundaries (stream idle); carves the BACK half
+// of the live pool when someone is waiting and the pool holds >= 2m nodes,
+// then blocks (bounded waits) until the thief's D2D copy acks.
+template <class NodeT>
+static void donate_if_wanted(SliceShare* share, DevCtl* host_ctl, DevCtl* ctl_d,
+                             NodeT* pool_d, unsigned long long m, hipStream_t s) {
+  if (!share || host_ctl->overflow) return;
+  if (host_ctl->size < 2 * m) return;
+  if (share->idle == 0) return;  // racy fast path; rechecked under the lock
+  std::unique_lock<std::mutex> lock(share->mu);
+  if (share->idle == 0 || share->offer_ready) return;
+  const unsigned long long half = host_ctl->size / 2;
+  const unsigned long long newsize = host_ctl->size - half;
+  // the stream is idle (caller synced), so rewriting ctl[0].size is safe
+  HIP_CHECK(hipMemcpyAsync(&ctl_d->size, &newsize, sizeof(newsize), hipMemcpyHostToDevice,
+                           s));
+  HIP_CHECK(hipStreamSynchronize(s));
+  share->offer_src = pool_d + newsize;
+  share->offer_n = half;
+  share->offer_best = host_ctl->best;
+  share->offer_ready = true;
+  share->offer_claimed = false;
+  share->offer_taken = false;
+  share->cv.notify_all();
+  const auto deadline = std::chrono::steady_clock::now() + std::chrono::seconds(10);
+  while (!share->offer_taken) {
+    if (!share->offer_claimed && share->idle == 0) {
+      // every waiter left (e.g. threw): withdraw the offer, restore the pool
+      share->offer_ready = false;
+      const unsigned long long restore = host_ctl->size;
+      HIP_CHECK(hipMemcpyAsync(&ctl_d->size, &restore, sizeof(restore),
+                               hipMemcpyHostToDevice, s));
+      HIP_CHECK(hipStreamSynchronize(s));
+      return;
+    }
+    if (share->offer_claimed && std::chrono::steady_clock::now() > deadline) {
+      // thief died mid-copy (an exception is already propagating): give the
+      // nodes up rather than double-count them
+      share->offer_ready = false;
+      host_ctl->size = newsize;
+      return;
+    }
+    share->cv.wait_for(lock, std::chrono::milliseconds(20));
+  }
+  share->offer_ready = false;
+  host_ctl->size = newsize;  // keep the loop's own view consistent
+}
+
+// Thief side: returns true with (src, n, best) filled when donated work was
+// claimed (caller must D2D-copy it and ack via ack_taken), false when no
+// runner remains. Caller holds no lock.
+struct StolenWork {
+  const void* src = nullptr;
+  unsigned long long n = 0;
+  int best = 0;
+};
+
+static bool wait_for_work(SliceShare& share, StolenWork& w) {
+  std::unique_lock<std::mutex> lock(share.mu);
+  share.idle++;
+  while (true) {
+    if (share.offer_ready && !share.offer_claimed) {
+      w.src = share.offer_src;
+      w.n = share.offer_n;
+      w.best = share.offer_best;
+      share.offer_claimed = true;  // ack via ack_taken after the copy
+      share.idle--;
+      return true;
+    }
+    if (share.runners == 0) {
+      share.idle--;
+      return false;
+    }
+    share.cv.wait_for(lock, std::chrono::milliseconds(20));
+  }
+}
+
+static void ack_taken(SliceShare& share) {
+  std::lock_guard<std::mutex> lock(share.mu);
+  share.offer_taken = true;
+  share.cv.notify_all();
+}
+
+struct RunnerScope {
+  SliceShare* s;
+  explicit RunnerScope(SliceShare* share) : s(share) {
+    if (s) {
+      std::lock_guard<std::mutex> l(s->mu);
+      s->runners++;
+    }
+  }
+  ~RunnerScope() {
+    if (s) {
+      std::lock_guard<std::mutex> l(s->mu);
+      s->runners--;
+      s->cv.notify_all();
+    }
+  }
+};
+
 static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices,
                                   std::atomic<int>& next_slice, int N, int g, int m, int M,
                                   int device, int finish, unsigned long long capacity,
-                                  bool allow_graph, std::vector<NQNode>& leftover) {
+                                  bool allow_graph, SliceShare* share,
+                                  std::vector<NQNode>& leftover) {
   HIP_CHECK(hipSetDevice(device));
   StreamGuard stream;
   SliceOut out;
@@ -437,26 +568,22 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
     launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, childbuf_d.p, pool_d.p, stride, G,
                       m, M, capacity, stream.s);
   };
+  ReadbackHook hook = [&](DevCtl* hc) {
+    donate_if_wanted(share, hc, ctl_d.p, pool_d.p, m, stream.s);
+  };
 
-  int si;
-  while ((si = next_slice.fetch_add(1)) < static_cast<int>(slices.size())) {
-    const std::vector<NQNode>& nodes = slices[si];
-    if (nodes.empty()) continue;
-    if (nodes.size() > capacity) throw std::runtime_error("devpool capacity too small");
-    // per-stream async copies: synchronous hipMemcpy runs on the NULL stream
-    // and would serialize every slice thread at each slice boundary
-    HIP_CHECK(hipMemcpyAsync(pool_d.p, nodes.data(), nodes.size() * sizeof(NQNode),
-                             hipMemcpyHostToDevice, stream.s));
+  auto run_pool = [&](unsigned long long init_size) {
     DevCtl ctl{};
-    ctl.size = nodes.size();
+    ctl.size = init_size;
     HIP_CHECK(hipMemcpyAsync(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
     HIP_CHECK(
         hipMemcpyAsync(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
     HIP_CHECK(hipStreamSynchronize(stream.s));  // ctl is a stack temporary
     r.h2d += 2;
-    r.h2d_bytes += nodes.size() * sizeof(NQNode) + 2 * sizeof(DevCtl);
+    r.h2d_bytes += 2 * sizeof(DevCtl);
+    RunnerScope runner(share);
     const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, nullptr,
-                                        allow_graph);
+                                        allow_graph, hook);
     out.fin.tree += fin.tree;
     out.fin.sol += fin.sol;
     r.gpu_iters += fin.iters;
@@ -469,6 +596,32 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
       r.d2h++;
       r.d2h_bytes += fin.size * sizeof(NQNode);
     }
+  };
+
+  int si;
+  while ((si = next_slice.fetch_add(1)) < static_cast<int>(slices.size())) {
+    const std::vector<NQNode>& nodes = slices[si];
+    if (nodes.empty()) continue;
+    if (nodes.size() > capacity) throw std::runtime_error("devpool capacity too small");
+    // per-stream async copies: synchronous hipMemcpy runs on the NULL stream
+    // and would serialize every slice thread at each slice boundary
+    HIP_CHECK(hipMemcpyAsync(pool_d.p, nodes.data(), nodes.size() * sizeof(NQNode),
+                             hipMemcpyHostToDevice, stream.s));
+    r.h2d++;
+    r.h2d_bytes += nodes.size() * sizeof(NQNode);
+    run_pool(nodes.size());
+  }
+
+  // queue drained: take donated halves of still-running slices until no
+  // runner remains (ROADMAP item 2, landed)
+  while (share) {
+    StolenWork w;
+    if (!wait_for_work(*share, w)) break;
+    HIP_CHECK(hipMemcpyAsync(pool_d.p, w.src, w.n * sizeof(NQNode), hipMemcpyDeviceToDevice,
+                             stream.s));
+    HIP_CHECK(hipStreamSynchronize(stream.s));
+    ack_taken(*share);
+    run_pool(w.n);
   }
   return out;
 }
@@ -478,7 +631,7 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
                                     const PfspDevTables& tb, int lbk, int best0,
                                     int m, int M, int device, unsigned long long capacity,
                                     std::atomic<int>* shared_best, bool allow_graph,
-                                    std::vector<PFSPNode>& leftover) {
+                                    SliceShare* share, std::vector<PFSPNode>& leftover) {
   HIP_CHECK(hipSetDevice(device));
   StreamGuard stream;
   SliceOut out;
@@ -506,26 +659,26 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
     launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, presum ? gsum_d.p : nullptr,
                         childbuf_d.p, pool_d.p, stride, G, m, M, capacity, stream.s);
   };
+  ReadbackHook hook = [&](DevCtl* hc) {
+    donate_if_wanted(share, hc, ctl_d.p, pool_d.p, m, stream.s);
+  };
 
-  int si;
-  while ((si = next_slice.fetch_add(1)) < static_cast<int>(slices.size())) {
-    const std::vector<PFSPNode>& nodes = slices[si];
-    if (nodes.empty()) continue;
-    if (nodes.size() > capacity) throw std::runtime_error("devpool capacity too small");
-    HIP_CHECK(hipMemcpyAsync(pool_d.p, nodes.data(), nodes.size() * sizeof(PFSPNode),
-                             hipMemcpyHostToDevice, stream.s));
+  auto run_pool = [&](unsigned long long init_size, int init_best) {
     DevCtl ctl{};
-    ctl.size = nodes.size();
-    // adopt the freshest incumbent before starting the slice
-    ctl.best = shared_best ? shared_best->load(std::memory_order_relaxed) : best0;
+    ctl.size = init_size;
+    // adopt the freshest incumbent before starting
+    const int sb_now =
+        shared_best ? shared_best->load(std::memory_order_relaxed) : init_best;
+    ctl.best = sb_now < init_best ? sb_now : init_best;
     HIP_CHECK(hipMemcpyAsync(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
     HIP_CHECK(
         hipMemcpyAsync(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
     HIP_CHECK(hipStreamSynchronize(stream.s));
     r.h2d += 2;
-    r.h2d_bytes += nodes.size() * sizeof(PFSPNode) + 2 * sizeof(DevCtl);
+    r.h2d_bytes += 2 * sizeof(DevCtl);
+    RunnerScope runner(share);
     const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best,
-                                        allow_graph);
+                                        allow_graph, hook);
     out.fin.tree += fin.tree;
     out.fin.sol += fin.sol;
     if (fin.best < out.fin.best) out.fin.best = fin.best;
@@ -540,6 +693,29 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
       r.d2h++;
       r.d2h_bytes += fin.size * sizeof(PFSPNode);
     }
+  };
+
+  int si;
+  while ((si = next_slice.fetch_add(1)) < static_cast<int>(slices.size())) {
+    const std::vector<PFSPNode>& nodes = slices[si];
+    if (nodes.empty()) continue;
+    if (nodes.size() > capacity) throw std::runtime_error("devpool capacity too small");
+    HIP_CHECK(hipMemcpyAsync(pool_d.p, nodes.data(), nodes.size() * sizeof(PFSPNode),
+                             hipMemcpyHostToDevice, stream.s));
+    r.h2d++;
+    r.h2d_bytes += nodes.size() * sizeof(PFSPNode);
+    run_pool(nodes.size(), best0);
+  }
+
+  // queue drained: take donated halves of still-running slices (ROADMAP #2)
+  while (share) {
+    StolenWork w;
+    if (!wait_for_work(*share, w)) break;
+    HIP_CHECK(hipMemcpyAsync(pool_d.p, w.src, w.n * sizeof(PFSPNode),
+                             hipMemcpyDeviceToDevice, stream.s));
+    HIP_CHECK(hipStreamSynchronize(stream.s));
+    ack_taken(*share);
+    run_pool(w.n, w.best);
   }
   return out;
 }
@@ -613,6 +789,8 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
       pool.clear();
     }
     std::atomic<int> next_slice{0};
+    SliceShare share;
+    SliceShare* sharep = (S > 1) ? &share : nullptr;
     std::vector<SliceOut> outs(S);
     std::vector<std::vector<NQNode>> lefts(S);
     std::vector<std::exception_ptr> errs(S);
@@ -621,7 +799,7 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
       threads.emplace_back([&, t] {
         try {
           outs[t] = devpool_thread_nq(slices, next_slice, N, g, m, M, device, finish,
-                                      capacity, S == 1, lefts[t]);
+                                      capacity, S == 1, sharep, lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
         }
@@ -741,6 +919,8 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     std::atomic<int> local_best{best};
     std::atomic<int>* sb = shared_best ? shared_best : &local_best;
     std::atomic<int> next_slice{0};
+    SliceShare share;
+    SliceShare* sharep = (S > 1) ? &share : nullptr;
     std::vector<SliceOut> outs(S);
     std::vector<std::vector<PFSPNode>> lefts(S);
     std::vector<std::exception_ptr> errs(S);
@@ -749,7 +929,8 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       threads.emplace_back([&, t] {
         try {
           outs[t] = devpool_thread_pfsp(slices, next_slice, I, tb_dev, lbk, best,
-                                        m, M, device, capacity, sb, S == 1, lefts[t]);
+                                        m, M, device, capacity, sb, S == 1, sharep,
+                                        lefts[t]);
         } catch (...) {
           errs[t] = std::current_exception();
         }
