@@ -450,7 +450,11 @@ template <class NodeT>
 static void donate_if_wanted(SliceShare* share, DevCtl* host_ctl, DevCtl* ctl_d,
                              NodeT* pool_d, unsigned long long m, hipStream_t s) {
   if (!share || host_ctl->overflow) return;
-  if (host_ctl->size < 2 * m) return;
+  // donation floor: a half-pool must be a worthwhile work grant (the thief
+  // pays ~1 ms of engine start-up) — 2m alone (the reference's steal
+  // threshold) caused donation thrash on large searches
+  const unsigned long long floor_sz = std::max<unsigned long long>(2 * m, 1u << 16);
+  if (host_ctl->size < floor_sz) return;
   if (share->idle == 0) return;  // racy fast path; rechecked under the lock
   std::unique_lock<std::mutex> lock(share->mu);
   if (share->idle == 0 || share->offer_ready) return;
