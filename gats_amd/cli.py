@@ -86,6 +86,9 @@ def add_common(p):
     p.add_argument("--mode", default="devpool", choices=["devpool", "hostpool"])
     p.add_argument("--capacity", type=int, default=1 << 27,
                    help="devpool device-pool capacity in nodes (288 GB HBM3E)")
+    p.add_argument("--perc", type=int, default=50,
+                   help="work-stealing percentage 1..99 (reference --perc; used by the "
+                        "multipool work-stealing engine; the devpool tier steals half)")
     p.add_argument("--stats-file", default=None,
                    help="append a result line (reference stats_*.dat parity)")
 
@@ -108,6 +111,8 @@ def main(argv=None):
     add_common(pf)
 
     args = ap.parse_args(argv)
+    if not (0 < args.perc < 100):
+        ap.error("unsupported work-stealing percentage (0 < --perc < 100)")
     c = gats_amd.core()
 
     if args.problem == "nqueens":
@@ -126,7 +131,8 @@ def main(argv=None):
             _phases(r)
         elif args.tier == "multigpu":
             gats_amd.require_gpu()
-            r = c.nqueens_multigpu(args.N, args.g, args.m, args.M, args.D, "devpool")
+            r = c.nqueens_multigpu(args.N, args.g, args.m, args.M, args.D, "devpool",
+                                   args.perc / 100.0)
             _phases(r)
         else:
             from gats_amd import dist
@@ -158,7 +164,8 @@ def main(argv=None):
             _phases(r)
         elif args.tier == "multigpu":
             gats_amd.require_gpu()
-            r = c.pfsp_multigpu(inst, args.lb, args.ub, args.m, args.M, args.D, "devpool", False)
+            r = c.pfsp_multigpu(inst, args.lb, args.ub, args.m, args.M, args.D, "devpool",
+                                False, args.perc / 100.0)
             _phases(r)
         else:
             from gats_amd import dist

@@ -35,9 +35,10 @@ Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst, const st
                           const std::string& mode, unsigned long long capacity);
 
 // engine_multi.cpp: in-process multi-GPU tier (eval = "gpu" or "cpu").
-Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& eval);
+Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& eval,
+                        double perc = 0.5);
 Result pfsp_multigpu(int inst, const std::string& lb, int ub, int m, int M, int D,
-                     const std::string& eval, bool share_best);
+                     const std::string& eval, bool share_best, double perc = 0.5);
 
 // Background-thread PFSP engine with a shared incumbent for mid-search
 // RCCL UB exchange (see engine_gpu.cpp).

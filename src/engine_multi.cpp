@@ -65,9 +65,9 @@ struct WorkerDiag {
 // EvalGen: (parents, n, own_pool, diag) -> void; must lock the pool around
 // its pushes.
 template <class NodeT, class EvalGen>
-void ws_worker(int id, int D, int m, int M, std::vector<ParPool<NodeT>>& pools,
-               std::vector<std::atomic<bool>>& states, std::atomic<bool>& all_idle_flag,
-               EvalGen&& eval_gen, WorkerDiag& diag) {
+void ws_worker(int id, int D, int m, int M, double perc,
+               std::vector<ParPool<NodeT>>& pools, std::vector<std::atomic<bool>>& states,
+               std::atomic<bool>& all_idle_flag, EvalGen&& eval_gen, WorkerDiag& diag) {
   ParPool<NodeT>& own = pools[id];
   std::vector<NodeT> parents(M);
   std::vector<NodeT> steal_buf;
@@ -94,7 +94,7 @@ void ws_worker(int id, int D, int m, int M, std::vector<ParPool<NodeT>>& pools,
         ParPool<NodeT>& victim = pools[v];
         for (int attempt = 0; attempt < 10; attempt++) {
           if (victim.tryLock()) {
-            size_t got = victim.popFrontHalfFree(m, steal_buf);
+            size_t got = victim.popFrontFracFree(m, perc, steal_buf);
             if (got > static_cast<size_t>(M)) {
               // cap like the reference's (m, M) window
               victim.pushBackBulkFree(steal_buf.data() + M, got - M);
@@ -362,8 +362,11 @@ static Result pfsp_multigpu_devpool(int inst, const std::string& lb_str, int ub,
   return r;
 }
 
-Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& eval) {
+Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& eval,
+                        double perc) {
   if (D < 1) throw std::invalid_argument("D must be >= 1");
+  if (!(perc > 0.0 && perc < 1.0))
+    throw std::invalid_argument("perc must be in (0,1) (reference: 0 < --perc < 100)");
   if (eval == "devpool") return nqueens_multigpu_devpool(N, g, m, M, D, 1ull << 27);
   Result r;
   Pool<NQNode> pool;
@@ -393,7 +396,7 @@ Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& ev
         if (eval == "gpu") {
           NqGpuCtx ctx(id % ndev, N, g, M);
           ws_worker<NQNode>(
-              id, D, m, M, pools, states, all_idle_flag,
+              id, D, m, M, perc, pools, states, all_idle_flag,
               [&](const NQNode* parents, size_t n, ParPool<NQNode>& own, WorkerDiag& dg) {
                 std::memcpy(ctx.parents_h, parents, n * sizeof(NQNode));
                 HIP_CHECK_M(hipMemcpyAsync(ctx.parents_d, ctx.parents_h, n * sizeof(NQNode),
@@ -416,7 +419,7 @@ Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& ev
               diags[id]);
         } else {
           ws_worker<NQNode>(
-              id, D, m, M, pools, states, all_idle_flag,
+              id, D, m, M, perc, pools, states, all_idle_flag,
               [&](const NQNode* parents, size_t n, ParPool<NQNode>& own, WorkerDiag& dg) {
                 own.acquireLock();
                 for (size_t i = 0; i < n; i++)
@@ -468,8 +471,10 @@ Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& ev
 }
 
 Result pfsp_multigpu(int inst, const std::string& lb_str, int ub, int m, int M, int D,
-                     const std::string& eval, bool share_best) {
+                     const std::string& eval, bool share_best, double perc) {
   if (D < 1) throw std::invalid_argument("D must be >= 1");
+  if (!(perc > 0.0 && perc < 1.0))
+    throw std::invalid_argument("perc must be in (0,1) (reference: 0 < --perc < 100)");
   if (eval == "devpool") return pfsp_multigpu_devpool(inst, lb_str, ub, m, M, D, 1ull << 27);
   const LbKind lb = lb_from_string(lb_str);
   PfspInstance I = make_pfsp_instance(inst, ub);
@@ -528,7 +533,7 @@ Result pfsp_multigpu(int inst, const std::string& lb_str, int ub, int m, int M, 
           HIP_CHECK_M(
               hipHostMalloc(reinterpret_cast<void**>(&parents_h), M * sizeof(PFSPNode)));
           ws_worker<PFSPNode>(
-              id, D, m, M, pools, states, all_idle_flag,
+              id, D, m, M, perc, pools, states, all_idle_flag,
               [&](const PFSPNode* parents, size_t n, ParPool<PFSPNode>& own,
                   WorkerDiag& dg) {
                 std::memcpy(parents_h, parents, n * sizeof(PFSPNode));
@@ -552,7 +557,7 @@ Result pfsp_multigpu(int inst, const std::string& lb_str, int ub, int m, int M, 
         } else {
           std::vector<int32_t> bounds(static_cast<size_t>(M) * I.jobs);
           ws_worker<PFSPNode>(
-              id, D, m, M, pools, states, all_idle_flag,
+              id, D, m, M, perc, pools, states, all_idle_flag,
               [&](const PFSPNode* parents, size_t n, ParPool<PFSPNode>& own,
                   WorkerDiag& dg) {
                 // CPU evaluator: same bounds the GPU kernels produce

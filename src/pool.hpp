@@ -154,15 +154,22 @@ class ParPool {
   }
   size_t popBackBulkFree(size_t m, size_t M, T* out) { return inner_.popBackBulk(m, M, out); }
 
-  // Steal half the victim's nodes from the FRONT (oldest = shallowest = biggest
-  // subtrees), only when the victim holds >= 2m nodes (Pool_par.chpl:180-191).
-  // Caller must hold the lock.
-  size_t popFrontHalfFree(size_t m, std::vector<T>& out) {
+  // Steal a FRACTION of the victim's nodes from the FRONT (oldest = shallowest
+  // = biggest subtrees), only when the victim holds >= 2m nodes — the
+  // reference's configurable steal percentage (popFrontBulkFree perc,
+  // pfsp_multigpu_cuda.c:539 / Pool_ext.c:138-147; Chapel fixes perc=0.5,
+  // Pool_par.chpl:180-191). Caller must hold the lock.
+  size_t popFrontFracFree(size_t m, double perc, std::vector<T>& out) {
     size_t sz = inner_.size();
     if (sz < 2 * m) return 0;
-    size_t n = sz / 2;
+    size_t n = static_cast<size_t>(sz * perc);
+    if (n == 0) return 0;
     out.resize(n);
     return inner_.popFrontBulk(n, out.data());
+  }
+
+  size_t popFrontHalfFree(size_t m, std::vector<T>& out) {
+    return popFrontFracFree(m, 0.5, out);
   }
 
   // Steal half from the BACK (Pool_par.chpl:153-165 popBackBulkFree(half)).

@@ -53,3 +53,26 @@ def test_pfsp_lb1_multipool_matches_seq(core):
     seq = core.pfsp_seq(2, "lb1", 0)
     r = core.pfsp_multigpu(2, "lb1", 0, 5, 128, 3, "cpu", True)
     assert r["optimum"] == seq["optimum"] == 1359
+
+
+def test_steal_fraction_perc(core):
+    # reference --perc: steal size*perc from the victim's front
+    # (pfsp_multigpu_cuda.c:126,539; Pool_ext.c:138-147). Counts must stay
+    # exact for any valid fraction.
+    seq = core.nqueens_seq(11, 1)
+    for perc in (0.1, 0.25, 0.75, 0.9):
+        r = core.nqueens_multigpu(11, 1, 5, 64, 4, "cpu", perc)
+        assert r["tree"] == seq["tree"]
+        assert r["sol"] == seq["sol"]
+    seq = core.pfsp_seq(14, "lb1_d", 1)
+    r = core.pfsp_multigpu(14, "lb1_d", 1, 25, 2000, 4, "cpu", False, 0.25)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+
+
+def test_steal_fraction_validation(core):
+    import pytest
+
+    for bad in (0.0, 1.0, -0.5, 1.5):
+        with pytest.raises(ValueError):
+            core.nqueens_multigpu(8, 1, 5, 64, 2, "cpu", bad)
