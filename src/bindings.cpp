@@ -112,7 +112,13 @@ py::tuple nq_bfs_frontier(int N, int g, size_t target) {
   Pool<NQNode> pool;
   pool.pushBack(nq_root());
   uint64_t tree = 0, sol = 0;
-  nq_bfs_until(N, g, target, pool, tree, sol);
+  // big frontiers (dist tier: every rank builds this redundantly) go through
+  // the parallel level-synchronous builder; small ones keep the serial
+  // popFront order
+  if (target >= 16384)
+    nq_bfs_level(N, g, target, pool, tree, sol);
+  else
+    nq_bfs_until(N, g, target, pool, tree, sol);
   return py::make_tuple(nodes_to_bytes(pool.data(), pool.size()), tree, sol);
 }
 

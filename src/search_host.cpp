@@ -2,7 +2,10 @@
 
 #include <chrono>
 #include <climits>
+#include <cstring>
 #include <stdexcept>
+#include <thread>
+#include <vector>
 
 #include "taillard.hpp"
 
@@ -79,6 +82,48 @@ void nq_bfs_until(int N, int g, size_t target, Pool<NQNode>& pool, uint64_t& tre
     if (!pool.popFront(parent)) break;
     nq_decompose(parent, N, g, tree, sol, pool);
   }
+}
+
+void nq_bfs_level(int N, int g, size_t target, Pool<NQNode>& pool, uint64_t& tree,
+                  uint64_t& sol) {
+  nq_check(N, g);
+  std::vector<NQNode> level(pool.size());
+  std::memcpy(level.data(), pool.data(), pool.size() * sizeof(NQNode));
+  pool.clear();
+  while (!level.empty() && level.size() < target) {
+    const size_t n = level.size();
+    unsigned T = std::thread::hardware_concurrency();
+    if (T == 0) T = 1;
+    if (T > 16) T = 16;
+    if (n < 4096) T = 1;  // thread spawn not worth it on shallow levels
+    std::vector<std::vector<NQNode>> childv(T);
+    std::vector<uint64_t> tcnt(T, 0), scnt(T, 0);
+    auto work = [&](unsigned t) {
+      const size_t lo = n * t / T, hi = n * (t + 1) / T;
+      Pool<NQNode> local;
+      for (size_t i = lo; i < hi; i++) nq_decompose(level[i], N, g, tcnt[t], scnt[t], local);
+      childv[t].assign(local.data(), local.data() + local.size());
+    };
+    if (T == 1) {
+      work(0);
+    } else {
+      std::vector<std::thread> th;
+      for (unsigned t = 1; t < T; t++) th.emplace_back(work, t);
+      work(0);
+      for (auto& x : th) x.join();
+    }
+    size_t total = 0;
+    for (unsigned t = 0; t < T; t++) {
+      total += childv[t].size();
+      tree += tcnt[t];
+      sol += scnt[t];
+    }
+    level.clear();
+    level.reserve(total);
+    for (unsigned t = 0; t < T; t++)
+      level.insert(level.end(), childv[t].begin(), childv[t].end());
+  }
+  if (!level.empty()) pool.pushBackBulk(level.data(), level.size());
 }
 
 void nq_generate_children(const NQNode* parents, size_t n, int N, const uint8_t* labels,

@@ -56,6 +56,15 @@ Result nqueens_seq(int N, int g);
 void nq_bfs_until(int N, int g, size_t target, Pool<NQNode>& pool, uint64_t& tree,
                   uint64_t& sol);
 
+// Deterministic PARALLEL level-synchronous BFS: expands whole depth levels
+// with up to 16 host threads until the frontier reaches `target` (may
+// overshoot by one level) or the tree is exhausted. The frontier is identical
+// for any thread count (children concatenated in parent order), so every rank
+// of the distributed tier computes the same partition redundantly
+// (pfsp_dist_multigpu_cuda.c:372-378's trick) without the serial-BFS cost.
+void nq_bfs_level(int N, int g, size_t target, Pool<NQNode>& pool, uint64_t& tree,
+                  uint64_t& sol);
+
 // ---------------- PFSP ----------------
 
 struct PfspInstance {

@@ -91,3 +91,28 @@ def test_pfsp_ub0_reaches_known_optimum(core, inst, lb):
         inst = 2
     r = core.pfsp_seq(inst, lb, 0)
     assert r["optimum"] == core.taillard_best_ub(inst)
+
+
+def test_nq_parallel_frontier_deterministic(core):
+    # target >= 16384 routes through the parallel level-synchronous BFS
+    # (search_host.cpp nq_bfs_level): frontier bytes must be identical across
+    # calls (any thread count) and totals must match the sequential search
+    a = core.nq_bfs_frontier(13, 1, 20000)
+    b = core.nq_bfs_frontier(13, 1, 20000)
+    assert a == b
+    nodes, tree1, sol1 = a
+    assert len(nodes) % NODE == 0 and len(nodes) // NODE >= 20000
+    seq = core.nqueens_seq(13, 1)
+    r = core.nqueens_seq_from_pool(nodes, 13, 1)
+    assert tree1 + r["tree"] == seq["tree"]
+    assert sol1 + r["sol"] == seq["sol"]
+
+
+def test_nq_parallel_frontier_exhausted_tree(core):
+    # target larger than the whole tree: BFS runs dry; frontier is empty and
+    # the counts alone must equal the full search
+    nodes, tree1, sol1 = core.nq_bfs_frontier(6, 1, 1 << 20)
+    seq = core.nqueens_seq(6, 1)
+    assert len(nodes) == 0
+    assert tree1 == seq["tree"]
+    assert sol1 == seq["sol"]
