@@ -87,21 +87,29 @@ void nq_bfs_until(int N, int g, size_t target, Pool<NQNode>& pool, uint64_t& tre
 void nq_bfs_level(int N, int g, size_t target, Pool<NQNode>& pool, uint64_t& tree,
                   uint64_t& sol) {
   nq_check(N, g);
-  std::vector<NQNode> level(pool.size());
-  std::memcpy(level.data(), pool.data(), pool.size() * sizeof(NQNode));
+  // Serial BFS with BLOCK-granular pops: take up to BLOCK nodes off the
+  // front, expand them in parallel (fixed chunk boundaries, children
+  // concatenated in chunk order -> identical frontier for any thread count),
+  // append the children at the back. Overshoot past `target` is bounded by
+  // one block's children (vs whole-level sync, which overshot 5-7x on the
+  // geometric N-Queens levels).
+  constexpr size_t BLOCK = 8192;
+  std::vector<NQNode> deque(pool.size());
+  std::memcpy(deque.data(), pool.data(), pool.size() * sizeof(NQNode));
   pool.clear();
-  while (!level.empty() && level.size() < target) {
-    const size_t n = level.size();
+  size_t head = 0;
+  while (deque.size() - head != 0 && deque.size() - head < target) {
+    const size_t n = std::min(BLOCK, deque.size() - head);
     unsigned T = std::thread::hardware_concurrency();
     if (T == 0) T = 1;
     if (T > 16) T = 16;
-    if (n < 4096) T = 1;  // thread spawn not worth it on shallow levels
+    if (n < 2048) T = 1;  // thread spawn not worth it on small blocks
     std::vector<std::vector<NQNode>> childv(T);
     std::vector<uint64_t> tcnt(T, 0), scnt(T, 0);
     auto work = [&](unsigned t) {
-      const size_t lo = n * t / T, hi = n * (t + 1) / T;
+      const size_t lo = head + n * t / T, hi = head + n * (t + 1) / T;
       Pool<NQNode> local;
-      for (size_t i = lo; i < hi; i++) nq_decompose(level[i], N, g, tcnt[t], scnt[t], local);
+      for (size_t i = lo; i < hi; i++) nq_decompose(deque[i], N, g, tcnt[t], scnt[t], local);
       childv[t].assign(local.data(), local.data() + local.size());
     };
     if (T == 1) {
@@ -112,18 +120,19 @@ void nq_bfs_level(int N, int g, size_t target, Pool<NQNode>& pool, uint64_t& tre
       work(0);
       for (auto& x : th) x.join();
     }
-    size_t total = 0;
+    head += n;
+    if (head > (1u << 20) && head > deque.size() - head) {  // compact dead front
+      deque.erase(deque.begin(), deque.begin() + head);
+      head = 0;
+    }
     for (unsigned t = 0; t < T; t++) {
-      total += childv[t].size();
       tree += tcnt[t];
       sol += scnt[t];
+      deque.insert(deque.end(), childv[t].begin(), childv[t].end());
     }
-    level.clear();
-    level.reserve(total);
-    for (unsigned t = 0; t < T; t++)
-      level.insert(level.end(), childv[t].begin(), childv[t].end());
   }
-  if (!level.empty()) pool.pushBackBulk(level.data(), level.size());
+  if (deque.size() - head != 0)
+    pool.pushBackBulk(deque.data() + head, deque.size() - head);
 }
 
 void nq_generate_children(const NQNode* parents, size_t n, int N, const uint8_t* labels,
