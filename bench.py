@@ -45,24 +45,36 @@ def main():
     args = parse_args()
     c = gats_amd.core()
     ndev = c.gpu_device_count()
-    if ndev == 0:
+    # GATS_BENCH_CPU=1: run the whole bench harness (torchrun rendezvous,
+    # dynamic slicing, reductions, JSON contract) with the CPU evaluator so
+    # CI covers the exact multi-rank path the driver launches on GPU nodes
+    cpu_mode = ndev == 0 and os.environ.get("GATS_BENCH_CPU") == "1"
+    if ndev == 0 and not cpu_mode:
         raise RuntimeError("bench.py needs an MI355X (no HIP device visible)")
-    local = int(os.environ.get("LOCAL_RANK", "0")) % ndev
-    torch.cuda.set_device(local)  # before NCCL/RCCL process-group init
+    if not cpu_mode:
+        local = int(os.environ.get("LOCAL_RANK", "0")) % ndev
+        torch.cuda.set_device(local)  # before NCCL/RCCL process-group init
+    else:
+        local = 0
     rank, world = gdist.init_dist()
     if world != args.gpus and rank == 0:
         print(f"# note: --gpus {args.gpus} but world_size {world}; using world_size",
               file=sys.stderr)
 
+    engine = "cpu" if cpu_mode else "gpu"
+
     def step():
         if args.problem == "nqueens":
-            return gdist.run_nqueens(args.N, args.g, args.m, args.M, args.mode)
-        return gdist.run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode)
+            return gdist.run_nqueens(args.N, args.g, args.m, args.M, args.mode,
+                                     engine=engine)
+        return gdist.run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode,
+                              engine=engine)
 
     def sync():
         if world > 1:
             torch.distributed.barrier()
-        torch.cuda.synchronize()
+        if not cpu_mode:
+            torch.cuda.synchronize()
 
     nodes_per_step = None
     for _ in range(args.warmup):
