@@ -155,11 +155,11 @@ def run_nqueens(N, g=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
     c = gats_amd.core()
     rank, world = init_dist()
     if frontier_target is None:
-        # scale with world so each dynamic sub-slice stays big enough for the
-        # engine's multi-slice mode (a 2048-node sub-slice forces S=1); the
-        # redundant per-rank BFS is cheap — nq_bfs_frontier switches to the
-        # parallel level-synchronous builder at this size
-        frontier_target = max(65536, 32768 * world)
+        # 65536 regardless of world: the parallel block-BFS builds it in ~8 ms
+        # (every rank redundantly, per step), and the engine's deep-frontier
+        # rule keeps even 2048-node dynamic sub-slices fully sliced, so a
+        # bigger (costlier) frontier buys nothing
+        frontier_target = 65536
     nodes, tree1, sol1 = c.nq_bfs_frontier(N, g, frontier_target)
     phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
     local = rank % max(1, c.gpu_device_count())

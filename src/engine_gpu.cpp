@@ -782,7 +782,15 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
     if (const char* e = std::getenv("GATS_NQ_FINISH")) finish = atoi(e);
     if (finish > 8) finish = 8;  // template recursion budget (NQ_FINISH_MAX)
     int S = devpool_slices();
-    while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;  // small search
+    // a frontier of DEEP nodes explodes immediately (a 2048-node N=17 dist
+    // sub-slice carries billion-node subtrees), so it deserves full slicing
+    // no matter how small the pool is; only genuinely small searches (the
+    // remaining levels bound the subtree) shrink S to skip slicing overhead
+    int maxd = 0;
+    for (size_t i = 0; i < pool.size(); i++)
+      maxd = std::max(maxd, static_cast<int>(pool.data()[i].depth));
+    if (N - maxd < 10)
+      while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;  // small search
     const int NS = (S == 1) ? 1 : S * 4;  // oversubscribe: ~4 queued slices/thread
     std::vector<std::vector<NQNode>> slices(NS);
     {
@@ -908,7 +916,11 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
     if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
       throw std::invalid_argument("devpool requires M * jobs <= 2^31");
     int S = devpool_slices();
-    while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;
+    int maxd = 0;  // same deep-frontier rule as N-Queens (12+ open jobs)
+    for (size_t i = 0; i < pool.size(); i++)
+      maxd = std::max(maxd, static_cast<int>(pool.data()[i].depth));
+    if (jobs - maxd < 12)
+      while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;
     const int NS = (S == 1) ? 1 : S * 4;
     std::vector<std::vector<PFSPNode>> slices(NS);
     {
