@@ -29,19 +29,19 @@ def _phases(r):
     names = ["Initial search on CPU", "Search on GPU", "Search on CPU"]
     for name, p in zip(names, r["phases"]):
         print(f"\n{name} completed")
-        print("Size of the explored tree: ", p["tree"])
-        print("Number of explored solutions: ", p["sol"])
-        print("Elapsed time: ", p["time"], " [s]")
+        print("Size of the explored tree:", p["tree"])
+        print("Number of explored solutions:", p["sol"])
+        print(f"Elapsed time: {p['time']:.4f} [s]")
 
 
 def _results(r, optimum=None, init_ub=None):
     print("\n=================================================")
-    print("Size of the explored tree: ", r["tree"])
-    print("Number of explored solutions: ", r["sol"])
+    print("Size of the explored tree:", r["tree"])
+    print("Number of explored solutions:", r["sol"])
     if optimum is not None:
         tag = " (improved)" if optimum < init_ub else " (not improved)"
-        print("Optimal makespan: ", optimum, tag)
-    print("Elapsed time: ", r["time"], " [s]")
+        print(f"Optimal makespan: {optimum}{tag}")
+    print(f"Elapsed time: {r['time']:.4f} [s]")
     print("=================================================\n")
 
 

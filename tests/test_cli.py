@@ -16,8 +16,8 @@ def test_nqueens_seq_output_contract(core):
     rc, out = run_cli(["nqueens", "--N", "10", "--tier", "seq"])
     assert rc == 0
     assert "Resolution of the 10-Queens instance" in out
-    assert "Size of the explored tree:  35538" in out
-    assert "Number of explored solutions:  724" in out
+    assert "Size of the explored tree: 35538" in out
+    assert "Number of explored solutions: 724" in out
     assert "Elapsed time:" in out
     assert "Exploration terminated." in out
 
@@ -31,7 +31,7 @@ def test_pfsp_seq_output_contract(core, tmp_path):
     assert "Initial upper bound: inf" in out
     assert "Lower bound function: lb2" in out
     assert "Branching rule: fwd" in out
-    assert "Optimal makespan:  1359  (improved)" in out
+    assert "Optimal makespan: 1359 (improved)" in out
     line = stats.read_text().strip()
     assert line.startswith("ta2 lb2 SEQ ") and line.endswith(" 33110 59 1359")
 
