@@ -127,3 +127,19 @@ def test_pfsp_instance_breadth_lb2(gpu, inst):
     # profiles/taillard_sweep_lb2.txt); optimum must be proven from ub=1
     r = gpu.pfsp_gpu(inst, "lb2", 1, 25, 50000, 0, "devpool", 1 << 25)
     assert r["optimum"] == gpu.taillard_best_ub(inst)
+
+
+def test_deep_frontier_small_slice_counts(gpu):
+    # a SMALL frontier of DEEP nodes (the dist tier's 2048-node dynamic
+    # sub-slices) keeps full multi-slice mode (engine_gpu.cpp deep-frontier
+    # rule) and must still produce exact counts
+    seq = gpu.nqueens_seq(14, 1)
+    nodes, tree1, sol1 = gpu.nq_bfs_frontier(14, 1, 2048)
+    r = gpu.nqueens_gpu_from_pool(nodes, 14, 1, 25, 50000, 0, "devpool", 1 << 24)
+    assert tree1 + r["tree"] == seq["tree"]
+    assert sol1 + r["sol"] == seq["sol"]
+    s = gpu.pfsp_seq(14, "lb1_d", 1)
+    pn, pt, ps, best = gpu.pfsp_bfs_frontier(14, "lb1_d", 1, 512)
+    p = gpu.pfsp_gpu_from_pool(pn, 14, "lb1_d", 1, best, 25, 50000, 0, "devpool", 1 << 24)
+    assert pt + p["tree"] == s["tree"]
+    assert ps + p["sol"] == s["sol"]
