@@ -13,7 +13,7 @@ from gats_amd import dist as gdist  # noqa: E402
 
 def main():
     rank, world = gdist.init_dist()
-    assert world == 2, world
+    assert world >= 2, world
     c = gats_amd.core()
 
     r = gdist.run_nqueens(11, engine="cpu", frontier_target=4096)

@@ -17,13 +17,19 @@ def free_port():
     return p
 
 
-def test_dist_gloo_world2():
+import pytest
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_dist_gloo(world):
+    # world 3: uneven dynamic-queue claims (12 sub-slices over 3 ranks) and a
+    # remainder-bearing round-robin partition
     env = dict(os.environ)
     env.pop("RANK", None)
     env.pop("WORLD_SIZE", None)
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
-        "--nnodes=1", "--nproc-per-node=2",
+        "--nnodes=1", f"--nproc-per-node={world}",
         "--master-addr", "127.0.0.1", "--master-port", str(free_port()),
         os.path.join(HERE, "helpers", "dist_check.py"),
     ]
