@@ -264,15 +264,21 @@ def run_pfsp_shared_ub(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
 
 def run_from_cli(args):
     """Entry for `gats-amd ... --tier dist` under torchrun; rank 0 returns the
-    combined stats dict, other ranks return None."""
+    combined stats dict, other ranks return None. GATS_DIST_ENGINE=cpu runs
+    the tier with the CPU evaluator (CI coverage of the full CLI dist path
+    without a GPU; the ub=0 mid-search UB exchange needs the GPU engine and
+    falls back to the end-of-search min-reduce)."""
     rank, world = init_dist()
+    engine = os.environ.get("GATS_DIST_ENGINE", "gpu")
     if args.problem == "nqueens":
-        r = run_nqueens(args.N, args.g, args.m, args.M, args.mode, args.capacity)
-    elif args.ub == 0:
+        r = run_nqueens(args.N, args.g, args.m, args.M, args.mode, args.capacity,
+                        engine=engine)
+    elif args.ub == 0 and engine == "gpu":
         # open upper bound: exchange the incumbent over RCCL during the search
         r = run_pfsp_shared_ub(args.inst, args.lb, args.ub, args.m, args.M, args.capacity)
     else:
-        r = run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode, args.capacity)
+        r = run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode, args.capacity,
+                     engine=engine)
     if world > 1:
         td.barrier()
     return r if rank == 0 else None

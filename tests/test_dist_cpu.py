@@ -37,3 +37,24 @@ def test_dist_gloo(world):
                        cwd=os.path.dirname(HERE))
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
     assert "DIST_CHECK_OK" in r.stdout
+
+
+def test_cli_dist_tier_cpu():
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    env["GATS_DIST_ENGINE"] = "cpu"
+    env["GATS_DIST_BACKEND"] = "gloo"
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node=2",
+        "--master-addr", "127.0.0.1", "--master-port", str(free_port()),
+        os.path.join(HERE, "helpers", "cli_dist_check.py"),
+    ]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env,
+                       cwd=os.path.dirname(HERE))
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "CLI_DIST_OK" in r.stdout
+    assert "Size of the explored tree: 35538" in r.stdout   # N=10 exact
+    assert "Size of the explored tree: 2573652" in r.stdout  # ta014 lb1_d ub1
+    assert "Optimal makespan: 1377 (not improved)" in r.stdout
