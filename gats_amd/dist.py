@@ -169,7 +169,10 @@ def run_nqueens(N, g=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
             return c.nqueens_gpu_from_pool(sl, N, g, m, M, local, mode, capacity)
         return c.nqueens_seq_from_pool(sl, N, g)  # CPU path for gloo CI
 
-    r = _run_dynamic(nodes, world, 4, run_slice)
+    # 2 claims per rank: N-Queens round-robin slices are statistically uniform,
+    # and every claim pays the engine spin-up (~2-3 ms) — PFSP (pruning-driven
+    # imbalance) keeps 4
+    r = _run_dynamic(nodes, world, 2, run_slice)
     if world == 1:
         r = dict(r)
         r["tree"] += phase1["tree"]
