@@ -4,20 +4,11 @@ slicing, MAX-over-ranks reduction and the one-line JSON contract are CI-tested
 without a GPU."""
 import json
 import os
-import socket
 import subprocess
 import sys
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 ROOT = os.path.dirname(HERE)
-
-
-def free_port():
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    p = s.getsockname()[1]
-    s.close()
-    return p
 
 
 def run_bench(nproc, extra):
@@ -28,7 +19,7 @@ def run_bench(nproc, extra):
     env["GATS_DIST_BACKEND"] = "gloo"
     cmd = [sys.executable, "-m", "torch.distributed.run",
            "--nnodes=1", f"--nproc-per-node={nproc}",
-           "--master-addr", "127.0.0.1", "--master-port", str(free_port()),
+           "--standalone", "--local-addr", "127.0.0.1",
            os.path.join(ROOT, "bench.py"), "--gpus", str(nproc)] + extra
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env,
                        cwd=ROOT)

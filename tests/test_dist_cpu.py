@@ -2,19 +2,10 @@
 Verifies the redundant-BFS + round-robin-slice partition and the end-of-search
 collectives reproduce sequential counts exactly."""
 import os
-import socket
 import subprocess
 import sys
 
 HERE = os.path.dirname(os.path.abspath(__file__))
-
-
-def free_port():
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    p = s.getsockname()[1]
-    s.close()
-    return p
 
 
 import pytest
@@ -30,7 +21,7 @@ def test_dist_gloo(world):
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", f"--nproc-per-node={world}",
-        "--master-addr", "127.0.0.1", "--master-port", str(free_port()),
+        "--standalone", "--local-addr", "127.0.0.1",
         os.path.join(HERE, "helpers", "dist_check.py"),
     ]
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env,
@@ -48,7 +39,7 @@ def test_cli_dist_tier_cpu():
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", "--nproc-per-node=2",
-        "--master-addr", "127.0.0.1", "--master-port", str(free_port()),
+        "--standalone", "--local-addr", "127.0.0.1",
         os.path.join(HERE, "helpers", "cli_dist_check.py"),
     ]
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env,
