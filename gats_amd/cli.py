@@ -150,7 +150,7 @@ def main(argv=None):
         tier_name = {"seq": "Sequential", "gpu": "Single-GPU", "multigpu": "Multi-GPU",
                      "dist": "Distributed multi-GPU"}[args.tier]
         _banner(f"{tier_name} MI355X (gats-amd)",
-                [f"Resolution of PFSP Taillard's instance: ta{inst:03d} "
+                [f"Resolution of PFSP Taillard's instance: ta{inst} "
                  f"(m = {machines}, n = {jobs})",
                  "Initial upper bound: " + ("opt" if args.ub == 1 else "inf"),
                  f"Lower bound function: {args.lb}",

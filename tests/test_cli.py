@@ -27,7 +27,7 @@ def test_pfsp_seq_output_contract(core, tmp_path):
     rc, out = run_cli(["pfsp", "--inst", "2", "--lb", "lb2", "--ub", "0",
                        "--tier", "seq", "--stats-file", str(stats)])
     assert rc == 0
-    assert "Resolution of PFSP Taillard's instance: ta002 (m = 5, n = 20)" in out
+    assert "Resolution of PFSP Taillard's instance: ta2 (m = 5, n = 20)" in out
     assert "Initial upper bound: inf" in out
     assert "Lower bound function: lb2" in out
     assert "Branching rule: fwd" in out
