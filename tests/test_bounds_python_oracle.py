@@ -113,3 +113,43 @@ def test_lb2_vs_python(core):
             child[depth], child[k] = child[k], child[depth]
             expect = py_lb2_bound(ptm, jobs, machines, child, limit1 + 1, mt, best)
             assert cpp[idx * jobs + k] == expect, (idx, k)
+
+
+def py_lb1d_child_bound(ptm, jobs, machines, prmu, limit1, min_tails, job):
+    # DIRECT recomputation of the child bound for appending `job` to the
+    # prefix — deliberately a different algorithm than the oracle's
+    # incremental lb1_children_bounds/add_front_and_bound pass
+    # (c_bound_simple.c:160-258), same math
+    front = [0] * machines
+    for i in range(limit1 + 1):
+        q = prmu[i]
+        front[0] += ptm[q]
+        for m in range(1, machines):
+            front[m] = max(front[m - 1], front[m]) + ptm[m * jobs + q]
+    front[0] += ptm[job]
+    for m in range(1, machines):
+        front[m] = max(front[m - 1], front[m]) + ptm[m * jobs + job]
+    scheduled = set(prmu[:limit1 + 1]) | {job}
+    remain = [sum(ptm[m * jobs + q] for q in range(jobs) if q not in scheduled)
+              for m in range(machines)]
+    tmp0 = front[0] + remain[0]
+    lb = tmp0 + min_tails[0]
+    for m in range(1, machines):
+        tmp1 = max(tmp0, front[m] + remain[m])
+        lb = max(lb, tmp1 + min_tails[m])
+        tmp0 = tmp1
+    return lb
+
+
+def test_lb1d_vs_python(core):
+    for inst in (2, 14):
+        jobs, machines = core.taillard_nb_jobs(inst), core.taillard_nb_machines(inst)
+        ptm = core.taillard_processing_times(inst)
+        mt = py_min_tails(ptm, jobs, machines)
+        nodes, _, _, best = core.pfsp_bfs_frontier(inst, "lb1_d", 1, 256)
+        cpp = core.pfsp_cpu_bounds(inst, "lb1_d", nodes, best)
+        for idx, (depth, limit1, prmu) in enumerate(nodes_of(nodes)):
+            for k in range(limit1 + 1, jobs):
+                expect = py_lb1d_child_bound(ptm, jobs, machines, prmu, limit1, mt,
+                                             prmu[k])
+                assert cpp[idx * jobs + k] == expect, (inst, idx, k)
