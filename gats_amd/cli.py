@@ -57,12 +57,19 @@ def _workload(r):
 
 def _stats_line(path, args, r):
     """Append-only results ledger, reference line format parity
-    (pfsp_gpu_cuda.c:140-148: `ta%d lb%d S-GPU %.4f %llu %llu %d`)."""
+    (pfsp_gpu_cuda.c:140-148: `ta%d lb%d S-GPU %.4f %llu %llu %d`). The
+    reference only writes this for PFSP; we honor the flag for N-Queens too
+    (same shape, `N%d g%d` in place of the instance/lb fields) rather than
+    silently dropping it."""
     tier_tag = {"seq": "SEQ", "gpu": "S-GPU", "multigpu": "M-GPU", "dist": "D-GPU"}[args.tier]
-    lb_num = {"lb1": 1, "lb1_d": 0, "lb2": 2}[args.lb]
     with open(path, "a") as f:
-        f.write(f"ta{args.inst} lb{lb_num} {tier_tag} {r['time']:.4f} "
-                f"{r['tree']} {r['sol']} {r['optimum']}\n")
+        if args.problem == "pfsp":
+            lb_num = {"lb1": 1, "lb1_d": 0, "lb2": 2}[args.lb]
+            f.write(f"ta{args.inst} lb{lb_num} {tier_tag} {r['time']:.4f} "
+                    f"{r['tree']} {r['sol']} {r['optimum']}\n")
+        else:
+            f.write(f"N{args.N} g{args.g} {tier_tag} {r['time']:.4f} "
+                    f"{r['tree']} {r['sol']}\n")
 
 
 def _diag(r):
@@ -132,7 +139,7 @@ def main(argv=None):
         elif args.tier == "multigpu":
             gats_amd.require_gpu()
             r = c.nqueens_multigpu(args.N, args.g, args.m, args.M, args.D, "devpool",
-                                   args.perc / 100.0)
+                                   args.perc / 100.0, args.capacity)
             _phases(r)
         else:
             from gats_amd import dist
@@ -143,6 +150,8 @@ def main(argv=None):
         _results(r)
         _workload(r)
         _diag(r)
+        if args.stats_file:
+            _stats_line(args.stats_file, args, r)
     else:
         inst = args.inst
         jobs, machines = c.taillard_nb_jobs(inst), c.taillard_nb_machines(inst)
@@ -165,7 +174,7 @@ def main(argv=None):
         elif args.tier == "multigpu":
             gats_amd.require_gpu()
             r = c.pfsp_multigpu(inst, args.lb, args.ub, args.m, args.M, args.D, "devpool",
-                                False, args.perc / 100.0)
+                                False, args.perc / 100.0, args.capacity)
             _phases(r)
         else:
             from gats_amd import dist

@@ -230,29 +230,33 @@ PYBIND11_MODULE(_core, mod) {
   mod.def("gpu_device_count", &gpu_device_count);
 
   mod.def("nqueens_multigpu",
-          [](int N, int g, int m, int M, int D, const std::string& eval, double perc) {
+          [](int N, int g, int m, int M, int D, const std::string& eval, double perc,
+             unsigned long long capacity) {
             Result r;
             {
               py::gil_scoped_release rel;
-              r = nqueens_multigpu(N, g, m, M, D, eval, perc);
+              r = nqueens_multigpu(N, g, m, M, D, eval, perc, capacity);
             }
             return result_to_dict(r);
           },
           py::arg("N") = 14, py::arg("g") = 1, py::arg("m") = 25, py::arg("M") = 50000,
-          py::arg("D") = 1, py::arg("eval") = "gpu", py::arg("perc") = 0.5);
+          py::arg("D") = 1, py::arg("eval") = "gpu", py::arg("perc") = 0.5,
+          py::arg("capacity") = (1ull << 27));
   mod.def("pfsp_multigpu",
           [](int inst, const std::string& lb, int ub, int m, int M, int D,
-             const std::string& eval, bool share_best, double perc) {
+             const std::string& eval, bool share_best, double perc,
+             unsigned long long capacity) {
             Result r;
             {
               py::gil_scoped_release rel;
-              r = pfsp_multigpu(inst, lb, ub, m, M, D, eval, share_best, perc);
+              r = pfsp_multigpu(inst, lb, ub, m, M, D, eval, share_best, perc, capacity);
             }
             return result_to_dict(r);
           },
           py::arg("inst") = 14, py::arg("lb") = "lb1", py::arg("ub") = 1, py::arg("m") = 25,
           py::arg("M") = 50000, py::arg("D") = 1, py::arg("eval") = "gpu",
-          py::arg("share_best") = false, py::arg("perc") = 0.5);
+          py::arg("share_best") = false, py::arg("perc") = 0.5,
+          py::arg("capacity") = (1ull << 27));
 
   mod.def("nqueens_gpu",
           [](int N, int g, int m, int M, int device, const std::string& mode,

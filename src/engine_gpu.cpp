@@ -12,7 +12,7 @@
 //   "devpool": MI355X-native fast path — pools live in HBM3E; each offload
 //       round is an expand-compact + gather kernel pair (kernels.hip), with
 //       ~16 frontier slices pulled off a queue by 4 worker threads on
-//       concurrent streams to fill the chip; the host polls 64 B control
+//       concurrent streams to fill the chip; the host polls 48 B control
 //       blocks every few iterations.
 #include <hip/hip_runtime.h>
 
@@ -275,7 +275,7 @@ int lbk_of(LbKind lb) {
 
 
 // Shared devpool driver: capture BATCH iterations into a hipGraph once, then
-// replay + poll the 64 B control block until the pool drops below m
+// replay + poll the 48 B control block until the pool drops below m
 // (graph replay ~10-16 us vs ~3.5 us host cost PER LAUNCH eager — the hot
 // loop is launch-bound at chunk sizes this small).
 // `shared_best` (optional): a cross-thread incumbent. Each readback publishes
@@ -632,11 +632,12 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
 
 static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& slices,
                                     std::atomic<int>& next_slice, const PfspInstance& I,
-                                    const PfspDevTables& tb, int lbk, int best0,
+                                    int lbk, int best0,
                                     int m, int M, int device, unsigned long long capacity,
                                     std::atomic<int>* shared_best, bool allow_graph,
                                     SliceShare* share, std::vector<PFSPNode>& leftover) {
   HIP_CHECK(hipSetDevice(device));
+  const PfspDevTables& tb = pfsp_tables_cached(I, device);
   StreamGuard stream;
   SliceOut out;
   out.fin.best = best0;
@@ -734,6 +735,148 @@ static void merge_slice_diag(Result& r, const Result& d) {
 }
 
 // ---------------------------------------------------------------------------
+// Multi-device devpool core (declared in engine_gpu.hpp): one shared slice
+// queue across ALL workers' threads. A worker whose slices finish early just
+// keeps claiming — the queue is the cross-device balancer (replacing the
+// static per-worker partition the reference's own CUDA multi-GPU baseline
+// uses, nqueens_multigpu_cuda.cu:268-277, which its README flags unstable).
+// Donation (SliceShare) stays within a worker group: the half-pool handoff is
+// a same-device D2D copy.
+// ---------------------------------------------------------------------------
+
+DevpoolMultiOut nq_devpool_multi(Pool<NQNode>& pool, int N, int g, int m, int M,
+                                 const std::vector<int>& devices,
+                                 unsigned long long capacity, Result& r) {
+  if (static_cast<unsigned long long>(M) * N > (1ull << 31))
+    throw std::invalid_argument("devpool requires M * N <= 2^31");
+  // depth of the in-thread bitmask subtree finisher (levels from the bottom)
+  int finish = 8;
+  if (const char* e = std::getenv("GATS_NQ_FINISH")) finish = atoi(e);
+  if (finish > 8) finish = 8;  // template recursion budget (NQ_FINISH_MAX)
+  const int D = static_cast<int>(devices.size());
+  int S = devpool_slices();
+  // a frontier of DEEP nodes explodes immediately (a 2048-node N=17 dist
+  // sub-slice carries billion-node subtrees), so it deserves full slicing
+  // no matter how small the pool is; only genuinely small searches (the
+  // remaining levels bound the subtree) shrink S to skip slicing overhead
+  int maxd = 0;
+  for (size_t i = 0; i < pool.size(); i++)
+    maxd = std::max(maxd, static_cast<int>(pool.data()[i].depth));
+  if (N - maxd < 10)
+    while (S > 1 && pool.size() < static_cast<size_t>(D) * S * 2048) S--;
+  const int T = D * S;
+  const int NS = (T == 1) ? 1 : T * 4;  // oversubscribe: ~4 queued slices/thread
+  std::vector<std::vector<NQNode>> slices(NS);
+  {
+    const NQNode* src = pool.data();
+    const size_t total = pool.size();
+    for (int t = 0; t < NS; t++) slices[t].reserve(total / NS + 1);
+    for (size_t i = 0; i < total; i++) slices[i % NS].push_back(src[i]);
+    pool.clear();
+  }
+  std::atomic<int> next_slice{0};
+  std::vector<SliceShare> shares(D);
+  std::vector<SliceOut> outs(T);
+  std::vector<std::vector<NQNode>> lefts(T);
+  std::vector<std::exception_ptr> errs(T);
+  std::vector<std::thread> threads;
+  const bool allow_graph = (T == 1);
+  for (int t = 0; t < T; t++) {
+    threads.emplace_back([&, t] {
+      try {
+        SliceShare* sh = (S > 1) ? &shares[t / S] : nullptr;
+        outs[t] = devpool_thread_nq(slices, next_slice, N, g, m, M, devices[t / S],
+                                    finish, capacity, allow_graph, sh, lefts[t]);
+      } catch (...) {
+        errs[t] = std::current_exception();
+      }
+    });
+  }
+  for (auto& th : threads) th.join();
+  for (auto& e : errs)
+    if (e) std::rethrow_exception(e);
+  DevpoolMultiOut o;
+  o.per_dev.assign(D, 0);
+  for (int t = 0; t < T; t++) {
+    o.tree += outs[t].fin.tree;
+    o.sol += outs[t].fin.sol;
+    o.per_dev[t / S] += outs[t].fin.tree;
+    merge_slice_diag(r, outs[t].diag);
+    if (!lefts[t].empty()) pool.pushBackBulk(lefts[t].data(), lefts[t].size());
+  }
+  return o;
+}
+
+DevpoolMultiOut pfsp_devpool_multi(const PfspInstance& I, Pool<PFSPNode>& pool, int lbk,
+                                   int best0, int m, int M,
+                                   const std::vector<int>& devices,
+                                   unsigned long long capacity,
+                                   std::atomic<int>* shared_best, Result& r) {
+  const int jobs = I.jobs;
+  if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
+    throw std::invalid_argument("devpool requires M * jobs <= 2^31");
+  const int D = static_cast<int>(devices.size());
+  int S = devpool_slices();
+  int maxd = 0;  // same deep-frontier rule as N-Queens (12+ open jobs)
+  for (size_t i = 0; i < pool.size(); i++)
+    maxd = std::max(maxd, static_cast<int>(pool.data()[i].depth));
+  if (jobs - maxd < 12)
+    while (S > 1 && pool.size() < static_cast<size_t>(D) * S * 2048) S--;
+  const int T = D * S;
+  const int NS = (T == 1) ? 1 : T * 4;
+  std::vector<std::vector<PFSPNode>> slices(NS);
+  {
+    const PFSPNode* src = pool.data();
+    const size_t total = pool.size();
+    for (int t = 0; t < NS; t++) slices[t].reserve(total / NS + 1);
+    for (size_t i = 0; i < total; i++) slices[i % NS].push_back(src[i]);
+    pool.clear();
+  }
+  // workers share the incumbent through this atomic even when no external
+  // one is plugged in (cross-worker pruning; identical counts at ub=1)
+  std::atomic<int> local_best{best0};
+  std::atomic<int>* sb = shared_best ? shared_best : &local_best;
+  std::atomic<int> next_slice{0};
+  std::vector<SliceShare> shares(D);
+  std::vector<SliceOut> outs(T);
+  std::vector<std::vector<PFSPNode>> lefts(T);
+  std::vector<std::exception_ptr> errs(T);
+  std::vector<std::thread> threads;
+  const bool allow_graph = (T == 1);
+  for (int t = 0; t < T; t++) {
+    threads.emplace_back([&, t] {
+      try {
+        SliceShare* sh = (S > 1) ? &shares[t / S] : nullptr;
+        outs[t] = devpool_thread_pfsp(slices, next_slice, I, lbk, best0, m, M,
+                                      devices[t / S], capacity, sb, allow_graph, sh,
+                                      lefts[t]);
+      } catch (...) {
+        errs[t] = std::current_exception();
+      }
+    });
+  }
+  for (auto& th : threads) th.join();
+  for (auto& e : errs)
+    if (e) std::rethrow_exception(e);
+  DevpoolMultiOut o;
+  o.best = best0;
+  o.per_dev.assign(D, 0);
+  for (int t = 0; t < T; t++) {
+    o.tree += outs[t].fin.tree;
+    o.sol += outs[t].fin.sol;
+    if (outs[t].fin.best < o.best) o.best = outs[t].fin.best;
+    o.per_dev[t / S] += outs[t].fin.tree;
+    merge_slice_diag(r, outs[t].diag);
+    if (!lefts[t].empty()) pool.pushBackBulk(lefts[t].data(), lefts[t].size());
+  }
+  {
+    const int sbv = sb->load(std::memory_order_relaxed);
+    if (sbv < o.best) o.best = sbv;
+  }
+  return o;
+}
+
+// ---------------------------------------------------------------------------
 // N-Queens
 // ---------------------------------------------------------------------------
 
@@ -747,13 +890,13 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
   uint64_t tree = tree0, sol = sol0;
 
   HIP_CHECK(hipSetDevice(device));
-  StreamGuard stream;
   const double t2 = now_sec();
 
   std::string mode_eff = mode;
   if (mode == "devpool" && N < 4) mode_eff = "hostpool";  // expand tiles assume N >= 4
 
   if (mode_eff == "hostpool") {
+    StreamGuard stream;
     PinnedGuard<NQNode> parents(M);
     PinnedGuard<uint8_t> labels(static_cast<size_t>(M) * N);
     DevGuard<NQNode> parents_d(M);
@@ -775,57 +918,9 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
       nq_generate_children(parents.p, n, N, labels.p, tree, sol, pool);
     }
   } else if (mode_eff == "devpool") {
-    if (static_cast<unsigned long long>(M) * N > (1ull << 31))
-      throw std::invalid_argument("devpool requires M * N <= 2^31");
-    // depth of the in-thread bitmask subtree finisher (levels from the bottom)
-    int finish = 8;
-    if (const char* e = std::getenv("GATS_NQ_FINISH")) finish = atoi(e);
-    if (finish > 8) finish = 8;  // template recursion budget (NQ_FINISH_MAX)
-    int S = devpool_slices();
-    // a frontier of DEEP nodes explodes immediately (a 2048-node N=17 dist
-    // sub-slice carries billion-node subtrees), so it deserves full slicing
-    // no matter how small the pool is; only genuinely small searches (the
-    // remaining levels bound the subtree) shrink S to skip slicing overhead
-    int maxd = 0;
-    for (size_t i = 0; i < pool.size(); i++)
-      maxd = std::max(maxd, static_cast<int>(pool.data()[i].depth));
-    if (N - maxd < 10)
-      while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;  // small search
-    const int NS = (S == 1) ? 1 : S * 4;  // oversubscribe: ~4 queued slices/thread
-    std::vector<std::vector<NQNode>> slices(NS);
-    {
-      const NQNode* src = pool.data();
-      const size_t total = pool.size();
-      for (int t = 0; t < NS; t++) slices[t].reserve(total / NS + 1);
-      for (size_t i = 0; i < total; i++) slices[i % NS].push_back(src[i]);
-      pool.clear();
-    }
-    std::atomic<int> next_slice{0};
-    SliceShare share;
-    SliceShare* sharep = (S > 1) ? &share : nullptr;
-    std::vector<SliceOut> outs(S);
-    std::vector<std::vector<NQNode>> lefts(S);
-    std::vector<std::exception_ptr> errs(S);
-    std::vector<std::thread> threads;
-    for (int t = 0; t < S; t++) {
-      threads.emplace_back([&, t] {
-        try {
-          outs[t] = devpool_thread_nq(slices, next_slice, N, g, m, M, device, finish,
-                                      capacity, S == 1, sharep, lefts[t]);
-        } catch (...) {
-          errs[t] = std::current_exception();
-        }
-      });
-    }
-    for (auto& th : threads) th.join();
-    for (auto& e : errs)
-      if (e) std::rethrow_exception(e);
-    for (int t = 0; t < S; t++) {
-      tree += outs[t].fin.tree;
-      sol += outs[t].fin.sol;
-      merge_slice_diag(r, outs[t].diag);
-      if (!lefts[t].empty()) pool.pushBackBulk(lefts[t].data(), lefts[t].size());
-    }
+    DevpoolMultiOut o = nq_devpool_multi(pool, N, g, m, M, {device}, capacity, r);
+    tree += o.tree;
+    sol += o.sol;
   } else {
     throw std::invalid_argument("mode must be hostpool or devpool");
   }
@@ -885,11 +980,11 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
   const int lbk = lbk_of(lb);
 
   HIP_CHECK(hipSetDevice(device));
-  StreamGuard stream;
-  const PfspDevTables& tb_dev = pfsp_tables_cached(I, device);
   const double t2 = now_sec();
 
   if (mode == "hostpool") {
+    StreamGuard stream;
+    const PfspDevTables& tb_dev = pfsp_tables_cached(I, device);
     PinnedGuard<PFSPNode> parents(M);
     PinnedGuard<int32_t> bounds(static_cast<size_t>(M) * jobs);
     DevGuard<PFSPNode> parents_d(M);
@@ -913,59 +1008,11 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       pfsp_generate_children(I, parents.p, n, bounds.p, tree, sol, best, pool);
     }
   } else if (mode == "devpool") {
-    if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
-      throw std::invalid_argument("devpool requires M * jobs <= 2^31");
-    int S = devpool_slices();
-    int maxd = 0;  // same deep-frontier rule as N-Queens (12+ open jobs)
-    for (size_t i = 0; i < pool.size(); i++)
-      maxd = std::max(maxd, static_cast<int>(pool.data()[i].depth));
-    if (jobs - maxd < 12)
-      while (S > 1 && pool.size() < static_cast<size_t>(S) * 2048) S--;
-    const int NS = (S == 1) ? 1 : S * 4;
-    std::vector<std::vector<PFSPNode>> slices(NS);
-    {
-      const PFSPNode* src = pool.data();
-      const size_t total = pool.size();
-      for (int t = 0; t < NS; t++) slices[t].reserve(total / NS + 1);
-      for (size_t i = 0; i < total; i++) slices[i % NS].push_back(src[i]);
-      pool.clear();
-    }
-    // slices share the incumbent through this atomic even when no external
-    // one is plugged in (cross-slice pruning; identical counts at ub=1)
-    std::atomic<int> local_best{best};
-    std::atomic<int>* sb = shared_best ? shared_best : &local_best;
-    std::atomic<int> next_slice{0};
-    SliceShare share;
-    SliceShare* sharep = (S > 1) ? &share : nullptr;
-    std::vector<SliceOut> outs(S);
-    std::vector<std::vector<PFSPNode>> lefts(S);
-    std::vector<std::exception_ptr> errs(S);
-    std::vector<std::thread> threads;
-    for (int t = 0; t < S; t++) {
-      threads.emplace_back([&, t] {
-        try {
-          outs[t] = devpool_thread_pfsp(slices, next_slice, I, tb_dev, lbk, best,
-                                        m, M, device, capacity, sb, S == 1, sharep,
-                                        lefts[t]);
-        } catch (...) {
-          errs[t] = std::current_exception();
-        }
-      });
-    }
-    for (auto& th : threads) th.join();
-    for (auto& e : errs)
-      if (e) std::rethrow_exception(e);
-    for (int t = 0; t < S; t++) {
-      tree += outs[t].fin.tree;
-      sol += outs[t].fin.sol;
-      if (outs[t].fin.best < best) best = outs[t].fin.best;
-      merge_slice_diag(r, outs[t].diag);
-      if (!lefts[t].empty()) pool.pushBackBulk(lefts[t].data(), lefts[t].size());
-    }
-    {
-      const int sbv = sb->load(std::memory_order_relaxed);
-      if (sbv < best) best = sbv;
-    }
+    DevpoolMultiOut o =
+        pfsp_devpool_multi(I, pool, lbk, best, m, M, {device}, capacity, shared_best, r);
+    tree += o.tree;
+    sol += o.sol;
+    if (o.best < best) best = o.best;
   } else {
     throw std::invalid_argument("mode must be hostpool or devpool");
   }

@@ -22,6 +22,28 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
                     int best0, double phase1_time, unsigned long long capacity,
                     std::atomic<int>* shared_best);
 
+// Multi-device devpool core: S slice threads per worker (device entry) pull
+// frontier slices off ONE shared queue — an oversubscribed dynamic partition
+// that load-balances across devices without pausing engines (the role of the
+// reference's intra-node stealing, pfsp_multigpu_chpl.chpl:438-479) — and
+// same-device threads additionally donate half-pools at readback boundaries.
+// Leftover nodes (pools that drained below m) are pushed back into `pool`
+// for the caller's CPU phase 3. `devices` may repeat physical devices
+// (D workers on fewer GPUs). Launch/copy counters merge into `diag`.
+struct DevpoolMultiOut {
+  uint64_t tree = 0, sol = 0;
+  int best = 0;
+  std::vector<uint64_t> per_dev;  // explored tree per worker (workload shares)
+};
+DevpoolMultiOut nq_devpool_multi(Pool<NQNode>& pool, int N, int g, int m, int M,
+                                 const std::vector<int>& devices,
+                                 unsigned long long capacity, Result& diag);
+DevpoolMultiOut pfsp_devpool_multi(const PfspInstance& I, Pool<PFSPNode>& pool, int lbk,
+                                   int best0, int m, int M,
+                                   const std::vector<int>& devices,
+                                   unsigned long long capacity,
+                                   std::atomic<int>* shared_best, Result& diag);
+
 Result nqueens_gpu(int N, int g, int m, int M, int device, const std::string& mode,
                    unsigned long long capacity);
 Result nqueens_gpu_from_pool(const std::vector<NQNode>& nodes, int N, int g, int m, int M,
@@ -34,11 +56,14 @@ Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst, const st
                           int ub, int best0, int m, int M, int device,
                           const std::string& mode, unsigned long long capacity);
 
-// engine_multi.cpp: in-process multi-GPU tier (eval = "gpu" or "cpu").
+// engine_multi.cpp: in-process multi-GPU tier (eval = "devpool", "gpu" or
+// "cpu"; capacity is the per-slice-thread device pool size in nodes, devpool
+// eval only).
 Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& eval,
-                        double perc = 0.5);
+                        double perc = 0.5, unsigned long long capacity = 1ull << 27);
 Result pfsp_multigpu(int inst, const std::string& lb, int ub, int m, int M, int D,
-                     const std::string& eval, bool share_best, double perc = 0.5);
+                     const std::string& eval, bool share_best, double perc = 0.5,
+                     unsigned long long capacity = 1ull << 27);
 
 // Background-thread PFSP engine with a shared incumbent for mid-search
 // RCCL UB exchange (see engine_gpu.cpp).

@@ -102,6 +102,15 @@ void ws_worker(int id, int D, int m, int M, double perc,
             }
             victim.releaseLock();
             if (got > 0) {
+              // go BUSY before publishing the stolen nodes: otherwise the
+              // all-idle scan can fire in the window between the refill and
+              // the state flip and the other workers exit early, leaving the
+              // thief to drain the stolen work alone (counts stay correct —
+              // only tail parallelism was lost)
+              if (idle) {
+                idle = false;
+                states[id].store(BUSY, std::memory_order_release);
+              }
               own.pushBackBulk(steal_buf.data(), got);
               diag.steals++;
               stolen = true;
@@ -232,12 +241,14 @@ struct PfspGpuCtx {
 
 }  // namespace
 
-// devpool-per-worker variant: each worker drives the full single-GPU devpool
-// engine (with its own stream slices) on its partition — the performance
-// path for the CLI multigpu tier. No inter-worker stealing (like the
-// reference's own CUDA multi-GPU baseline, nqueens_multigpu_cuda.cu:287-313);
-// the deep round-robin partition keeps workers statistically balanced, and
-// PFSP workers share the incumbent through one atomic.
+// devpool multi-worker tier: D workers (one per GPU, wrapping when D > device
+// count) whose slice threads all pull from ONE shared frontier queue
+// (nq_devpool_multi / pfsp_devpool_multi) — dynamic cross-worker balancing,
+// the role of the reference's intra-node work stealing
+// (pfsp_multigpu_chpl.chpl:438-479), without pausing running engines. PFSP
+// workers share the incumbent through one atomic. Leftovers merge into the
+// parent pool and the CPU phase-3 drain finishes them (reference flow,
+// nqueens_multigpu_chpl.chpl:315-330).
 static Result nqueens_multigpu_devpool(int N, int g, int m, int M, int D,
                                        unsigned long long capacity) {
   Result r;
@@ -252,48 +263,24 @@ static Result nqueens_multigpu_devpool(int N, int g, int m, int M, int D,
 
   const int ndev = gpu_device_count();
   if (ndev == 0) throw std::runtime_error("no HIP device visible");
-  std::vector<Pool<NQNode>> parts(D);
-  {
-    const NQNode* src = pool.data();
-    for (size_t i = 0; i < pool.size(); i++) parts[i % D].pushBack(src[i]);
-    pool.clear();
-  }
-  std::vector<Result> outs(D);
-  std::vector<std::exception_ptr> errs(D);
-  std::vector<std::thread> threads;
+  std::vector<int> devices(D);
+  for (int d = 0; d < D; d++) devices[d] = d % ndev;
   const double t2 = now_sec();
-  for (int d = 0; d < D; d++) {
-    threads.emplace_back([&, d] {
-      try {
-        outs[d] = nqueens_gpu_run(parts[d], N, g, m, M, d % ndev, "devpool", 0, 0, 0.0,
-                                  capacity);
-      } catch (...) {
-        errs[d] = std::current_exception();
-      }
-    });
-  }
-  for (auto& t : threads) t.join();
-  for (auto& e : errs)
-    if (e) std::rethrow_exception(e);
-  uint64_t tree2 = tree, sol2 = sol;
-  for (auto& o : outs) {
-    tree2 += o.tree;
-    sol2 += o.sol;
-    r.per_worker.push_back(o.tree);
-    r.kernel_launch += o.kernel_launch;
-    r.h2d += o.h2d;
-    r.d2h += o.d2h;
-    r.h2d_bytes += o.h2d_bytes;
-    r.d2h_bytes += o.d2h_bytes;
-    r.gpu_iters += o.gpu_iters;
-  }
+  DevpoolMultiOut o = nq_devpool_multi(pool, N, g, m, M, devices, capacity, r);
+  const uint64_t tree2 = tree + o.tree, sol2 = sol + o.sol;
+  r.per_worker = o.per_dev;
   const double t3 = now_sec();
   r.phases.push_back({tree2 - tree, sol2 - sol, t3 - t2});
-  r.phases.push_back({0, 0, 0.0});  // workers drained their own phase 3
-  r.tree = tree2;
-  r.sol = sol2;
+
+  uint64_t tree3 = tree2, sol3 = sol2;
+  NQNode parent;
+  while (pool.popBack(parent)) nq_decompose(parent, N, g, tree3, sol3, pool);
+  const double t4 = now_sec();
+  r.phases.push_back({tree3 - tree2, sol3 - sol2, t4 - t3});
+  r.tree = tree3;
+  r.sol = sol3;
   r.gpu_time = t3 - t2;
-  r.time = p1 + (t3 - t2);
+  r.time = p1 + (t4 - t2);
   return r;
 }
 
@@ -314,60 +301,37 @@ static Result pfsp_multigpu_devpool(int inst, const std::string& lb_str, int ub,
 
   const int ndev = gpu_device_count();
   if (ndev == 0) throw std::runtime_error("no HIP device visible");
-  std::vector<Pool<PFSPNode>> parts(D);
-  {
-    const PFSPNode* src = pool.data();
-    for (size_t i = 0; i < pool.size(); i++) parts[i % D].pushBack(src[i]);
-    pool.clear();
-  }
-  std::atomic<int> shared_best{best};
-  std::vector<Result> outs(D);
-  std::vector<std::exception_ptr> errs(D);
-  std::vector<std::thread> threads;
+  std::vector<int> devices(D);
+  for (int d = 0; d < D; d++) devices[d] = d % ndev;
+  const int lbk = (lb == LbKind::LB1_D) ? 0 : (lb == LbKind::LB1 ? 1 : 2);
   const double t2 = now_sec();
-  for (int d = 0; d < D; d++) {
-    threads.emplace_back([&, d] {
-      try {
-        outs[d] = pfsp_gpu_run(I, lb, parts[d], m, M, d % ndev, "devpool", 0, 0, best, 0.0,
-                               capacity, &shared_best);
-      } catch (...) {
-        errs[d] = std::current_exception();
-      }
-    });
-  }
-  for (auto& t : threads) t.join();
-  for (auto& e : errs)
-    if (e) std::rethrow_exception(e);
-  uint64_t tree2 = tree, sol2 = sol;
-  for (auto& o : outs) {
-    tree2 += o.tree;
-    sol2 += o.sol;
-    if (o.optimum < best) best = o.optimum;
-    r.per_worker.push_back(o.tree);
-    r.kernel_launch += o.kernel_launch;
-    r.h2d += o.h2d;
-    r.d2h += o.d2h;
-    r.h2d_bytes += o.h2d_bytes;
-    r.d2h_bytes += o.d2h_bytes;
-    r.gpu_iters += o.gpu_iters;
-  }
+  DevpoolMultiOut o =
+      pfsp_devpool_multi(I, pool, lbk, best, m, M, devices, capacity, nullptr, r);
+  const uint64_t tree2 = tree + o.tree, sol2 = sol + o.sol;
+  if (o.best < best) best = o.best;
+  r.per_worker = o.per_dev;
   const double t3 = now_sec();
   r.phases.push_back({tree2 - tree, sol2 - sol, t3 - t2});
-  r.phases.push_back({0, 0, 0.0});
-  r.tree = tree2;
-  r.sol = sol2;
+
+  uint64_t tree3 = tree2, sol3 = sol2;
+  PFSPNode parent;
+  while (pool.popBack(parent)) pfsp_decompose(I, lb, parent, tree3, sol3, best, pool);
+  const double t4 = now_sec();
+  r.phases.push_back({tree3 - tree2, sol3 - sol2, t4 - t3});
+  r.tree = tree3;
+  r.sol = sol3;
   r.optimum = best;
   r.gpu_time = t3 - t2;
-  r.time = p1 + (t3 - t2);
+  r.time = p1 + (t4 - t2);
   return r;
 }
 
 Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& eval,
-                        double perc) {
+                        double perc, unsigned long long capacity) {
   if (D < 1) throw std::invalid_argument("D must be >= 1");
   if (!(perc > 0.0 && perc < 1.0))
     throw std::invalid_argument("perc must be in (0,1) (reference: 0 < --perc < 100)");
-  if (eval == "devpool") return nqueens_multigpu_devpool(N, g, m, M, D, 1ull << 27);
+  if (eval == "devpool") return nqueens_multigpu_devpool(N, g, m, M, D, capacity);
   Result r;
   Pool<NQNode> pool;
   pool.pushBack(nq_root());
@@ -471,11 +435,12 @@ Result nqueens_multigpu(int N, int g, int m, int M, int D, const std::string& ev
 }
 
 Result pfsp_multigpu(int inst, const std::string& lb_str, int ub, int m, int M, int D,
-                     const std::string& eval, bool share_best, double perc) {
+                     const std::string& eval, bool share_best, double perc,
+                     unsigned long long capacity) {
   if (D < 1) throw std::invalid_argument("D must be >= 1");
   if (!(perc > 0.0 && perc < 1.0))
     throw std::invalid_argument("perc must be in (0,1) (reference: 0 < --perc < 100)");
-  if (eval == "devpool") return pfsp_multigpu_devpool(inst, lb_str, ub, m, M, D, 1ull << 27);
+  if (eval == "devpool") return pfsp_multigpu_devpool(inst, lb_str, ub, m, M, D, capacity);
   const LbKind lb = lb_from_string(lb_str);
   PfspInstance I = make_pfsp_instance(inst, ub);
   Result r;

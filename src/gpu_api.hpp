@@ -11,17 +11,21 @@ typedef struct ihipStream_t* hipStream_t;
 
 namespace gats {
 
-// Device-resident control block for devpool mode. One per engine; the host
-// reads it back (64 B) every few iterations instead of synchronizing per
-// offload round like the reference does (pfsp_gpu_chpl.chpl:373-396).
+// Device-resident control block for devpool mode (48 B). Each engine keeps a
+// parity-alternating PAIR of these: iteration i's kernels read ctl[parity]
+// and only the LAST block of gather2(i) writes ctl[1-parity] (single-writer;
+// no atomics on size — the one exception is `best`, which expand kernels
+// atomicMin into ctl[parity] and gather2 copies forward). The host reads one
+// block back every few iterations instead of synchronizing per offload round
+// like the reference does (pfsp_gpu_chpl.chpl:373-396).
 struct DevCtl {
-  unsigned long long size = 0;   // live pool size; expand kernels atomicAdd on it
+  unsigned long long size = 0;   // live pool size (written by gather2's last block)
   unsigned long long chunk = 0;  // current iteration's popped chunk
   unsigned long long tree = 0;
   unsigned long long sol = 0;
   unsigned long long iters = 0;  // productive iterations
   int best = 0;                  // PFSP incumbent (atomicMin); unused for N-Queens
-  int overflow = 0;              // pool capacity exceeded -> host aborts
+  int overflow = 0;              // pool capacity exceeded -> host spills or aborts
 };
 
 // Device pointers of the PFSP bound tables (int16/uint8 compressed; the

@@ -97,13 +97,33 @@ __device__ inline void emit_pfsp_child(PFSPNode* pool, unsigned long long slot,
 // ---------------------------------------------------------------------------
 
 // Diagonal safety of placing row `q` at column `depth` against columns [0,depth).
+// g repeats every check (the reference's artificial-work knob,
+// nqueens_gpu_cuda.cu:137-164); the empty asm is an optimization barrier so
+// the repeats are real work the compiler cannot hoist out of the loop.
 __device__ inline uint8_t nq_safe(const uint8_t* board, int depth, int q, int g) {
   uint8_t safe = 1;
   for (int i = 0; i < depth; i++) {
     const int o = board[i];
-    for (int r = 0; r < g; r++) safe &= (o != q - (depth - i)) & (o != q + (depth - i));
+    for (int r = 0; r < g; r++) {
+      int qq = q;
+      if (g > 1) asm volatile("" : "+v"(qq));
+      safe &= (o != qq - (depth - i)) & (o != qq + (depth - i));
+    }
   }
   return safe;
+}
+
+// Mask-based safety test repeated g times (devpool path's g semantics: the
+// per-child evaluation is the O(1) mask test, so g scales exactly that).
+__device__ inline bool nq_mask_unsafe(uint32_t b, uint32_t occ, int g) {
+  if (g == 1) return (b & occ) != 0;
+  bool unsafe = false;
+  for (int r = 0; r < g; r++) {
+    uint32_t o = occ;
+    asm volatile("" : "+v"(o));  // force a real re-test per repeat
+    unsafe |= (b & o) != 0;
+  }
+  return unsafe;
 }
 
 // hostpool mode: one thread per (parent, k), labels out
@@ -469,12 +489,24 @@ __device__ inline unsigned long long block_reduce_u64(unsigned long long v) {
 // keeps every stack frame in registers (no runtime-indexed arrays).
 template <int B>
 __device__ inline void nq_dfs(uint32_t cols, uint32_t d1, uint32_t d2, int placed, int N,
-                              unsigned long long& tree, unsigned long long& sol) {
+                              int g, unsigned long long& tree, unsigned long long& sol) {
   if constexpr (B == 0) {
     return;
   } else {
     const uint32_t msk = (1u << N) - 1u;
-    uint32_t free = ~(cols | d1 | d2) & msk;
+    uint32_t free;
+    if (g == 1) {
+      free = ~(cols | d1 | d2) & msk;
+    } else {
+      // repeat the node's whole safety evaluation g times (reference knob);
+      // barrier keeps each repeat from being folded away
+      free = 0;
+      for (int r = 0; r < g; r++) {
+        uint32_t occ = cols | d1 | d2;
+        asm volatile("" : "+v"(occ));
+        free = ~occ & msk;
+      }
+    }
     while (free) {
       const uint32_t bit = free & (0u - free);
       free ^= bit;
@@ -483,7 +515,7 @@ __device__ inline void nq_dfs(uint32_t cols, uint32_t d1, uint32_t d2, int place
         sol++;
       } else {
         nq_dfs<B - 1>(cols | bit, ((d1 | bit) << 1) & msk, (d2 | bit) >> 1, placed + 1, N,
-                      tree, sol);
+                      g, tree, sol);
       }
     }
   }
@@ -537,8 +569,7 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
         uint32_t cols = 0, d1 = 0, d2 = 0;
         const int depth = p.depth;
         for (int i = 0; i < depth && i < N; i++) {
-          uint32_t b = 1u << p.board[i];
-          for (int r = 1; r < g; r++) b = 1u << p.board[i];  // artificial work knob
+          const uint32_t b = 1u << p.board[i];
           cols |= b;
           d1 = ((d1 | b) << 1) & msk;
           d2 = (d2 | b) >> 1;
@@ -566,7 +597,7 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
           const uint32_t d1 = pmask[pid - first][1];
           const uint32_t d2 = pmask[pid - first][2];
           const uint32_t b = 1u << p.board[k];
-          if (!(b & (cols | d1 | d2))) {  // == nq_safe (diagonal masks)
+          if (!nq_mask_unsafe(b, cols | d1 | d2, g)) {  // == nq_safe (diag masks)
             const int rem = N - (depth + 1);  // levels below the child
             if (rem <= finish) {
               // child + its whole subtree counted here, nothing pushed
@@ -576,7 +607,7 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
               } else {
                 const uint32_t msk = (1u << N) - 1u;
                 nq_dfs<NQ_FINISH_MAX>(cols | b, ((d1 | b) << 1) & msk, (d2 | b) >> 1,
-                                      depth + 1, N, extra, sols);
+                                      depth + 1, N, g, extra, sols);
               }
             } else {
               lab[j] = 1;

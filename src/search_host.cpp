@@ -30,7 +30,11 @@ bool nq_is_safe(const uint8_t* board, int depth, int row_pos, int g) {
   for (int i = 0; i < depth; i++) {
     const int other = board[i];
     for (int r = 0; r < g; r++) {
-      if (other == row_pos - (depth - i) || other == row_pos + (depth - i)) safe = 0;
+      int rp = row_pos;
+      // barrier keeps the g repeats real work (the reference's artificial-work
+      // knob); without it the loop is invariant and folds to one check
+      if (g > 1) asm volatile("" : "+r"(rp));
+      if (other == rp - (depth - i) || other == rp + (depth - i)) safe = 0;
     }
   }
   return safe != 0;
