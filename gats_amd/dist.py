@@ -160,9 +160,14 @@ def run_nqueens(N, g=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
         # rule keeps even 2048-node dynamic sub-slices fully sliced, so a
         # bigger (costlier) frontier buys nothing
         frontier_target = 65536
-    nodes, tree1, sol1 = c.nq_bfs_frontier(N, g, frontier_target)
-    phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
     local = rank % max(1, c.gpu_device_count())
+    if engine == "gpu":
+        # device-built frontier (~0.2 ms vs ~8 ms CPU at 65536); every rank
+        # builds it redundantly and deterministically
+        nodes, tree1, sol1 = c.nq_gpu_frontier(N, g, frontier_target, local)
+    else:
+        nodes, tree1, sol1 = c.nq_bfs_frontier(N, g, frontier_target)
+    phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
 
     def run_slice(sl):
         if engine == "gpu":
@@ -191,9 +196,12 @@ def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 
         # PFSP 20-job trees are small (ta014 lb1 ~2.6M nodes); a deep frontier
         # would move a large share of the search onto the single-threaded CPU
         frontier_target = max(2048, 2048 * world)
-    nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
-    phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
     local = rank % max(1, c.gpu_device_count())
+    if engine == "gpu":
+        nodes, tree1, sol1, best = c.pfsp_gpu_frontier(inst, lb, ub, frontier_target, local)
+    else:
+        nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
+    phase1 = {"tree": tree1 if rank == 0 else 0, "sol": sol1 if rank == 0 else 0, "time": 0.0}
     best_so_far = [best]
 
     def run_slice(sl):
@@ -235,9 +243,11 @@ def run_pfsp_shared_ub(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
     rank, world = init_dist()
     if frontier_target is None:
         frontier_target = max(2048, 2048 * world)
-    nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
-    my = slice_frontier(nodes, rank, world)
     local = rank % max(1, c.gpu_device_count())
+    # device-built frontier; no leaf can be reached at these depths so the
+    # build is deterministic (identical on every rank) even with ub=0
+    nodes, tree1, sol1, best = c.pfsp_gpu_frontier(inst, lb, ub, frontier_target, local)
+    my = slice_frontier(nodes, rank, world)
     t0 = time.perf_counter()
     eng = c.PfspAsyncEngine(my, inst, lb, ub, best, m, M, local, capacity)
     dev = _backend_device(f"cuda:{local}")

@@ -122,6 +122,31 @@ py::tuple nq_bfs_frontier(int N, int g, size_t target) {
   return py::make_tuple(nodes_to_bytes(pool.data(), pool.size()), tree, sol);
 }
 
+py::tuple nq_gpu_frontier_py(int N, int g, size_t target, int device) {
+  uint64_t tree = 0, sol = 0;
+  std::vector<NQNode> nodes;
+  {
+    py::gil_scoped_release rel;
+    nodes = nq_gpu_frontier(N, g, target, device, tree, sol);
+  }
+  return py::make_tuple(nodes_to_bytes(nodes.data(), nodes.size()), tree, sol);
+}
+
+py::tuple pfsp_gpu_frontier_py(int inst, const std::string& lb_str, int ub, size_t target,
+                               int device) {
+  const LbKind lb = lb_from_string(lb_str);
+  PfspInstance I = make_pfsp_instance(inst, ub);
+  const int lbk = (lb == LbKind::LB1_D) ? 0 : (lb == LbKind::LB1 ? 1 : 2);
+  uint64_t tree = 0, sol = 0;
+  int best = I.init_ub;
+  std::vector<PFSPNode> nodes;
+  {
+    py::gil_scoped_release rel;
+    nodes = pfsp_gpu_frontier(I, lbk, target, device, I.init_ub, tree, sol, best);
+  }
+  return py::make_tuple(nodes_to_bytes(nodes.data(), nodes.size()), tree, sol, best);
+}
+
 py::tuple pfsp_bfs_frontier(int inst, const std::string& lb_str, int ub, size_t target) {
   const LbKind lb = lb_from_string(lb_str);
   PfspInstance I = make_pfsp_instance(inst, ub);
@@ -327,6 +352,11 @@ PYBIND11_MODULE(_core, mod) {
           py::arg("target") = 1024);
   mod.def("pfsp_bfs_frontier", &pfsp_bfs_frontier, py::arg("inst"), py::arg("lb") = "lb1",
           py::arg("ub") = 1, py::arg("target") = 1024);
+  mod.def("nq_gpu_frontier", &nq_gpu_frontier_py, py::arg("N"), py::arg("g") = 1,
+          py::arg("target") = 1024, py::arg("device") = 0);
+  mod.def("pfsp_gpu_frontier", &pfsp_gpu_frontier_py, py::arg("inst"),
+          py::arg("lb") = "lb1", py::arg("ub") = 1, py::arg("target") = 1024,
+          py::arg("device") = 0);
 
   mod.def("nq_cpu_labels", &nq_cpu_labels);
   mod.def("pfsp_cpu_bounds", &pfsp_cpu_bounds);

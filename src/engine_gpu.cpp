@@ -21,6 +21,7 @@
 #include <cstdlib>
 #include <chrono>
 #include <condition_variable>
+#include <deque>
 #include <functional>
 #include <map>
 #include <mutex>
@@ -183,6 +184,72 @@ struct PfspTablesGuard {
   }
 };
 
+// Parked worker threads reused across searches (the persistent-engine thread
+// component): spawning slice threads per search costs ~50-100 us each and the
+// dist tier re-arms an engine per frontier claim, so threads park on a
+// condvar between tasks instead of being joined. Threads are created on
+// demand up to the high-water mark of concurrent tasks; the pool object is
+// intentionally leaked so detached workers never touch destroyed statics at
+// process exit. HIP device selection is per-thread and every engine task sets
+// it on entry, so reuse across devices is safe.
+class WorkerPool {
+ public:
+  static WorkerPool& instance() {
+    static WorkerPool* p = new WorkerPool();  // leaked by design
+    return *p;
+  }
+  void submit(std::function<void()> fn) {
+    std::lock_guard<std::mutex> l(mu_);
+    q_.push_back(std::move(fn));
+    // avail_ counts workers inside (or re-acquiring from) cv_.wait: each will
+    // take one queued task on its next lock acquisition, so spawn only when
+    // the queue outgrew the parked set — never serializes two tasks onto one
+    // worker
+    if (q_.size() > static_cast<size_t>(avail_)) std::thread([this] { loop(); }).detach();
+    cv_.notify_one();
+  }
+
+ private:
+  void loop() {
+    std::unique_lock<std::mutex> l(mu_);
+    for (;;) {
+      while (q_.empty()) {
+        avail_++;
+        cv_.wait(l);
+        avail_--;
+      }
+      std::function<void()> fn = std::move(q_.front());
+      q_.pop_front();
+      l.unlock();
+      fn();  // tasks are noexcept by contract (engine lambdas catch all)
+      l.lock();
+    }
+  }
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::deque<std::function<void()>> q_;
+  int avail_ = 0;
+};
+
+// Run T engine tasks on pooled threads and wait for all of them.
+template <class MakeTask>
+static void run_on_pool(int T, MakeTask&& make_task) {
+  std::atomic<int> remaining{T};
+  std::mutex done_mu;
+  std::condition_variable done_cv;
+  for (int t = 0; t < T; t++) {
+    WorkerPool::instance().submit([t, &remaining, &done_mu, &done_cv, &make_task] {
+      make_task(t);
+      if (remaining.fetch_sub(1, std::memory_order_acq_rel) == 1) {
+        std::lock_guard<std::mutex> l(done_mu);
+        done_cv.notify_all();
+      }
+    });
+  }
+  std::unique_lock<std::mutex> l(done_mu);
+  done_cv.wait(l, [&] { return remaining.load(std::memory_order_acquire) == 0; });
+}
+
 // Stream recycling is OFF by default: reusing streams measured a reproducible
 // 40% N-Queens regression (146 vs 108 ms per N=17 search, same box, ROCm 7.2)
 // — fresh streams apparently schedule better across the slice threads. Keep
@@ -283,19 +350,35 @@ int lbk_of(LbKind lb) {
 // other ranks/threads by writing ctl->best on the (synchronized) stream —
 // the RCCL incumbent-UB exchange of the distributed tier plugs in here.
 // enqueue_iter(parity): parity alternates 0/1 per iteration — iteration i
-// reads ctl[parity] and gather2 writes ctl[1-parity]. BATCH is even, so the
-// host always reads (and patches the incumbent of) ctl[0].
-// `readback_hook(host_ctl, done_soon)`: called after every batch readback with
-// the stream idle; may reduce host_ctl->size after carving the pool (the
+// reads ctl[parity] and gather2 writes ctl[1-parity]. The loop tracks the
+// live parity so hooks and readbacks always touch the block the last gather
+// wrote.
+// `readback_hook(host_ctl, live_ctl_d)`: called after every batch readback
+// with the stream idle; may reduce host_ctl->size after carving the pool (the
 // donor path of SliceShare) — it must then also write the device ctl itself.
-using ReadbackHook = std::function<void(DevCtl*)>;
+// `spill_hook(host_ctl, live_ctl_d)`: called (stream idle) when the next
+// iteration's worst-case growth no longer fits `capacity`; carves part of the
+// pool to host storage and rewrites both sizes. The caller re-runs spilled
+// nodes after the loop, so counts stay exact (Pool.chpl:28-31's unbounded-
+// growth contract, met by spilling instead of growing).
+using ReadbackHook = std::function<void(DevCtl*, DevCtl*)>;
+
+struct DevLoopCfg {
+  unsigned long long m = 1;          // loop exits when size < m
+  unsigned long long init_size = 0;  // pool size at entry
+  unsigned long long growth = 0;     // worst-case net pool growth per iteration
+  unsigned long long capacity = 0;   // pool capacity in nodes
+  unsigned long long stop_size = 0;  // frontier builder: stop once size >= this
+  int kernels_per_iter = 2;
+  bool allow_graph = true;
+};
 
 template <class EnqueueIter>
-static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long m,
-                               int kernels_per_iter, EnqueueIter&& enqueue_iter, Result& r,
+static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, const DevLoopCfg& cfg,
+                               EnqueueIter&& enqueue_iter, Result& r,
                                std::atomic<int>* shared_best = nullptr,
-                               bool allow_graph = true,
-                               const ReadbackHook& readback_hook = {}) {
+                               const ReadbackHook& readback_hook = {},
+                               const ReadbackHook& spill_hook = {}) {
   PinnedGuard<DevCtl> ctl_h(1);
   const int BATCH = 16;
   // GATS_NO_GRAPH=1 falls back to eager launches (rocprofv3 crashes tracing
@@ -308,14 +391,41 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
   // with concurrent async work from the other slice threads no matter the
   // capture mode; with >1 slice in flight the host launch latency is hidden
   // by the other slices anyway
-  const bool graph_allowed = allow_graph && std::getenv("GATS_NO_GRAPH") == nullptr;
+  const bool graph_allowed = cfg.allow_graph && cfg.stop_size == 0 &&
+                             std::getenv("GATS_NO_GRAPH") == nullptr;
   const int EAGER_BATCHES = 4;
   int batches = 0;
+  int par = 0;  // parity of the NEXT iteration == index of the live ctl block
+  unsigned long long last_size = cfg.init_size;
   hipGraph_t graph = nullptr;
   hipGraphExec_t exec = nullptr;
   bool overflow = false;
+  *ctl_h.p = DevCtl{};
+  ctl_h.p->size = cfg.init_size;
   while (true) {
-    if (graph_allowed && batches >= EAGER_BATCHES && exec == nullptr) {
+    // capacity-aware batch: never launch more iterations than worst-case
+    // growth allows; spill when even one iteration might not fit
+    int b = (cfg.stop_size > 0) ? 1 : BATCH;
+    if (cfg.growth > 0 && cfg.capacity > 0) {
+      const unsigned long long room =
+          cfg.capacity > last_size ? cfg.capacity - last_size : 0;
+      const unsigned long long bmax = room / cfg.growth;
+      if (bmax == 0 && spill_hook && last_size >= 4 * cfg.m && batches > 0) {
+        spill_hook(ctl_h.p, ctl_d + par);
+        last_size = ctl_h.p->size;
+        continue;
+      }
+      if (bmax == 0) {
+        // worst-case doesn't fit but the actual child count usually does;
+        // run one iteration — the gather kernel's exact-fit guard flags a
+        // REAL overflow, which is fatal (the iteration state is torn)
+        b = 1;
+      } else if (bmax < static_cast<unsigned long long>(b)) {
+        b = static_cast<int>(bmax);
+      }
+    }
+    if (graph_allowed && batches >= EAGER_BATCHES && exec == nullptr && b == BATCH &&
+        par == 0) {
       // one capture at a time: concurrent captures from the slice threads
       // race inside ROCm 7.2 ("previous error during capture"); relaxed mode
       // lets the other slices keep launching meanwhile. A failed capture is
@@ -324,7 +434,7 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
       std::lock_guard<std::mutex> lock(capture_mu);
       hipError_t ce = hipStreamBeginCapture(s, hipStreamCaptureModeRelaxed);
       if (ce == hipSuccess) {
-        for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
+        for (int i = 0; i < BATCH; i++) enqueue_iter(i & 1);
         ce = hipStreamEndCapture(s, &graph);
         if (ce == hipSuccess) {
           if (hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0) != hipSuccess) {
@@ -337,39 +447,35 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
         }
         // a successfully captured batch was not EXECUTED; replay it below or
         // re-run eagerly — either way the work happens exactly once
-        if (exec == nullptr) {
-          (void)hipGetLastError();
-          for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
-          batches++;
-          HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d, sizeof(DevCtl), hipMemcpyDeviceToHost, s));
-          HIP_CHECK(hipStreamSynchronize(s));
-          if (ctl_h.p->overflow) {
-            overflow = true;
-            break;
-          }
-          if (ctl_h.p->size < m) break;
-          continue;
-        }
+        if (exec == nullptr) (void)hipGetLastError();
       } else {
         (void)hipGetLastError();
       }
     }
-    if (exec != nullptr) {
-      HIP_CHECK(hipGraphLaunch(exec, s));
+    if (exec != nullptr && b == BATCH && par == 0) {
+      HIP_CHECK(hipGraphLaunch(exec, s));  // even count: parity unchanged
     } else {
-      for (int b = 0; b < BATCH; b++) enqueue_iter(b & 1);
+      for (int i = 0; i < b; i++) {
+        enqueue_iter(par);
+        par ^= 1;
+      }
     }
     batches++;
-    HIP_CHECK(hipMemcpyAsync(ctl_h.p, ctl_d, sizeof(DevCtl), hipMemcpyDeviceToHost, s));
+    HIP_CHECK(
+        hipMemcpyAsync(ctl_h.p, ctl_d + par, sizeof(DevCtl), hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
-    r.kernel_launch += static_cast<uint64_t>(kernels_per_iter) * BATCH;
+    r.kernel_launch += static_cast<uint64_t>(cfg.kernels_per_iter) * b;
     r.d2h++;
     r.d2h_bytes += sizeof(DevCtl);
+    last_size = ctl_h.p->size;
     if (ctl_h.p->overflow) {
       overflow = true;
       break;
     }
-    if (readback_hook) readback_hook(ctl_h.p);
+    if (readback_hook) {
+      readback_hook(ctl_h.p, ctl_d + par);
+      last_size = ctl_h.p->size;
+    }
     if (shared_best) {
       int mine = ctl_h.p->best;
       int cur = shared_best->load(std::memory_order_relaxed);
@@ -380,17 +486,22 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, unsigned long long 
       if (cur < mine) {
         // adopt the lower incumbent; the stream is idle (synchronized above),
         // so no device atomicMin can race this write
-        HIP_CHECK(hipMemcpyAsync(&ctl_d->best, &cur, sizeof(int), hipMemcpyHostToDevice, s));
+        HIP_CHECK(hipMemcpyAsync(&(ctl_d + par)->best, &cur, sizeof(int),
+                                 hipMemcpyHostToDevice, s));
         HIP_CHECK(hipStreamSynchronize(s));
       }
     }
-    if (ctl_h.p->size < m) break;
+    if (cfg.stop_size > 0 && ctl_h.p->size >= cfg.stop_size) break;
+    if (ctl_h.p->size < cfg.m) break;
   }
   if (exec != nullptr) {
     (void)hipGraphExecDestroy(exec);
     (void)hipGraphDestroy(graph);
   }
-  if (overflow) throw std::runtime_error("device pool overflow; raise capacity");
+  if (overflow)
+    throw std::runtime_error(
+        "device pool overflow: capacity too small for one offload iteration even after "
+        "spilling; raise --capacity (or lower --M)");
   return *ctl_h.p;
 }
 
@@ -563,6 +674,7 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
   DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G);
   DevGuard<unsigned long long> bs_d(G), be_d(G);
+  std::vector<NQNode> spilled;  // capacity-pressure spill, re-run after the slice
 
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
@@ -572,9 +684,30 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
     launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, childbuf_d.p, pool_d.p, stride, G,
                       m, M, capacity, stream.s);
   };
-  ReadbackHook hook = [&](DevCtl* hc) {
-    donate_if_wanted(share, hc, ctl_d.p, pool_d.p, m, stream.s);
+  ReadbackHook hook = [&](DevCtl* hc, DevCtl* live) {
+    donate_if_wanted(share, hc, live, pool_d.p, m, stream.s);
   };
+  ReadbackHook spill = [&](DevCtl* hc, DevCtl* live) {
+    const unsigned long long half = hc->size / 2;
+    if (half == 0) return;
+    const unsigned long long newsize = hc->size - half;
+    const size_t base = spilled.size();
+    spilled.resize(base + half);
+    HIP_CHECK(hipMemcpyAsync(spilled.data() + base, pool_d.p + newsize,
+                             half * sizeof(NQNode), hipMemcpyDeviceToHost, stream.s));
+    HIP_CHECK(hipMemcpyAsync(&live->size, &newsize, sizeof(newsize),
+                             hipMemcpyHostToDevice, stream.s));
+    HIP_CHECK(hipStreamSynchronize(stream.s));
+    hc->size = newsize;
+    r.d2h++;
+    r.d2h_bytes += half * sizeof(NQNode);
+  };
+
+  DevLoopCfg cfg;
+  cfg.m = m;
+  cfg.capacity = capacity;
+  cfg.growth = static_cast<unsigned long long>(M) * N;  // worst-case children/iter
+  cfg.allow_graph = allow_graph;
 
   auto run_pool = [&](unsigned long long init_size) {
     DevCtl ctl{};
@@ -585,9 +718,10 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
     HIP_CHECK(hipStreamSynchronize(stream.s));  // ctl is a stack temporary
     r.h2d += 2;
     r.h2d_bytes += 2 * sizeof(DevCtl);
+    cfg.init_size = init_size;
     RunnerScope runner(share);
-    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, nullptr,
-                                        allow_graph, hook);
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, cfg, iter, r, nullptr, hook,
+                                        spill);
     out.fin.tree += fin.tree;
     out.fin.sol += fin.sol;
     r.gpu_iters += fin.iters;
@@ -599,6 +733,20 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
       HIP_CHECK(hipStreamSynchronize(stream.s));
       r.d2h++;
       r.d2h_bytes += fin.size * sizeof(NQNode);
+    }
+  };
+  // re-run anything the capacity spill pushed out (it may spill again; the
+  // stack shrinks by at least half the capacity per round)
+  auto drain_spilled = [&]() {
+    while (!spilled.empty()) {
+      const size_t take = std::min<size_t>(spilled.size(), capacity / 2);
+      std::vector<NQNode> chunk(spilled.end() - take, spilled.end());
+      spilled.resize(spilled.size() - take);
+      HIP_CHECK(hipMemcpyAsync(pool_d.p, chunk.data(), take * sizeof(NQNode),
+                               hipMemcpyHostToDevice, stream.s));
+      r.h2d++;
+      r.h2d_bytes += take * sizeof(NQNode);
+      run_pool(take);
     }
   };
 
@@ -614,6 +762,7 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
     r.h2d++;
     r.h2d_bytes += nodes.size() * sizeof(NQNode);
     run_pool(nodes.size());
+    drain_spilled();
   }
 
   // queue drained: take donated halves of still-running slices until no
@@ -626,6 +775,7 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
     HIP_CHECK(hipStreamSynchronize(stream.s));
     ack_taken(*share);
     run_pool(w.n);
+    drain_spilled();
   }
   return out;
 }
@@ -655,6 +805,8 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   const bool presum = (lbk == 2);
   DevGuard<uint32_t> gsum_d(presum ? (G + 255) / 256 : 1);
 
+  std::vector<PFSPNode> spilled;  // capacity-pressure spill, re-run after the slice
+
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
@@ -664,9 +816,30 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
     launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, presum ? gsum_d.p : nullptr,
                         childbuf_d.p, pool_d.p, stride, G, m, M, capacity, stream.s);
   };
-  ReadbackHook hook = [&](DevCtl* hc) {
-    donate_if_wanted(share, hc, ctl_d.p, pool_d.p, m, stream.s);
+  ReadbackHook hook = [&](DevCtl* hc, DevCtl* live) {
+    donate_if_wanted(share, hc, live, pool_d.p, m, stream.s);
   };
+  ReadbackHook spill = [&](DevCtl* hc, DevCtl* live) {
+    const unsigned long long half = hc->size / 2;
+    if (half == 0) return;
+    const unsigned long long newsize = hc->size - half;
+    const size_t base = spilled.size();
+    spilled.resize(base + half);
+    HIP_CHECK(hipMemcpyAsync(spilled.data() + base, pool_d.p + newsize,
+                             half * sizeof(PFSPNode), hipMemcpyDeviceToHost, stream.s));
+    HIP_CHECK(hipMemcpyAsync(&live->size, &newsize, sizeof(newsize),
+                             hipMemcpyHostToDevice, stream.s));
+    HIP_CHECK(hipStreamSynchronize(stream.s));
+    hc->size = newsize;
+    r.d2h++;
+    r.d2h_bytes += half * sizeof(PFSPNode);
+  };
+
+  DevLoopCfg cfg;
+  cfg.m = m;
+  cfg.capacity = capacity;
+  cfg.growth = static_cast<unsigned long long>(M) * jobs;
+  cfg.allow_graph = allow_graph;
 
   auto run_pool = [&](unsigned long long init_size, int init_best) {
     DevCtl ctl{};
@@ -681,9 +854,10 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
     HIP_CHECK(hipStreamSynchronize(stream.s));
     r.h2d += 2;
     r.h2d_bytes += 2 * sizeof(DevCtl);
+    cfg.init_size = init_size;
     RunnerScope runner(share);
-    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, m, 2, iter, r, shared_best,
-                                        allow_graph, hook);
+    const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, cfg, iter, r, shared_best,
+                                        hook, spill);
     out.fin.tree += fin.tree;
     out.fin.sol += fin.sol;
     if (fin.best < out.fin.best) out.fin.best = fin.best;
@@ -699,6 +873,18 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
       r.d2h_bytes += fin.size * sizeof(PFSPNode);
     }
   };
+  auto drain_spilled = [&]() {
+    while (!spilled.empty()) {
+      const size_t take = std::min<size_t>(spilled.size(), capacity / 2);
+      std::vector<PFSPNode> chunk(spilled.end() - take, spilled.end());
+      spilled.resize(spilled.size() - take);
+      HIP_CHECK(hipMemcpyAsync(pool_d.p, chunk.data(), take * sizeof(PFSPNode),
+                               hipMemcpyHostToDevice, stream.s));
+      r.h2d++;
+      r.h2d_bytes += take * sizeof(PFSPNode);
+      run_pool(take, out.fin.best);
+    }
+  };
 
   int si;
   while ((si = next_slice.fetch_add(1)) < static_cast<int>(slices.size())) {
@@ -710,6 +896,7 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
     r.h2d++;
     r.h2d_bytes += nodes.size() * sizeof(PFSPNode);
     run_pool(nodes.size(), best0);
+    drain_spilled();
   }
 
   // queue drained: take donated halves of still-running slices (ROADMAP #2)
@@ -721,6 +908,7 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
     HIP_CHECK(hipStreamSynchronize(stream.s));
     ack_taken(*share);
     run_pool(w.n, w.best);
+    drain_spilled();
   }
   return out;
 }
@@ -779,20 +967,16 @@ DevpoolMultiOut nq_devpool_multi(Pool<NQNode>& pool, int N, int g, int m, int M,
   std::vector<SliceOut> outs(T);
   std::vector<std::vector<NQNode>> lefts(T);
   std::vector<std::exception_ptr> errs(T);
-  std::vector<std::thread> threads;
   const bool allow_graph = (T == 1);
-  for (int t = 0; t < T; t++) {
-    threads.emplace_back([&, t] {
-      try {
-        SliceShare* sh = (S > 1) ? &shares[t / S] : nullptr;
-        outs[t] = devpool_thread_nq(slices, next_slice, N, g, m, M, devices[t / S],
-                                    finish, capacity, allow_graph, sh, lefts[t]);
-      } catch (...) {
-        errs[t] = std::current_exception();
-      }
-    });
-  }
-  for (auto& th : threads) th.join();
+  run_on_pool(T, [&](int t) {
+    try {
+      SliceShare* sh = (S > 1) ? &shares[t / S] : nullptr;
+      outs[t] = devpool_thread_nq(slices, next_slice, N, g, m, M, devices[t / S],
+                                  finish, capacity, allow_graph, sh, lefts[t]);
+    } catch (...) {
+      errs[t] = std::current_exception();
+    }
+  });
   for (auto& e : errs)
     if (e) std::rethrow_exception(e);
   DevpoolMultiOut o;
@@ -841,21 +1025,17 @@ DevpoolMultiOut pfsp_devpool_multi(const PfspInstance& I, Pool<PFSPNode>& pool, 
   std::vector<SliceOut> outs(T);
   std::vector<std::vector<PFSPNode>> lefts(T);
   std::vector<std::exception_ptr> errs(T);
-  std::vector<std::thread> threads;
   const bool allow_graph = (T == 1);
-  for (int t = 0; t < T; t++) {
-    threads.emplace_back([&, t] {
-      try {
-        SliceShare* sh = (S > 1) ? &shares[t / S] : nullptr;
-        outs[t] = devpool_thread_pfsp(slices, next_slice, I, lbk, best0, m, M,
-                                      devices[t / S], capacity, sb, allow_graph, sh,
-                                      lefts[t]);
-      } catch (...) {
-        errs[t] = std::current_exception();
-      }
-    });
-  }
-  for (auto& th : threads) th.join();
+  run_on_pool(T, [&](int t) {
+    try {
+      SliceShare* sh = (S > 1) ? &shares[t / S] : nullptr;
+      outs[t] = devpool_thread_pfsp(slices, next_slice, I, lbk, best0, m, M,
+                                    devices[t / S], capacity, sb, allow_graph, sh,
+                                    lefts[t]);
+    } catch (...) {
+      errs[t] = std::current_exception();
+    }
+  });
   for (auto& e : errs)
     if (e) std::rethrow_exception(e);
   DevpoolMultiOut o;
@@ -1075,6 +1255,118 @@ Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst,
   const int best = (best0 > 0) ? best0 : I.init_ub;
   return pfsp_gpu_run(I, lb, pool, m, M, device, mode, 0, 0, best, 0.0, capacity,
                       nullptr);
+}
+
+// ---------------------------------------------------------------------------
+// Device-built frontiers (declared in engine_gpu.hpp)
+// ---------------------------------------------------------------------------
+
+std::vector<NQNode> nq_gpu_frontier(int N, int g, size_t target, int device,
+                                    uint64_t& tree, uint64_t& sol) {
+  HIP_CHECK(hipSetDevice(device));
+  StreamGuard stream;
+  Result r;
+  const unsigned long long M = target;  // one level per iteration below target
+  const unsigned long long capacity = target * (MAX_JOBS + 1) + 64;  // branching <= 20
+  DevGuard<NQNode> pool_d(capacity);
+  DevGuard<DevCtl> ctl_d(2);
+  const int G = devpool_grid(M, N, 1);
+  const int stride = devpool_stride(1);
+  DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
+  DevGuard<uint32_t> bc_d(G);
+  DevGuard<unsigned long long> bs_d(G), be_d(G);
+  int finish = 8;
+  if (const char* e = std::getenv("GATS_NQ_FINISH")) finish = atoi(e);
+  if (finish > 8) finish = 8;
+
+  NQNode root = nq_root();
+  DevCtl ctl{};
+  ctl.size = 1;
+  HIP_CHECK(hipMemcpyAsync(pool_d.p, &root, sizeof(NQNode), hipMemcpyHostToDevice, stream.s));
+  HIP_CHECK(hipMemcpyAsync(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
+  HIP_CHECK(hipMemcpyAsync(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
+  HIP_CHECK(hipStreamSynchronize(stream.s));
+
+  auto iter = [&](int parity) {
+    DevCtl* cur = ctl_d.p + parity;
+    DevCtl* next = ctl_d.p + (1 - parity);
+    launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, 1, M,
+                stream.s);
+    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, childbuf_d.p, pool_d.p, stride, G,
+                      1, M, capacity, stream.s);
+  };
+  DevLoopCfg cfg;
+  cfg.m = 1;
+  cfg.init_size = 1;
+  cfg.stop_size = target;
+  cfg.allow_graph = false;
+  const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, cfg, iter, r);
+  tree = fin.tree;
+  sol = fin.sol;
+  std::vector<NQNode> nodes(fin.size);
+  if (fin.size > 0) {
+    HIP_CHECK(hipMemcpyAsync(nodes.data(), pool_d.p, fin.size * sizeof(NQNode),
+                             hipMemcpyDeviceToHost, stream.s));
+    HIP_CHECK(hipStreamSynchronize(stream.s));
+  }
+  return nodes;
+}
+
+std::vector<PFSPNode> pfsp_gpu_frontier(const PfspInstance& I, int lbk, size_t target,
+                                        int device, int best0, uint64_t& tree,
+                                        uint64_t& sol, int& best_out) {
+  HIP_CHECK(hipSetDevice(device));
+  const PfspDevTables& tb = pfsp_tables_cached(I, device);
+  StreamGuard stream;
+  Result r;
+  const int jobs = I.jobs, machines = I.machines;
+  const unsigned long long M = target;
+  const unsigned long long capacity = target * (MAX_JOBS + 1) + 64;
+  DevGuard<PFSPNode> pool_d(capacity);
+  DevGuard<DevCtl> ctl_d(2);
+  const int G = devpool_grid(M, jobs, lbk);
+  const int stride = devpool_stride(lbk);
+  DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
+  DevGuard<uint32_t> bc_d(G);
+  DevGuard<unsigned long long> bs_d(G);
+  const bool presum = (lbk == 2);
+  DevGuard<uint32_t> gsum_d(presum ? (G + 255) / 256 : 1);
+
+  PFSPNode root = pfsp_root();
+  DevCtl ctl{};
+  ctl.size = 1;
+  ctl.best = best0;
+  HIP_CHECK(hipMemcpyAsync(pool_d.p, &root, sizeof(PFSPNode), hipMemcpyHostToDevice,
+                           stream.s));
+  HIP_CHECK(hipMemcpyAsync(ctl_d.p, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
+  HIP_CHECK(hipMemcpyAsync(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
+  HIP_CHECK(hipStreamSynchronize(stream.s));
+
+  auto iter = [&](int parity) {
+    DevCtl* cur = ctl_d.p + parity;
+    DevCtl* next = ctl_d.p + (1 - parity);
+    launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk, tb, 1,
+                  M, stream.s);
+    if (presum) launch_presum(bc_d.p, gsum_d.p, G, stream.s);
+    launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, presum ? gsum_d.p : nullptr,
+                        childbuf_d.p, pool_d.p, stride, G, 1, M, capacity, stream.s);
+  };
+  DevLoopCfg cfg;
+  cfg.m = 1;
+  cfg.init_size = 1;
+  cfg.stop_size = target;
+  cfg.allow_graph = false;
+  const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, cfg, iter, r);
+  tree = fin.tree;
+  sol = fin.sol;
+  best_out = fin.best;
+  std::vector<PFSPNode> nodes(fin.size);
+  if (fin.size > 0) {
+    HIP_CHECK(hipMemcpyAsync(nodes.data(), pool_d.p, fin.size * sizeof(PFSPNode),
+                             hipMemcpyDeviceToHost, stream.s));
+    HIP_CHECK(hipStreamSynchronize(stream.s));
+  }
+  return nodes;
 }
 
 // ---------------------------------------------------------------------------

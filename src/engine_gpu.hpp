@@ -85,6 +85,19 @@ class PfspAsyncEngine {
   std::exception_ptr err_;
 };
 
+// Device-built phase-1 frontiers: run the devpool expand from the root with
+// m=1 (each iteration pops the whole pool while it is below M — level-
+// synchronous BFS) until the pool reaches `target` nodes or the search
+// exhausts. Deterministic for a fixed (problem, target): every rank of the
+// distributed tier builds the same frontier redundantly in ~0.1-0.5 ms where
+// the CPU builder needs ~3-10 ms (pfsp_dist_multigpu_cuda.c:372-378's
+// redundant-BFS trick, moved onto the GPU).
+std::vector<NQNode> nq_gpu_frontier(int N, int g, size_t target, int device,
+                                    uint64_t& tree, uint64_t& sol);
+std::vector<PFSPNode> pfsp_gpu_frontier(const PfspInstance& I, int lbk, size_t target,
+                                        int device, int best0, uint64_t& tree,
+                                        uint64_t& sol, int& best_out);
+
 std::vector<uint8_t> nq_gpu_labels(int N, int g, const std::vector<NQNode>& nodes,
                                    int device);
 std::vector<int32_t> pfsp_gpu_bounds(int inst, const std::string& lb,
