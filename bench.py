@@ -1,11 +1,14 @@
 #!/usr/bin/env python3
 """Driver benchmark contract: flagship tree-search throughput on N MI355X GPUs.
 
-Headline metric (BASELINE.json): Mnodes/sec (whole node), N-Queens N=17 with
-the reference's default offload window m=25, M=50000. A "step" is one complete
-search of the instance (the non-neural analog of a training pass over a fixed
-input). Strong scaling: the frontier of ONE search is round-robin partitioned
-across ranks (gats_amd.dist), so total work is fixed as N grows.
+Headline metric (BASELINE.json): Mnodes/sec (whole node) on "N-Queens N=17 AND
+PFSP ta014 lb1" with the reference's default offload window m=25, M=50000. The
+default step therefore runs BOTH searches back to back (one complete N=17
+search + one complete ta014 lb1 ub=1 proof — the non-neural analog of a
+training pass over a fixed input); value = summed explored nodes / elapsed.
+--problem nqueens|pfsp isolates one config. Strong scaling: the frontier of
+each search is round-robin partitioned across ranks (gats_amd.dist), so total
+work is fixed as N grows.
 
 Launched by the driver as
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
@@ -29,7 +32,8 @@ def parse_args():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
-    ap.add_argument("--problem", default="nqueens", choices=["nqueens", "pfsp"])
+    ap.add_argument("--problem", default="headline",
+                    choices=["headline", "nqueens", "pfsp"])
     ap.add_argument("--N", type=int, default=17)
     ap.add_argument("--g", type=int, default=1)
     ap.add_argument("--inst", type=int, default=14)
@@ -67,8 +71,15 @@ def main():
         if args.problem == "nqueens":
             return gdist.run_nqueens(args.N, args.g, args.m, args.M, args.mode,
                                      engine=engine)
-        return gdist.run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode,
-                              engine=engine)
+        if args.problem == "pfsp":
+            return gdist.run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode,
+                                  engine=engine)
+        # headline: both BASELINE.json configs in one step
+        r1 = gdist.run_nqueens(args.N, args.g, args.m, args.M, args.mode, engine=engine)
+        r2 = gdist.run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode,
+                            engine=engine)
+        return {"tree": r1["tree"] + r2["tree"], "nqueens_tree": r1["tree"],
+                "pfsp_tree": r2["tree"], "optimum": r2.get("optimum", 0)}
 
     def sync():
         if world > 1:
@@ -106,10 +117,16 @@ def main():
         if args.problem == "nqueens":
             cfg = {"model": f"nqueens-N{args.N}", "N": args.N, "g": args.g, "m": args.m,
                    "M": args.M, "parallelism": f"multipool-dp{world}"}
-        else:
+        elif args.problem == "pfsp":
             cfg = {"model": f"pfsp-ta{args.inst:03d}-{args.lb}", "inst": args.inst,
                    "lb": args.lb, "ub": args.ub, "m": args.m, "M": args.M,
                    "parallelism": f"multipool-dp{world}"}
+        else:
+            cfg = {"model": f"nqueens-N{args.N}+pfsp-ta{args.inst:03d}-{args.lb}",
+                   "N": args.N, "inst": args.inst, "lb": args.lb, "ub": args.ub,
+                   "m": args.m, "M": args.M, "parallelism": f"multipool-dp{world}",
+                   "nqueens_tree_per_step": r.get("nqueens_tree"),
+                   "pfsp_tree_per_step": r.get("pfsp_tree")}
         print(json.dumps({
             "metric": "Mnodes_per_sec",
             "value": value,

@@ -33,7 +33,8 @@ def run_bench(nproc, extra):
 
 
 def test_bench_json_contract_world2():
-    out = run_bench(2, ["--steps", "2", "--warmup", "1", "--N", "10"])
+    out = run_bench(2, ["--steps", "2", "--warmup", "1", "--N", "10",
+                        "--problem", "nqueens"])
     assert out["metric"] == "Mnodes_per_sec"
     assert out["n_gpus"] == 2
     assert out["steps"] == 2
@@ -54,3 +55,14 @@ def test_bench_json_contract_world1_pfsp():
     assert out["n_gpus"] == 1
     assert out["config"]["model"] == "pfsp-ta014-lb1_d"
     assert out["explored_tree_per_step"] == 2573652
+
+
+def test_bench_headline_combines_both_configs():
+    # default --problem headline runs BOTH BASELINE.json configs per step
+    # (small stand-ins here: N=10 + ta014 lb1_d)
+    out = run_bench(1, ["--steps", "1", "--warmup", "0", "--N", "10",
+                        "--inst", "14", "--lb", "lb1_d"])
+    assert out["config"]["model"] == "nqueens-N10+pfsp-ta014-lb1_d"
+    assert out["config"]["nqueens_tree_per_step"] == 35538
+    assert out["config"]["pfsp_tree_per_step"] == 2573652
+    assert out["explored_tree_per_step"] == 35538 + 2573652

@@ -1,0 +1,152 @@
+"""Round-2 GPU coverage: capacity spill, device-built frontiers, real g work,
+multigpu shared-queue balancing, and multi-device paths (skip on 1 GPU)."""
+import os
+import subprocess
+import sys
+import time
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+ROOT = os.path.dirname(HERE)
+
+
+def test_capacity_spill_nqueens(gpu):
+    # capacity far below the peak pool size: the engine must spill to host
+    # and still produce exact counts (Pool.chpl:28-31 unbounded-growth parity)
+    seq = gpu.nqueens_seq(14, 1)
+    r = gpu.nqueens_gpu(14, 1, 25, 1000, 0, "devpool", 1 << 15)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+
+
+def test_capacity_spill_pfsp(gpu):
+    seq = gpu.pfsp_seq(14, "lb1", 1)
+    r = gpu.pfsp_gpu(14, "lb1", 1, 25, 1000, 0, "devpool", 1 << 15)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    assert r["optimum"] == 1377
+
+
+def test_gpu_frontier_nq_counts(gpu):
+    # device-built phase 1 + engine phase 2/3 must reproduce sequential totals
+    seq = gpu.nqueens_seq(13, 1)
+    nodes, tree1, sol1 = gpu.nq_gpu_frontier(13, 1, 4096, 0)
+    assert len(nodes) % 24 == 0 and len(nodes) // 24 >= 4096
+    r = gpu.nqueens_gpu_from_pool(nodes, 13, 1, 25, 50000, 0, "devpool", 1 << 24)
+    assert tree1 + r["tree"] == seq["tree"]
+    assert sol1 + r["sol"] == seq["sol"]
+
+
+def test_gpu_frontier_deterministic(gpu):
+    # every dist rank rebuilds the frontier redundantly; bytes must be identical
+    a = gpu.nq_gpu_frontier(15, 1, 8192, 0)
+    b = gpu.nq_gpu_frontier(15, 1, 8192, 0)
+    assert a == b
+    pa = gpu.pfsp_gpu_frontier(14, "lb1", 1, 2048, 0)
+    pb = gpu.pfsp_gpu_frontier(14, "lb1", 1, 2048, 0)
+    assert pa == pb
+
+
+def test_gpu_frontier_pfsp_counts(gpu):
+    seq = gpu.pfsp_seq(14, "lb2", 1)
+    nodes, tree1, sol1, best = gpu.pfsp_gpu_frontier(14, "lb2", 1, 2048, 0)
+    r = gpu.pfsp_gpu_from_pool(nodes, 14, "lb2", 1, best, 25, 50000, 0, "devpool", 1 << 24)
+    assert tree1 + r["tree"] == seq["tree"]
+    assert sol1 + r["sol"] == seq["sol"]
+    assert r["optimum"] == 1377
+
+
+def test_gpu_frontier_exhausts_small_tree(gpu):
+    # target larger than the whole tree: the builder finishes the search and
+    # returns an empty frontier with full counts
+    seq = gpu.nqueens_seq(8, 1)
+    nodes, tree1, sol1 = gpu.nq_gpu_frontier(8, 1, 1 << 20, 0)
+    r = gpu.nqueens_gpu_from_pool(nodes, 8, 1, 25, 50000, 0, "devpool", 1 << 22)
+    assert tree1 + r["tree"] == seq["tree"]
+    assert sol1 + r["sol"] == seq["sol"]
+
+
+def test_g_knob_scales_devpool_time(gpu):
+    # g repeats every safety evaluation; devpool kernel time must scale with
+    # it (the round-1 knob was dead code). Generous threshold: g=8 >= 2x g=1.
+    seq = gpu.nqueens_seq(15, 1)
+    t0 = time.perf_counter()
+    r1 = gpu.nqueens_gpu(15, 1, 25, 50000, 0, "devpool", 1 << 26)
+    t1 = time.perf_counter() - t0
+    t0 = time.perf_counter()
+    r8 = gpu.nqueens_gpu(15, 8, 25, 50000, 0, "devpool", 1 << 26)
+    t8 = time.perf_counter() - t0
+    assert r1["tree"] == r8["tree"] == seq["tree"]
+    assert r1["sol"] == r8["sol"] == seq["sol"]
+    assert t8 > 2.0 * t1, (t1, t8)
+
+
+def test_multigpu_shared_queue_skewed(gpu):
+    # shared-queue balancing: with 4 workers on a deliberately small worker
+    # count vs slice count, all workers report work and counts stay exact
+    seq = gpu.nqueens_seq(15, 1)
+    r = gpu.nqueens_multigpu(15, 1, 25, 50000, 4, "devpool")
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    assert len(r["per_worker_tree"]) == 4
+    assert all(w > 0 for w in r["per_worker_tree"])
+
+
+def test_multigpu_capacity_flag(gpu):
+    # --capacity reaches the multigpu tier (was hardwired); small value forces
+    # spilling inside workers, counts stay exact
+    seq = gpu.nqueens_seq(14, 1)
+    r = gpu.nqueens_multigpu(14, 1, 25, 1000, 2, "devpool", 0.5, 1 << 15)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+
+
+# ---------------- multi-device (driver may land on an 8-GPU node) ----------
+
+
+def two_gpus(core):
+    return core.gpu_device_count() >= 2
+
+
+def test_multigpu_two_devices(gpu):
+    if not two_gpus(gpu):
+        pytest.skip("needs >= 2 HIP devices")
+    seq = gpu.nqueens_seq(15, 1)
+    r = gpu.nqueens_multigpu(15, 1, 25, 50000, 2, "devpool")
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    assert all(w > 0 for w in r["per_worker_tree"])
+
+
+def test_pfsp_multigpu_two_devices(gpu):
+    if not two_gpus(gpu):
+        pytest.skip("needs >= 2 HIP devices")
+    seq = gpu.pfsp_seq(14, "lb2", 1)
+    r = gpu.pfsp_multigpu(14, "lb2", 1, 25, 50000, 2, "devpool", False)
+    assert r["tree"] == seq["tree"]
+    assert r["optimum"] == 1377
+
+
+def test_dist_world2_rccl(gpu):
+    # the real dist tier over RCCL: one process per GPU, nccl backend
+    if not two_gpus(gpu):
+        pytest.skip("needs >= 2 HIP devices")
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node=2",
+        "--standalone", "--local-addr", "127.0.0.1",
+        os.path.join(HERE, "helpers", "dist_gpu_check.py"),
+    ]
+    for attempt in range(2):
+        r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env,
+                           cwd=ROOT)
+        if r.returncode == 0:
+            break
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "DIST_GPU_OK" in r.stdout
