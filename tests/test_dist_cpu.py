@@ -55,7 +55,8 @@ def test_cli_dist_tier_cpu():
     ]
     r = run_torchrun(cmd, env)
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
-    assert "CLI_DIST_OK" in r.stdout
-    assert "Size of the explored tree: 35538" in r.stdout   # N=10 exact
-    assert "Size of the explored tree: 2573652" in r.stdout  # ta014 lb1_d ub1
-    assert "Optimal makespan: 1377 (not improved)" in r.stdout
+    out = f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "CLI_DIST_OK" in r.stdout, out
+    assert "Size of the explored tree: 35538" in r.stdout, out   # N=10 exact
+    assert "Size of the explored tree: 2573652" in r.stdout, out  # ta014 lb1_d ub1
+    assert "Optimal makespan: 1377 (not improved)" in r.stdout, out
