@@ -161,7 +161,7 @@ def run_nqueens(N, g=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
         # bigger (costlier) frontier buys nothing
         frontier_target = 65536
     local = rank % max(1, c.gpu_device_count())
-    if engine == "gpu":
+    if engine == "gpu" and os.environ.get("GATS_CPU_FRONTIER") != "1":
         # device-built frontier (~0.2 ms vs ~8 ms CPU at 65536); every rank
         # builds it redundantly and deterministically
         nodes, tree1, sol1 = c.nq_gpu_frontier(N, g, frontier_target, local)
@@ -197,7 +197,7 @@ def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 
         # would move a large share of the search onto the single-threaded CPU
         frontier_target = max(2048, 2048 * world)
     local = rank % max(1, c.gpu_device_count())
-    if engine == "gpu":
+    if engine == "gpu" and os.environ.get("GATS_CPU_FRONTIER") != "1":
         nodes, tree1, sol1, best = c.pfsp_gpu_frontier(inst, lb, ub, frontier_target, local)
     else:
         nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)

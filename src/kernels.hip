@@ -113,17 +113,17 @@ __device__ inline uint8_t nq_safe(const uint8_t* board, int depth, int q, int g)
   return safe;
 }
 
-// Mask-based safety test repeated g times (devpool path's g semantics: the
-// per-child evaluation is the O(1) mask test, so g scales exactly that).
-__device__ inline bool nq_mask_unsafe(uint32_t b, uint32_t occ, int g) {
-  if (g == 1) return (b & occ) != 0;
-  bool unsafe = false;
-  for (int r = 0; r < g; r++) {
-    uint32_t o = occ;
-    asm volatile("" : "+v"(o));  // force a real re-test per repeat
-    unsafe |= (b & o) != 0;
+// Occupancy mask re-evaluated g times as a LIVE dependency chain (devpool
+// path's g semantics: the per-node evaluation is the O(1) mask combine, so g
+// scales exactly that). Each repeat's asm makes the mask opaque, so the next
+// repeat's ORs must really execute; the value is the identity cols|d1|d2.
+__device__ inline uint32_t nq_occ_g(uint32_t cols, uint32_t d1, uint32_t d2, int g) {
+  uint32_t occ = cols | d1 | d2;
+  for (int r = 1; r < g; r++) {
+    asm volatile("" : "+v"(occ));
+    occ |= cols | d1 | d2;  // identity; non-foldable through the barrier
   }
-  return unsafe;
+  return occ;
 }
 
 // hostpool mode: one thread per (parent, k), labels out
@@ -487,7 +487,10 @@ __device__ inline unsigned long long block_reduce_u64(unsigned long long v) {
 // values) and safety is the two diagonal masks — identical node set, and
 // N-Queens counts are traversal-order independent. Template depth budget B
 // keeps every stack frame in registers (no runtime-indexed arrays).
-template <int B>
+// G1=true is the hot path (no g argument threaded through the recursion at
+// all — a runtime g branch per node measured ~30% on N=17); G1=false repeats
+// every node's mask evaluation g times via nq_occ_g.
+template <int B, bool G1>
 __device__ inline void nq_dfs(uint32_t cols, uint32_t d1, uint32_t d2, int placed, int N,
                               int g, unsigned long long& tree, unsigned long long& sol) {
   if constexpr (B == 0) {
@@ -495,18 +498,10 @@ __device__ inline void nq_dfs(uint32_t cols, uint32_t d1, uint32_t d2, int place
   } else {
     const uint32_t msk = (1u << N) - 1u;
     uint32_t free;
-    if (g == 1) {
+    if constexpr (G1)
       free = ~(cols | d1 | d2) & msk;
-    } else {
-      // repeat the node's whole safety evaluation g times (reference knob);
-      // barrier keeps each repeat from being folded away
-      free = 0;
-      for (int r = 0; r < g; r++) {
-        uint32_t occ = cols | d1 | d2;
-        asm volatile("" : "+v"(occ));
-        free = ~occ & msk;
-      }
-    }
+    else
+      free = ~nq_occ_g(cols, d1, d2, g) & msk;
     while (free) {
       const uint32_t bit = free & (0u - free);
       free ^= bit;
@@ -514,8 +509,8 @@ __device__ inline void nq_dfs(uint32_t cols, uint32_t d1, uint32_t d2, int place
       if (placed + 1 == N) {
         sol++;
       } else {
-        nq_dfs<B - 1>(cols | bit, ((d1 | bit) << 1) & msk, (d2 | bit) >> 1, placed + 1, N,
-                      g, tree, sol);
+        nq_dfs<B - 1, G1>(cols | bit, ((d1 | bit) << 1) & msk, (d2 | bit) >> 1, placed + 1,
+                          N, g, tree, sol);
       }
     }
   }
@@ -539,6 +534,7 @@ __device__ inline unsigned long long derive_chunk(const DevCtl* ctl, unsigned lo
 // A phase-A pass computes each PARENT's (cols, d1, d2) diagonal masks once
 // (one thread per staged parent), so a child's safety test and the finisher's
 // starting masks are O(1) instead of an O(depth) walk per child.
+template <bool G1>
 __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                        uint32_t* blockCounts, unsigned long long* blockSols,
                        unsigned long long* blockExtra, int N, int g, int finish,
@@ -597,7 +593,8 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
           const uint32_t d1 = pmask[pid - first][1];
           const uint32_t d2 = pmask[pid - first][2];
           const uint32_t b = 1u << p.board[k];
-          if (!nq_mask_unsafe(b, cols | d1 | d2, g)) {  // == nq_safe (diag masks)
+          const uint32_t occ = G1 ? (cols | d1 | d2) : nq_occ_g(cols, d1, d2, g);
+          if (!(b & occ)) {  // == nq_safe (diagonal masks)
             const int rem = N - (depth + 1);  // levels below the child
             if (rem <= finish) {
               // child + its whole subtree counted here, nothing pushed
@@ -606,8 +603,8 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                 sols += 1;
               } else {
                 const uint32_t msk = (1u << N) - 1u;
-                nq_dfs<NQ_FINISH_MAX>(cols | b, ((d1 | b) << 1) & msk, (d2 | b) >> 1,
-                                      depth + 1, N, g, extra, sols);
+                nq_dfs<NQ_FINISH_MAX, G1>(cols | b, ((d1 | b) << 1) & msk, (d2 | b) >> 1,
+                                          depth + 1, N, g, extra, sols);
               }
             } else {
               lab[j] = 1;
@@ -1075,8 +1072,14 @@ void launch_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
                  uint32_t* blockCounts, unsigned long long* blockSols,
                  unsigned long long* blockExtra, int N, int g, int finish,
                  unsigned long long m, unsigned long long M, hipStream_t s) {
-  hipLaunchKernelGGL(k_nq_x, dim3(devpool_grid(M, N, 1)), dim3(BLOCK), 0, s, ctl, pool,
-                     childbuf, blockCounts, blockSols, blockExtra, N, g, finish, m, M);
+  if (g == 1)
+    hipLaunchKernelGGL((k_nq_x<true>), dim3(devpool_grid(M, N, 1)), dim3(BLOCK), 0, s, ctl,
+                       pool, childbuf, blockCounts, blockSols, blockExtra, N, g, finish, m,
+                       M);
+  else
+    hipLaunchKernelGGL((k_nq_x<false>), dim3(devpool_grid(M, N, 1)), dim3(BLOCK), 0, s, ctl,
+                       pool, childbuf, blockCounts, blockSols, blockExtra, N, g, finish, m,
+                       M);
 }
 
 template <int MM>
