@@ -111,12 +111,15 @@ def _run_dynamic(nodes, world, per_rank_slices, run_slice):
 
 def slice_frontier(nodes: bytes, rank: int, world: int) -> bytes:
     """Round-robin node slice (rank, rank+world, ...), the reference's static
-    interleaved partition (nqueens_dist_multigpu_chpl.chpl:223-227)."""
-    n = len(nodes) // NODE_BYTES
-    out = bytearray()
-    for i in range(rank, n, world):
-        out += nodes[i * NODE_BYTES:(i + 1) * NODE_BYTES]
-    return bytes(out)
+    interleaved partition (nqueens_dist_multigpu_chpl.chpl:223-227).
+    Vectorized: a Python per-node loop cost milliseconds per claim on the
+    ~300k-node N=17 frontier."""
+    if world == 1:
+        return nodes
+    import numpy as np
+
+    a = np.frombuffer(nodes, dtype=np.uint8).reshape(-1, NODE_BYTES)
+    return a[rank::world].tobytes()
 
 
 def _backend_device(device):

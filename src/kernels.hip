@@ -113,17 +113,20 @@ __device__ inline uint8_t nq_safe(const uint8_t* board, int depth, int q, int g)
   return safe;
 }
 
-// Occupancy mask re-evaluated g times as a LIVE dependency chain (devpool
-// path's g semantics: the per-node evaluation is the O(1) mask combine, so g
-// scales exactly that). Each repeat's asm makes the mask opaque, so the next
-// repeat's ORs must really execute; the value is the identity cols|d1|d2.
-__device__ inline uint32_t nq_occ_g(uint32_t cols, uint32_t d1, uint32_t d2, int g) {
-  uint32_t occ = cols | d1 | d2;
-  for (int r = 1; r < g; r++) {
+// Free-column mask evaluated g times as a LIVE dependency chain (devpool
+// path's g semantics: the per-node safety evaluation is the mask combine +
+// negate, so g repeats exactly that). Each repeat's asm makes `occ` opaque,
+// so the ORs/ANDN must really execute every round; the result is the
+// identity ~(cols|d1|d2) & msk.
+__device__ inline uint32_t nq_free_g(uint32_t cols, uint32_t d1, uint32_t d2, uint32_t msk,
+                                     int g) {
+  uint32_t free = 0, occ = cols | d1 | d2;
+  for (int r = 0; r < g; r++) {
     asm volatile("" : "+v"(occ));
     occ |= cols | d1 | d2;  // identity; non-foldable through the barrier
+    free |= ~occ & msk;     // idempotent accumulate keeps every repeat live
   }
-  return occ;
+  return free;
 }
 
 // hostpool mode: one thread per (parent, k), labels out
@@ -501,7 +504,7 @@ __device__ inline void nq_dfs(uint32_t cols, uint32_t d1, uint32_t d2, int place
     if constexpr (G1)
       free = ~(cols | d1 | d2) & msk;
     else
-      free = ~nq_occ_g(cols, d1, d2, g) & msk;
+      free = nq_free_g(cols, d1, d2, msk, g);
     while (free) {
       const uint32_t bit = free & (0u - free);
       free ^= bit;
@@ -593,8 +596,10 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
           const uint32_t d1 = pmask[pid - first][1];
           const uint32_t d2 = pmask[pid - first][2];
           const uint32_t b = 1u << p.board[k];
-          const uint32_t occ = G1 ? (cols | d1 | d2) : nq_occ_g(cols, d1, d2, g);
-          if (!(b & occ)) {  // == nq_safe (diagonal masks)
+          const uint32_t msk = (1u << N) - 1u;
+          const uint32_t freemask =
+              G1 ? (~(cols | d1 | d2) & msk) : nq_free_g(cols, d1, d2, msk, g);
+          if (b & freemask) {  // == nq_safe (diagonal masks)
             const int rem = N - (depth + 1);  // levels below the child
             if (rem <= finish) {
               // child + its whole subtree counted here, nothing pushed
@@ -602,7 +607,6 @@ __global__ void k_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,
               if (rem == 0) {
                 sols += 1;
               } else {
-                const uint32_t msk = (1u << N) - 1u;
                 nq_dfs<NQ_FINISH_MAX, G1>(cols | b, ((d1 | b) << 1) & msk, (d2 | b) >> 1,
                                           depth + 1, N, g, extra, sols);
               }
