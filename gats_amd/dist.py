@@ -278,6 +278,224 @@ def run_pfsp_shared_ub(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
     return _reduce_stats(r, phase1, f"cuda:{local}", world, best=True)
 
 
+class CpuPfspEngine:
+    """CPU twin of the C++ PfspAsyncEngine (same protocol surface) so the
+    live-steal protocol is CI-testable on gloo without a GPU: a daemon thread
+    drains the queued frontier one node at a time (so extraction can always
+    interrupt between subtrees); extraction hands over half the remaining
+    queue. Counts are exact because nodes move, never copy."""
+
+    def __init__(self, core, inst, lb, ub, best0):
+        import threading
+
+        self.c, self.inst, self.lb, self.ub = core, inst, lb, ub
+        self._best = best0 if best0 > 0 else (1 << 30)
+        self._lock = threading.Lock()
+        self._q = []
+        self._tree = 0
+        self._sol = 0
+        self._busy = False
+        self._stop = False
+        self._want = False
+        self._ready = None  # bytes once an extract request is answered
+        self._th = threading.Thread(target=self._loop, daemon=True)
+        self._th.start()
+
+    def _loop(self):
+        import time as _t
+
+        nb = NODE_BYTES
+        while True:
+            node = None
+            with self._lock:
+                if self._want:
+                    half = len(self._q) // 2
+                    if half:
+                        self._ready = b"".join(self._q[-half:])
+                        del self._q[-half:]
+                    else:
+                        self._ready = b""
+                    self._want = False
+                if self._q:
+                    node = self._q.pop()
+                    self._busy = True
+                else:
+                    self._busy = False
+                    if self._stop:
+                        return
+            if node is None:
+                _t.sleep(0.001)
+                continue
+            r = self.c.pfsp_seq_from_pool(node, self.inst, self.lb, self.ub, self._best)
+            with self._lock:
+                self._tree += r["tree"]
+                self._sol += r["sol"]
+                if r.get("optimum"):
+                    self._best = min(self._best, r["optimum"])
+
+    def submit(self, nodes: bytes, best0: int = 0):
+        nb = NODE_BYTES
+        parts = [nodes[i:i + nb] for i in range(0, len(nodes), nb)]
+        with self._lock:
+            if best0 > 0:
+                self._best = min(self._best, best0)
+            self._q.extend(parts)
+            if parts:
+                self._busy = True
+
+    def best(self):
+        with self._lock:
+            return self._best
+
+    def update_best(self, b):
+        with self._lock:
+            if 0 < b < self._best:
+                self._best = b
+
+    def done(self):
+        with self._lock:
+            return not self._busy and not self._q
+
+    def pool_size(self):
+        with self._lock:
+            return len(self._q)
+
+    def request_extract(self):
+        with self._lock:
+            if not self._q and not self._busy:
+                self._ready = b""
+            else:
+                self._want = True
+
+    def extract_ready(self):
+        with self._lock:
+            return self._ready is not None
+
+    def extract_pending(self):
+        with self._lock:
+            return self._want
+
+    def take_extract(self):
+        with self._lock:
+            out, self._ready = self._ready, None
+            return out or b""
+
+    def join(self):
+        import time as _t
+
+        with self._lock:
+            self._stop = True
+        while self._th.is_alive():
+            self._th.join(timeout=0.05)
+        with self._lock:
+            return {"tree": self._tree, "sol": self._sol, "optimum": self._best,
+                    "time": 0.0, "diag": {}}
+
+
+def run_pfsp_live(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
+                  frontier_target=None, poll_s=0.005, engine="gpu"):
+    """Distributed PFSP with BOTH mid-search incumbent exchange and
+    engine-pausing inter-rank work stealing (the reference dist tier's remote
+    half-pool steal, nqueens_dist_multigpu_chpl.chpl:332-377, as an explicit
+    fixed-cadence protocol over RCCL):
+
+    every round, every rank joins ONE all_gather of its [best, done, pool,
+    ready, pending] state, then all ranks derive the SAME transfer pairing
+    (ready donors -> idle thieves, sorted) and steal-request assignment
+    (largest busy pools -> remaining idle ranks) from that snapshot — so the
+    collective schedule is identical everywhere and no rank can block. Nodes
+    MOVE between engines (donor carves half its device pool at a readback
+    boundary), so counts stay exact at ub=1."""
+    import time
+
+    c = gats_amd.core()
+    rank, world = init_dist()
+    if frontier_target is None:
+        frontier_target = max(2048, 2048 * world)
+    local = rank % max(1, c.gpu_device_count())
+    if engine == "gpu" and os.environ.get("GATS_CPU_FRONTIER") != "1":
+        nodes, tree1, sol1, best = c.pfsp_gpu_frontier(inst, lb, ub, frontier_target, local)
+    else:
+        nodes, tree1, sol1, best = c.pfsp_bfs_frontier(inst, lb, ub, frontier_target)
+    my = slice_frontier(nodes, rank, world)
+    t0 = time.perf_counter()
+    if engine == "gpu":
+        eng = c.PfspAsyncEngine(inst, lb, ub, m, M, local, capacity)
+    else:
+        eng = CpuPfspEngine(c, inst, lb, ub, best)
+    eng.submit(my, best)
+    steals = 0
+    if world > 1:
+        dev = _backend_device(f"cuda:{local}" if engine == "gpu" else "cpu")
+        STEAL_MIN = max(4 * m, 256)  # don't move trivial pools
+        gathered = [torch.zeros(5, dtype=torch.int64, device=dev) for _ in range(world)]
+        while True:
+            st = torch.tensor([eng.best(), 1 if eng.done() else 0,
+                               int(eng.pool_size()),
+                               1 if eng.extract_ready() else 0,
+                               1 if eng.extract_pending() else 0],
+                              dtype=torch.int64, device=dev)
+            td.all_gather(gathered, st)
+            snap = [[int(x) for x in g] for g in gathered]
+            eng.update_best(min(s[0] for s in snap))
+            if all(s[1] == 1 for s in snap) and not any(s[3] or s[4] for s in snap):
+                break
+            # transfers: ready donors -> idle thieves, same pairing everywhere
+            donors = [i for i, s in enumerate(snap) if s[3] == 1]
+            thieves = [i for i, s in enumerate(snap) if s[1] == 1]
+            paired_thieves = set()
+            for d, t in zip(donors, thieves):
+                paired_thieves.add(t)
+                if d == t:
+                    if rank == d:
+                        payload = eng.take_extract()
+                        if payload:
+                            eng.submit(payload, eng.best())
+                    continue
+                if rank == d:
+                    payload = eng.take_extract()
+                    n = torch.tensor([len(payload)], dtype=torch.int64, device=dev)
+                    td.send(n, dst=t)
+                    if len(payload):
+                        buf = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
+                        td.send(buf.to(dev), dst=t)
+                elif rank == t:
+                    n = torch.zeros(1, dtype=torch.int64, device=dev)
+                    td.recv(n, src=d)
+                    if int(n.item()):
+                        buf = torch.zeros(int(n.item()), dtype=torch.uint8, device=dev)
+                        td.recv(buf, src=d)
+                        eng.submit(bytes(buf.cpu().numpy().tobytes()), eng.best())
+                        steals += 1
+            # new steal requests: hungriest idle ranks target the fattest
+            # running pools (one request per victim outstanding)
+            hungry = [i for i, s in enumerate(snap)
+                      if s[1] == 1 and i not in paired_thieves]
+            victims = sorted((i for i, s in enumerate(snap)
+                              if s[1] == 0 and s[2] >= STEAL_MIN
+                              and not s[3] and not s[4]),
+                             key=lambda i: -snap[i][2])
+            for v, h in zip(victims, hungry):
+                if rank == v:
+                    eng.request_extract()
+            time.sleep(poll_s)
+    r = eng.join()
+    elapsed = time.perf_counter() - t0
+    r = dict(r)
+    r["time"] = elapsed
+    if world == 1:
+        r["tree"] += tree1
+        r["sol"] += sol1
+        return r
+    phase1 = {"tree": tree1, "sol": sol1, "time": 0.0}
+    dev_name = f"cuda:{local}" if engine == "gpu" else "cpu"
+    out = _reduce_stats(r, phase1, dev_name, world, best=True)
+    s = torch.tensor([steals], dtype=torch.int64, device=_backend_device(dev_name))
+    td.all_reduce(s, op=td.ReduceOp.SUM)
+    out["steals"] = int(s.item())
+    return out
+
+
 def run_from_cli(args):
     """Entry for `gats-amd ... --tier dist` under torchrun; rank 0 returns the
     combined stats dict, other ranks return None. GATS_DIST_ENGINE=cpu runs
@@ -286,12 +504,15 @@ def run_from_cli(args):
     falls back to the end-of-search min-reduce)."""
     rank, world = init_dist()
     engine = os.environ.get("GATS_DIST_ENGINE", "gpu")
+    live = os.environ.get("GATS_DIST_LIVE") == "1"
     if args.problem == "nqueens":
         r = run_nqueens(args.N, args.g, args.m, args.M, args.mode, args.capacity,
                         engine=engine)
-    elif args.ub == 0 and engine == "gpu":
-        # open upper bound: exchange the incumbent over RCCL during the search
-        r = run_pfsp_shared_ub(args.inst, args.lb, args.ub, args.m, args.M, args.capacity)
+    elif args.ub == 0 or live:
+        # open upper bound (or opt-in): persistent engines + mid-search RCCL
+        # incumbent exchange + engine-pausing inter-rank stealing
+        r = run_pfsp_live(args.inst, args.lb, args.ub, args.m, args.M, args.capacity,
+                          engine=engine)
     else:
         r = run_pfsp(args.inst, args.lb, args.ub, args.m, args.M, args.mode, args.capacity,
                      engine=engine)

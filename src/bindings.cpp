@@ -378,8 +378,10 @@ PYBIND11_MODULE(_core, mod) {
   mod.def("pool_selftest", &pool_selftest);
 
   py::class_<PfspAsyncEngine>(mod, "PfspAsyncEngine",
-                              "Background PFSP devpool search with a shared incumbent "
-                              "(mid-search RCCL UB exchange hook)")
+                              "Persistent per-rank PFSP devpool engine: successive "
+                              "frontier submits, shared incumbent (mid-search RCCL UB "
+                              "exchange), engine-pausing work extraction for inter-rank "
+                              "steals")
       .def(py::init([](const py::bytes& nodes, int inst, const std::string& lb, int ub,
                        int best0, int m, int M, int device, unsigned long long capacity) {
              return new PfspAsyncEngine(nodes_from_bytes<PFSPNode>(nodes), inst, lb, ub,
@@ -388,9 +390,30 @@ PYBIND11_MODULE(_core, mod) {
            py::arg("nodes"), py::arg("inst"), py::arg("lb") = "lb1", py::arg("ub") = 1,
            py::arg("best0") = 0, py::arg("m") = 25, py::arg("M") = 50000,
            py::arg("device") = 0, py::arg("capacity") = (1ull << 24))
+      .def(py::init([](int inst, const std::string& lb, int ub, int m, int M, int device,
+                       unsigned long long capacity) {
+             return new PfspAsyncEngine(inst, lb, ub, m, M, device, capacity);
+           }),
+           py::arg("inst"), py::arg("lb") = "lb1", py::arg("ub") = 1, py::arg("m") = 25,
+           py::arg("M") = 50000, py::arg("device") = 0,
+           py::arg("capacity") = (1ull << 24))
+      .def("submit",
+           [](PfspAsyncEngine& e, const py::bytes& nodes, int best0) {
+             e.submit(nodes_from_bytes<PFSPNode>(nodes), best0);
+           },
+           py::arg("nodes"), py::arg("best0") = 0)
       .def("best", &PfspAsyncEngine::best)
       .def("update_best", &PfspAsyncEngine::update_best)
       .def("done", &PfspAsyncEngine::done)
+      .def("pool_size", &PfspAsyncEngine::pool_size)
+      .def("request_extract", &PfspAsyncEngine::request_extract)
+      .def("extract_ready", &PfspAsyncEngine::extract_ready)
+      .def("extract_pending", &PfspAsyncEngine::extract_pending)
+      .def("take_extract",
+           [](PfspAsyncEngine& e) {
+             auto v = e.take_extract();
+             return nodes_to_bytes(v.data(), v.size());
+           })
       .def("join", [](PfspAsyncEngine& e) {
         Result r;
         {

@@ -785,7 +785,8 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
                                     int lbk, int best0,
                                     int m, int M, int device, unsigned long long capacity,
                                     std::atomic<int>* shared_best, bool allow_graph,
-                                    SliceShare* share, std::vector<PFSPNode>& leftover) {
+                                    SliceShare* share, std::vector<PFSPNode>& leftover,
+                                    ExtractShare* extract) {
   HIP_CHECK(hipSetDevice(device));
   const PfspDevTables& tb = pfsp_tables_cached(I, device);
   StreamGuard stream;
@@ -818,6 +819,32 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   };
   ReadbackHook hook = [&](DevCtl* hc, DevCtl* live) {
     donate_if_wanted(share, hc, live, pool_d.p, m, stream.s);
+    if (extract) {
+      extract->live_size.store(hc->size, std::memory_order_relaxed);
+      if (extract->want.load(std::memory_order_relaxed) &&
+          hc->size >= 2 * static_cast<unsigned long long>(m) &&
+          extract->want.exchange(false)) {
+        // engine-pausing inter-rank steal: carve the back half to the host
+        // at this (stream-idle) readback boundary; nodes MOVE, never copy,
+        // so counts stay exact
+        const unsigned long long half = hc->size / 2;
+        const unsigned long long newsize = hc->size - half;
+        {
+          std::lock_guard<std::mutex> l(extract->mu);
+          extract->taken.resize(half);
+          HIP_CHECK(hipMemcpyAsync(extract->taken.data(), pool_d.p + newsize,
+                                   half * sizeof(PFSPNode), hipMemcpyDeviceToHost,
+                                   stream.s));
+        }
+        HIP_CHECK(hipMemcpyAsync(&live->size, &newsize, sizeof(newsize),
+                                 hipMemcpyHostToDevice, stream.s));
+        HIP_CHECK(hipStreamSynchronize(stream.s));
+        hc->size = newsize;
+        r.d2h++;
+        r.d2h_bytes += half * sizeof(PFSPNode);
+        extract->ready.store(true, std::memory_order_release);
+      }
+    }
   };
   ReadbackHook spill = [&](DevCtl* hc, DevCtl* live) {
     const unsigned long long half = hc->size / 2;
@@ -995,7 +1022,8 @@ DevpoolMultiOut pfsp_devpool_multi(const PfspInstance& I, Pool<PFSPNode>& pool, 
                                    int best0, int m, int M,
                                    const std::vector<int>& devices,
                                    unsigned long long capacity,
-                                   std::atomic<int>* shared_best, Result& r) {
+                                   std::atomic<int>* shared_best, Result& r,
+                                   ExtractShare* extract) {
   const int jobs = I.jobs;
   if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
     throw std::invalid_argument("devpool requires M * jobs <= 2^31");
@@ -1031,7 +1059,7 @@ DevpoolMultiOut pfsp_devpool_multi(const PfspInstance& I, Pool<PFSPNode>& pool, 
       SliceShare* sh = (S > 1) ? &shares[t / S] : nullptr;
       outs[t] = devpool_thread_pfsp(slices, next_slice, I, lbk, best0, m, M,
                                     devices[t / S], capacity, sb, allow_graph, sh,
-                                    lefts[t]);
+                                    lefts[t], extract);
     } catch (...) {
       errs[t] = std::current_exception();
     }
@@ -1151,7 +1179,8 @@ Result nqueens_gpu_from_pool(const std::vector<NQNode>& nodes, int N, int g, int
 Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int m, int M,
                     int device, const std::string& mode, uint64_t tree0, uint64_t sol0,
                     int best0, double phase1_time, unsigned long long capacity,
-                    std::atomic<int>* shared_best /*= nullptr*/) {
+                    std::atomic<int>* shared_best /*= nullptr*/,
+                    ExtractShare* extract /*= nullptr*/) {
   Result r;
   r.phases.push_back({tree0, sol0, phase1_time});
   uint64_t tree = tree0, sol = sol0;
@@ -1188,8 +1217,8 @@ Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int 
       pfsp_generate_children(I, parents.p, n, bounds.p, tree, sol, best, pool);
     }
   } else if (mode == "devpool") {
-    DevpoolMultiOut o =
-        pfsp_devpool_multi(I, pool, lbk, best, m, M, {device}, capacity, shared_best, r);
+    DevpoolMultiOut o = pfsp_devpool_multi(I, pool, lbk, best, m, M, {device}, capacity,
+                                           shared_best, r, extract);
     tree += o.tree;
     sol += o.sol;
     if (o.best < best) best = o.best;
@@ -1423,29 +1452,91 @@ std::vector<int32_t> pfsp_gpu_bounds(int inst, const std::string& lb_str,
 // always sound since any incumbent >= the optimum is a valid UB).
 // ---------------------------------------------------------------------------
 
+PfspAsyncEngine::PfspAsyncEngine(int inst, const std::string& lb_str, int ub, int m,
+                                 int M, int device, unsigned long long capacity)
+    : inst_(inst), ub_(ub), m_(m), M_(M), device_(device), lb_(lb_str),
+      capacity_(capacity), shared_best_(0) {
+  const PfspInstance& I = pfsp_instance_cached(inst, ub);
+  shared_best_.store(I.init_ub);
+  result_.optimum = I.init_ub;
+  th_ = std::thread([this] { loop(); });
+}
+
 PfspAsyncEngine::PfspAsyncEngine(std::vector<PFSPNode> nodes, int inst,
                                  const std::string& lb_str, int ub, int best0, int m, int M,
                                  int device, unsigned long long capacity)
-    : shared_best_(0), done_(false) {
-  const LbKind lb = lb_from_string(lb_str);
-  const PfspInstance& I = pfsp_instance_cached(inst, ub);
-  const int b0 = (best0 > 0) ? best0 : I.init_ub;
-  shared_best_.store(b0);
-  th_ = std::thread([this, nodes = std::move(nodes), &I, lb, m, M, device, capacity,
-                     b0]() mutable {
-    try {
+    : PfspAsyncEngine(inst, lb_str, ub, m, M, device, capacity) {
+  submit(std::move(nodes), best0);
+}
+
+void PfspAsyncEngine::loop() {
+  try {
+    const PfspInstance& I = pfsp_instance_cached(inst_, ub_);
+    const LbKind lb = lb_from_string(lb_);
+    while (true) {
+      std::vector<PFSPNode> nodes;
+      int b0 = 0;
+      {
+        std::unique_lock<std::mutex> l(mu_);
+        cv_.wait(l, [&] { return finish_ || !q_.empty(); });
+        if (q_.empty()) break;  // finish requested and drained
+        nodes = std::move(q_.front().first);
+        b0 = q_.front().second;
+        q_.pop_front();
+        idle_.store(false, std::memory_order_release);
+      }
+      queued_nodes_.fetch_sub(nodes.size(), std::memory_order_relaxed);
       Pool<PFSPNode> pool;
       pool.pushBackBulk(nodes.data(), nodes.size());
-      result_ = pfsp_gpu_run(I, lb, pool, m, M, device, "devpool", 0, 0, b0, 0.0, capacity,
-                             &shared_best_);
-    } catch (...) {
-      err_ = std::current_exception();
+      const int start = (b0 > 0) ? b0 : I.init_ub;
+      update_best(start);  // non-increasing incumbent across submits
+      Result r = pfsp_gpu_run(I, lb, pool, m_, M_, device_, "devpool", 0, 0,
+                              shared_best_.load(std::memory_order_relaxed), 0.0,
+                              capacity_, &shared_best_, &ex_);
+      result_.tree += r.tree;
+      result_.sol += r.sol;
+      if (r.optimum > 0 &&
+          (result_.optimum == 0 || r.optimum < result_.optimum))
+        result_.optimum = r.optimum;
+      result_.kernel_launch += r.kernel_launch;
+      result_.h2d += r.h2d;
+      result_.d2h += r.d2h;
+      result_.h2d_bytes += r.h2d_bytes;
+      result_.d2h_bytes += r.d2h_bytes;
+      result_.gpu_iters += r.gpu_iters;
+      {
+        std::lock_guard<std::mutex> l(mu_);
+        if (q_.empty()) {
+          idle_.store(true, std::memory_order_release);
+          ex_.live_size.store(0, std::memory_order_relaxed);
+          // an unanswered steal request meets an empty engine: answer "none"
+          if (ex_.want.exchange(false)) ex_.ready.store(true, std::memory_order_release);
+        }
+      }
     }
-    done_.store(true, std::memory_order_release);
-  });
+    // finishing with a pending request: answer empty so no rank blocks
+    if (ex_.want.exchange(false)) ex_.ready.store(true, std::memory_order_release);
+  } catch (...) {
+    err_ = std::current_exception();
+    idle_.store(true, std::memory_order_release);
+    if (ex_.want.exchange(false)) ex_.ready.store(true, std::memory_order_release);
+  }
+}
+
+void PfspAsyncEngine::submit(std::vector<PFSPNode> nodes, int best0) {
+  queued_nodes_.fetch_add(nodes.size(), std::memory_order_relaxed);
+  std::lock_guard<std::mutex> l(mu_);
+  idle_.store(false, std::memory_order_release);
+  q_.emplace_back(std::move(nodes), best0);
+  cv_.notify_one();
 }
 
 PfspAsyncEngine::~PfspAsyncEngine() {
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    finish_ = true;
+    cv_.notify_all();
+  }
   if (th_.joinable()) th_.join();
 }
 
@@ -1458,9 +1549,63 @@ void PfspAsyncEngine::update_best(int b) {
   }
 }
 
-bool PfspAsyncEngine::done() const { return done_.load(std::memory_order_acquire); }
+bool PfspAsyncEngine::done() const {
+  return idle_.load(std::memory_order_acquire) &&
+         queued_nodes_.load(std::memory_order_relaxed) == 0;
+}
+
+unsigned long long PfspAsyncEngine::pool_size() const {
+  const unsigned long long live =
+      done() ? 0 : ex_.live_size.load(std::memory_order_relaxed);
+  return live + queued_nodes_.load(std::memory_order_relaxed);
+}
+
+void PfspAsyncEngine::request_extract() {
+  {
+    // cheapest grant first: an un-started queued frontier moves host-to-host
+    std::lock_guard<std::mutex> l(mu_);
+    if (!q_.empty()) {
+      std::vector<PFSPNode> nodes = std::move(q_.back().first);
+      q_.pop_back();
+      queued_nodes_.fetch_sub(nodes.size(), std::memory_order_relaxed);
+      std::lock_guard<std::mutex> le(ex_.mu);
+      ex_.taken = std::move(nodes);
+      ex_.ready.store(true, std::memory_order_release);
+      return;
+    }
+  }
+  if (done()) {  // nothing to give: answer immediately so no rank waits
+    ex_.ready.store(true, std::memory_order_release);
+    return;
+  }
+  ex_.ready.store(false, std::memory_order_relaxed);
+  ex_.want.store(true, std::memory_order_release);
+  // re-check: the engine may have gone idle between the checks
+  if (done() && ex_.want.exchange(false)) ex_.ready.store(true, std::memory_order_release);
+}
+
+bool PfspAsyncEngine::extract_ready() const {
+  return ex_.ready.load(std::memory_order_acquire);
+}
+
+bool PfspAsyncEngine::extract_pending() const {
+  return ex_.want.load(std::memory_order_acquire);
+}
+
+std::vector<PFSPNode> PfspAsyncEngine::take_extract() {
+  std::vector<PFSPNode> out;
+  if (!ex_.ready.exchange(false)) return out;
+  std::lock_guard<std::mutex> l(ex_.mu);
+  out.swap(ex_.taken);
+  return out;
+}
 
 Result PfspAsyncEngine::join() {
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    finish_ = true;
+    cv_.notify_all();
+  }
   if (th_.joinable()) th_.join();
   if (err_) std::rethrow_exception(err_);
   return result_;

@@ -1,6 +1,9 @@
 // Public API of the single-GPU engines (implemented in engine_gpu.cpp).
 #pragma once
 #include <atomic>
+#include <condition_variable>
+#include <deque>
+#include <mutex>
 #include <string>
 #include <thread>
 #include <vector>
@@ -17,10 +20,24 @@ int gpu_device_count();
 Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int device,
                        const std::string& mode, uint64_t tree0, uint64_t sol0,
                        double phase1_time, unsigned long long capacity);
+// Extraction channel for engine-pausing inter-rank steals
+// (nqueens_dist_multigpu_chpl.chpl:332-377's remote half-pool steal, as an
+// explicit host protocol): the dist tier posts a request; the next readback
+// of a running slice thread holding >= 2m nodes carves the BACK half of its
+// device pool into `taken` and flags `ready`. `live_size` is the latest
+// readback's pool size (victim-selection heuristic, approximate).
+struct ExtractShare {
+  std::atomic<bool> want{false};
+  std::atomic<bool> ready{false};
+  std::atomic<unsigned long long> live_size{0};
+  std::mutex mu;  // guards taken
+  std::vector<PFSPNode> taken;
+};
+
 Result pfsp_gpu_run(const PfspInstance& I, LbKind lb, Pool<PFSPNode>& pool, int m, int M,
                     int device, const std::string& mode, uint64_t tree0, uint64_t sol0,
                     int best0, double phase1_time, unsigned long long capacity,
-                    std::atomic<int>* shared_best);
+                    std::atomic<int>* shared_best, ExtractShare* extract = nullptr);
 
 // Multi-device devpool core: S slice threads per worker (device entry) pull
 // frontier slices off ONE shared queue — an oversubscribed dynamic partition
@@ -42,7 +59,8 @@ DevpoolMultiOut pfsp_devpool_multi(const PfspInstance& I, Pool<PFSPNode>& pool, 
                                    int best0, int m, int M,
                                    const std::vector<int>& devices,
                                    unsigned long long capacity,
-                                   std::atomic<int>* shared_best, Result& diag);
+                                   std::atomic<int>* shared_best, Result& diag,
+                                   ExtractShare* extract = nullptr);
 
 Result nqueens_gpu(int N, int g, int m, int M, int device, const std::string& mode,
                    unsigned long long capacity);
@@ -65,22 +83,49 @@ Result pfsp_multigpu(int inst, const std::string& lb, int ub, int m, int M, int 
                      const std::string& eval, bool share_best, double perc = 0.5,
                      unsigned long long capacity = 1ull << 27);
 
-// Background-thread PFSP engine with a shared incumbent for mid-search
-// RCCL UB exchange (see engine_gpu.cpp).
+// Persistent per-rank PFSP engine: a background thread that accepts
+// successive frontiers (submit), runs the devpool search on each, exposes a
+// shared incumbent for mid-search RCCL UB exchange, and supports
+// engine-pausing work extraction so a starving rank can steal half of a
+// RUNNING rank's device pool (the reference dist tier's remote steal,
+// nqueens_dist_multigpu_chpl.chpl:332-377). Device buffers come from the
+// process-level cache and the thread parks between submits, so repeated
+// frontier claims pay no re-arm cost.
 class PfspAsyncEngine {
  public:
+  PfspAsyncEngine(int inst, const std::string& lb, int ub, int m, int M, int device,
+                  unsigned long long capacity);
+  // one-shot convenience (round-1 API): submits `nodes` immediately
   PfspAsyncEngine(std::vector<PFSPNode> nodes, int inst, const std::string& lb, int ub,
                   int best0, int m, int M, int device, unsigned long long capacity);
   ~PfspAsyncEngine();
+  void submit(std::vector<PFSPNode> nodes, int best0);
   int best() const;
   void update_best(int b);
-  bool done() const;
-  Result join();
+  bool done() const;                     // parked with nothing queued
+  unsigned long long pool_size() const;  // approx live device pool + queued
+  // steal protocol: request -> (engine carves at a readback, or answers
+  // empty when idle) -> take. `extract_pending` = request not yet answered.
+  void request_extract();
+  bool extract_ready() const;
+  bool extract_pending() const;
+  std::vector<PFSPNode> take_extract();
+  Result join();  // stop accepting work, drain, join, return accumulated
 
  private:
+  void loop();
+  int inst_, ub_, m_, M_, device_;
+  std::string lb_;
+  unsigned long long capacity_;
   std::thread th_;
+  mutable std::mutex mu_;
+  std::condition_variable cv_;
+  std::deque<std::pair<std::vector<PFSPNode>, int>> q_;
+  bool finish_ = false;
+  std::atomic<bool> idle_{true};
+  std::atomic<unsigned long long> queued_nodes_{0};
   std::atomic<int> shared_best_;
-  std::atomic<bool> done_;
+  ExtractShare ex_;
   Result result_;
   std::exception_ptr err_;
 };

@@ -34,6 +34,12 @@ def main():
     u = gdist.run_pfsp_shared_ub(14, "lb1", 0)
     assert u["optimum"] == 1377, u["optimum"]
 
+    # live-steal protocol over real RCCL (ub=1 count frozen in
+    # profiles/taillard_sweep_lb2.txt)
+    v = gdist.run_pfsp_live(6, "lb2", 1, engine="gpu")
+    assert v["tree"] == 116837138, v["tree"]
+    assert v["optimum"] == 1195, v["optimum"]
+
     td.barrier()
     if rank == 0:
         print("DIST_GPU_OK")

@@ -41,6 +41,24 @@ def test_dist_gloo(world):
     assert "DIST_CHECK_OK" in r.stdout
 
 
+@pytest.mark.parametrize("world", [2, 3])
+def test_live_steal_gloo(world):
+    # engine-pausing inter-rank stealing with a skewed partition: counts
+    # exact, and at least one steal actually happened
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", f"--nproc-per-node={world}",
+        "--standalone", "--local-addr", "127.0.0.1",
+        os.path.join(HERE, "helpers", "dist_live_check.py"),
+    ]
+    r = run_torchrun(cmd, env)
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "LIVE_STEAL_OK" in r.stdout
+
+
 def test_cli_dist_tier_cpu():
     env = dict(os.environ)
     env.pop("RANK", None)

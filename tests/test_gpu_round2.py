@@ -104,6 +104,64 @@ def test_multigpu_capacity_flag(gpu):
     assert r["sol"] == seq["sol"]
 
 
+def test_async_engine_live_extract(gpu):
+    # engine-pausing steal, single process: carve half the device pool out of
+    # a RUNNING engine, finish both halves separately, total must equal the
+    # frozen ub=1 count (profiles/taillard_sweep_lb2.txt: ta006 lb2)
+    nodes, tree1, sol1, best = gpu.pfsp_gpu_frontier(6, "lb2", 1, 2048, 0)
+    eng = gpu.PfspAsyncEngine(6, "lb2", 1, 25, 50000, 0, 1 << 24)
+    eng.submit(nodes, best)
+    time.sleep(0.05)
+    eng.request_extract()
+    t0 = time.time()
+    while not eng.extract_ready() and time.time() - t0 < 20:
+        time.sleep(0.002)
+    assert eng.extract_ready()
+    stolen = eng.take_extract()
+    r1 = eng.join()
+    assert len(stolen) > 0 and len(stolen) % 24 == 0  # carve happened mid-run
+    r2 = gpu.pfsp_gpu_from_pool(stolen, 6, "lb2", 1, r1["optimum"], 25, 50000, 0,
+                                "devpool", 1 << 24)
+    assert tree1 + r1["tree"] + r2["tree"] == 116837138
+    assert min(r1["optimum"], r2["optimum"]) == 1195
+
+
+def test_async_engine_multiple_submits(gpu):
+    # persistent engine: successive frontier claims accumulate exactly
+    nodes, tree1, sol1, best = gpu.pfsp_gpu_frontier(14, "lb1_d", 1, 2048, 0)
+    half = (len(nodes) // 24 // 2) * 24
+    eng = gpu.PfspAsyncEngine(14, "lb1_d", 1, 25, 50000, 0, 1 << 24)
+    eng.submit(nodes[:half], best)
+    eng.submit(nodes[half:], best)
+    r = eng.join()
+    seq = gpu.pfsp_seq(14, "lb1_d", 1)
+    assert tree1 + r["tree"] == seq["tree"]
+    assert sol1 + r["sol"] == seq["sol"]
+    assert r["optimum"] == 1377
+
+
+def test_dist_live_steal_one_gpu(gpu):
+    # the full cross-rank live-steal protocol with gloo collectives and GPU
+    # engines (both ranks drive device 0 — works on a 1-GPU box)
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    env["GATS_DIST_BACKEND"] = "gloo"
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node=2",
+        "--standalone", "--local-addr", "127.0.0.1",
+        os.path.join(HERE, "helpers", "dist_live_gpu_check.py"),
+    ]
+    for attempt in range(2):
+        r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env,
+                           cwd=ROOT)
+        if r.returncode == 0:
+            break
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "LIVE_GPU_OK" in r.stdout
+
+
 # ---------------- multi-device (driver may land on an 8-GPU node) ----------
 
 
