@@ -425,6 +425,9 @@ def run_pfsp_live(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
         eng = CpuPfspEngine(c, inst, lb, ub, best)
     eng.submit(my, best)
     steals = 0
+    sent_nodes = 0
+    recv_nodes = 0
+    no_steal = os.environ.get("GATS_NO_STEAL") == "1"
     if world > 1:
         dev = _backend_device(f"cuda:{local}" if engine == "gpu" else "cpu")
         STEAL_MIN = max(4 * m, 256)  # don't move trivial pools
@@ -454,6 +457,7 @@ def run_pfsp_live(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
                     continue
                 if rank == d:
                     payload = eng.take_extract()
+                    sent_nodes += len(payload) // NODE_BYTES
                     n = torch.tensor([len(payload)], dtype=torch.int64, device=dev)
                     td.send(n, dst=t)
                     if len(payload):
@@ -466,23 +470,29 @@ def run_pfsp_live(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
                         buf = torch.zeros(int(n.item()), dtype=torch.uint8, device=dev)
                         td.recv(buf, src=d)
                         eng.submit(bytes(buf.cpu().numpy().tobytes()), eng.best())
+                        recv_nodes += int(n.item()) // NODE_BYTES
                         steals += 1
             # new steal requests: hungriest idle ranks target the fattest
             # running pools (one request per victim outstanding)
-            hungry = [i for i, s in enumerate(snap)
-                      if s[1] == 1 and i not in paired_thieves]
-            victims = sorted((i for i, s in enumerate(snap)
-                              if s[1] == 0 and s[2] >= STEAL_MIN
-                              and not s[3] and not s[4]),
-                             key=lambda i: -snap[i][2])
-            for v, h in zip(victims, hungry):
-                if rank == v:
-                    eng.request_extract()
+            if not no_steal:
+                hungry = [i for i, s in enumerate(snap)
+                          if s[1] == 1 and i not in paired_thieves]
+                victims = sorted((i for i, s in enumerate(snap)
+                                  if s[1] == 0 and s[2] >= STEAL_MIN
+                                  and not s[3] and not s[4]),
+                                 key=lambda i: -snap[i][2])
+                for v, h in zip(victims, hungry):
+                    if rank == v:
+                        eng.request_extract()
             time.sleep(poll_s)
     r = eng.join()
     elapsed = time.perf_counter() - t0
     r = dict(r)
     r["time"] = elapsed
+    if os.environ.get("GATS_LIVE_DEBUG") == "1":
+        print(f"[live rank {rank}] tree={r['tree']} sol={r['sol']} steals={steals} "
+              f"sent={sent_nodes} recv={recv_nodes} my={len(my) // NODE_BYTES} "
+              f"tree1={tree1}", flush=True)
     if world == 1:
         r["tree"] += tree1
         r["sol"] += sol1

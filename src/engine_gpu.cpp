@@ -467,6 +467,11 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, const DevLoopCfg& c
     r.kernel_launch += static_cast<uint64_t>(cfg.kernels_per_iter) * b;
     r.d2h++;
     r.d2h_bytes += sizeof(DevCtl);
+    static const bool chk = std::getenv("GATS_DEVPOOL_CHECK") != nullptr;
+    if (chk && cfg.growth > 0 &&
+        ctl_h.p->size > last_size + static_cast<unsigned long long>(b) * cfg.growth)
+      fprintf(stderr, "DEVPOOL VIOLATION: size %llu -> %llu after %d iters (growth %llu)\n",
+              last_size, ctl_h.p->size, b, cfg.growth);
     last_size = ctl_h.p->size;
     if (ctl_h.p->overflow) {
       overflow = true;
@@ -560,6 +565,8 @@ struct SliceShare {
 template <class NodeT>
 static void donate_if_wanted(SliceShare* share, DevCtl* host_ctl, DevCtl* ctl_d,
                              NodeT* pool_d, unsigned long long m, hipStream_t s) {
+  static const bool off = std::getenv("GATS_NO_DONATE") != nullptr;  // bisect knob
+  if (off) return;
   if (!share || host_ctl->overflow) return;
   // donation floor: a half-pool must be a worthwhile work grant (the thief
   // pays ~1 ms of engine start-up) — 2m alone (the reference's steal
@@ -821,12 +828,15 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
     donate_if_wanted(share, hc, live, pool_d.p, m, stream.s);
     if (extract) {
       extract->live_size.store(hc->size, std::memory_order_relaxed);
-      if (extract->want.load(std::memory_order_relaxed) &&
+      int want = ExtractShare::WANTED;
+      if (extract->state.load(std::memory_order_relaxed) == want &&
           hc->size >= 2 * static_cast<unsigned long long>(m) &&
-          extract->want.exchange(false)) {
+          extract->state.compare_exchange_strong(want, ExtractShare::CARVING,
+                                                 std::memory_order_acq_rel)) {
         // engine-pausing inter-rank steal: carve the back half to the host
         // at this (stream-idle) readback boundary; nodes MOVE, never copy,
-        // so counts stay exact
+        // so counts stay exact. The CARVING state keeps the request visibly
+        // pending until the nodes are READY to take.
         const unsigned long long half = hc->size / 2;
         const unsigned long long newsize = hc->size - half;
         {
@@ -842,7 +852,10 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
         hc->size = newsize;
         r.d2h++;
         r.d2h_bytes += half * sizeof(PFSPNode);
-        extract->ready.store(true, std::memory_order_release);
+        if (std::getenv("GATS_CARVE_LOG"))
+          fprintf(stderr, "CARVE pool=%p size %llu -> %llu tree=%llu iters=%llu\n",
+                  (void*)pool_d.p, newsize + half, newsize, hc->tree, hc->iters);
+        extract->state.store(ExtractShare::READY, std::memory_order_release);
       }
     }
   };
@@ -1483,9 +1496,9 @@ void PfspAsyncEngine::loop() {
         nodes = std::move(q_.front().first);
         b0 = q_.front().second;
         q_.pop_front();
-        idle_.store(false, std::memory_order_release);
+        running_ = true;
+        queued_nodes_.fetch_sub(nodes.size(), std::memory_order_relaxed);
       }
-      queued_nodes_.fetch_sub(nodes.size(), std::memory_order_relaxed);
       Pool<PFSPNode> pool;
       pool.pushBackBulk(nodes.data(), nodes.size());
       const int start = (b0 > 0) ? b0 : I.init_ub;
@@ -1506,27 +1519,39 @@ void PfspAsyncEngine::loop() {
       result_.gpu_iters += r.gpu_iters;
       {
         std::lock_guard<std::mutex> l(mu_);
+        running_ = false;
         if (q_.empty()) {
-          idle_.store(true, std::memory_order_release);
           ex_.live_size.store(0, std::memory_order_relaxed);
-          // an unanswered steal request meets an empty engine: answer "none"
-          if (ex_.want.exchange(false)) ex_.ready.store(true, std::memory_order_release);
+          answer_want_empty();  // unanswered steal request meets an empty engine
         }
       }
     }
-    // finishing with a pending request: answer empty so no rank blocks
-    if (ex_.want.exchange(false)) ex_.ready.store(true, std::memory_order_release);
+    answer_want_empty();  // finishing with a pending request: no rank may block
   } catch (...) {
     err_ = std::current_exception();
-    idle_.store(true, std::memory_order_release);
-    if (ex_.want.exchange(false)) ex_.ready.store(true, std::memory_order_release);
+    {
+      // drain so done() turns true and the error surfaces at join() instead
+      // of wedging pollers that wait for done()
+      std::lock_guard<std::mutex> l(mu_);
+      for (auto& it : q_) queued_nodes_.fetch_sub(it.first.size(), std::memory_order_relaxed);
+      q_.clear();
+      running_ = false;
+    }
+    answer_want_empty();
   }
 }
 
+// WANTED -> READY with an empty grant (the "nothing to give" answer). A
+// CARVING state is left alone: the carver will set READY itself.
+void PfspAsyncEngine::answer_want_empty() {
+  int want = ExtractShare::WANTED;
+  ex_.state.compare_exchange_strong(want, ExtractShare::READY,
+                                    std::memory_order_acq_rel);
+}
+
 void PfspAsyncEngine::submit(std::vector<PFSPNode> nodes, int best0) {
-  queued_nodes_.fetch_add(nodes.size(), std::memory_order_relaxed);
   std::lock_guard<std::mutex> l(mu_);
-  idle_.store(false, std::memory_order_release);
+  queued_nodes_.fetch_add(nodes.size(), std::memory_order_relaxed);
   q_.emplace_back(std::move(nodes), best0);
   cv_.notify_one();
 }
@@ -1550,53 +1575,59 @@ void PfspAsyncEngine::update_best(int b) {
 }
 
 bool PfspAsyncEngine::done() const {
-  return idle_.load(std::memory_order_acquire) &&
-         queued_nodes_.load(std::memory_order_relaxed) == 0;
+  std::lock_guard<std::mutex> l(mu_);
+  return !running_ && q_.empty();
 }
 
 unsigned long long PfspAsyncEngine::pool_size() const {
+  std::lock_guard<std::mutex> l(mu_);
   const unsigned long long live =
-      done() ? 0 : ex_.live_size.load(std::memory_order_relaxed);
+      running_ ? ex_.live_size.load(std::memory_order_relaxed) : 0;
   return live + queued_nodes_.load(std::memory_order_relaxed);
 }
 
 void PfspAsyncEngine::request_extract() {
   {
-    // cheapest grant first: an un-started queued frontier moves host-to-host
     std::lock_guard<std::mutex> l(mu_);
+    if (ex_.state.load(std::memory_order_acquire) != ExtractShare::IDLE)
+      return;  // one outstanding request at a time
+    // cheapest grant first: an un-started queued frontier moves host-to-host
     if (!q_.empty()) {
       std::vector<PFSPNode> nodes = std::move(q_.back().first);
       q_.pop_back();
       queued_nodes_.fetch_sub(nodes.size(), std::memory_order_relaxed);
       std::lock_guard<std::mutex> le(ex_.mu);
       ex_.taken = std::move(nodes);
-      ex_.ready.store(true, std::memory_order_release);
+      ex_.state.store(ExtractShare::READY, std::memory_order_release);
       return;
     }
+    if (!running_) {  // nothing to give: answer immediately so no rank waits
+      ex_.state.store(ExtractShare::READY, std::memory_order_release);
+      return;
+    }
+    ex_.state.store(ExtractShare::WANTED, std::memory_order_release);
   }
-  if (done()) {  // nothing to give: answer immediately so no rank waits
-    ex_.ready.store(true, std::memory_order_release);
-    return;
-  }
-  ex_.ready.store(false, std::memory_order_relaxed);
-  ex_.want.store(true, std::memory_order_release);
-  // re-check: the engine may have gone idle between the checks
-  if (done() && ex_.want.exchange(false)) ex_.ready.store(true, std::memory_order_release);
+  // re-check: the run may have completed between the checks
+  if (done()) answer_want_empty();
 }
 
 bool PfspAsyncEngine::extract_ready() const {
-  return ex_.ready.load(std::memory_order_acquire);
+  return ex_.state.load(std::memory_order_acquire) == ExtractShare::READY;
 }
 
 bool PfspAsyncEngine::extract_pending() const {
-  return ex_.want.load(std::memory_order_acquire);
+  const int s = ex_.state.load(std::memory_order_acquire);
+  return s == ExtractShare::WANTED || s == ExtractShare::CARVING;
 }
 
 std::vector<PFSPNode> PfspAsyncEngine::take_extract() {
   std::vector<PFSPNode> out;
-  if (!ex_.ready.exchange(false)) return out;
-  std::lock_guard<std::mutex> l(ex_.mu);
-  out.swap(ex_.taken);
+  if (ex_.state.load(std::memory_order_acquire) != ExtractShare::READY) return out;
+  {
+    std::lock_guard<std::mutex> l(ex_.mu);
+    out.swap(ex_.taken);
+  }
+  ex_.state.store(ExtractShare::IDLE, std::memory_order_release);
   return out;
 }
 

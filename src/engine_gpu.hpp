@@ -24,11 +24,17 @@ Result nqueens_gpu_run(Pool<NQNode>& pool, int N, int g, int m, int M, int devic
 // (nqueens_dist_multigpu_chpl.chpl:332-377's remote half-pool steal, as an
 // explicit host protocol): the dist tier posts a request; the next readback
 // of a running slice thread holding >= 2m nodes carves the BACK half of its
-// device pool into `taken` and flags `ready`. `live_size` is the latest
-// readback's pool size (victim-selection heuristic, approximate).
+// device pool into `taken` and flags READY. The request LIFECYCLE is one
+// atomic — IDLE -> WANTED -> CARVING -> READY -> IDLE — because a carve in
+// flight must still read as "pending" to the requester: treating
+// consumed-want as idle let a second request trigger a second carve that
+// overwrote the first one's nodes before they were taken (nodes already off
+// the donor's pool => silently lost; found via count deficit + carve log).
+// `live_size` is the latest readback's pool size (victim-selection
+// heuristic, approximate).
 struct ExtractShare {
-  std::atomic<bool> want{false};
-  std::atomic<bool> ready{false};
+  static constexpr int IDLE = 0, WANTED = 1, CARVING = 2, READY = 3;
+  std::atomic<int> state{IDLE};
   std::atomic<unsigned long long> live_size{0};
   std::mutex mu;  // guards taken
   std::vector<PFSPNode> taken;
@@ -114,6 +120,7 @@ class PfspAsyncEngine {
 
  private:
   void loop();
+  void answer_want_empty();
   int inst_, ub_, m_, M_, device_;
   std::string lb_;
   unsigned long long capacity_;
@@ -122,7 +129,7 @@ class PfspAsyncEngine {
   std::condition_variable cv_;
   std::deque<std::pair<std::vector<PFSPNode>, int>> q_;
   bool finish_ = false;
-  std::atomic<bool> idle_{true};
+  bool running_ = false;  // a run is executing (guarded by mu_)
   std::atomic<unsigned long long> queued_nodes_{0};
   std::atomic<int> shared_best_;
   ExtractShare ex_;

@@ -143,6 +143,17 @@ def test_async_engine_multiple_submits(gpu):
     assert r["optimum"] == 1377
 
 
+def test_repeated_extraction_exact(gpu):
+    # regression for the carve-lifecycle race: rapid repeated extraction out
+    # of a running engine (IDLE->WANTED->CARVING->READY) must never lose a
+    # carved chunk; totals pinned to the frozen ta006 lb2 count
+    r = subprocess.run([sys.executable, os.path.join(HERE, "helpers",
+                                                     "extract_loop_check.py")],
+                       capture_output=True, text=True, timeout=300, cwd=ROOT)
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "EXTRACT_LOOP_OK" in r.stdout
+
+
 def test_dist_live_steal_one_gpu(gpu):
     # the full cross-rank live-steal protocol with gloo collectives and GPU
     # engines (both ranks drive device 0 — works on a 1-GPU box)
