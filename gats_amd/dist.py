@@ -281,9 +281,14 @@ def run_pfsp_shared_ub(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
 class CpuPfspEngine:
     """CPU twin of the C++ PfspAsyncEngine (same protocol surface) so the
     live-steal protocol is CI-testable on gloo without a GPU: a daemon thread
-    drains the queued frontier one node at a time (so extraction can always
-    interrupt between subtrees); extraction hands over half the remaining
-    queue. Counts are exact because nodes move, never copy."""
+    advances a host pool in bounded steps (pfsp_seq_step, ~tens of ms each),
+    so extraction interrupts quickly AND the shared incumbent is re-read
+    every step (a whole-subtree run with a stale ub=0 incumbent could take
+    minutes). Extraction hands over the FRONT half of the remaining pool
+    (shallow nodes, like the reference's steal-from-front); counts are exact
+    because nodes move, never copy."""
+
+    STEP_NODES = 50000
 
     def __init__(self, core, inst, lb, ub, best0):
         import threading
@@ -291,12 +296,13 @@ class CpuPfspEngine:
         self.c, self.inst, self.lb, self.ub = core, inst, lb, ub
         self._best = best0 if best0 > 0 else (1 << 30)
         self._lock = threading.Lock()
-        self._q = []
+        self._pool = b""
         self._tree = 0
         self._sol = 0
         self._busy = False
         self._stop = False
         self._want = False
+        self._inflight = 0  # nodes inside the current bounded step
         self._ready = None  # bytes once an extract request is answered
         self._th = threading.Thread(target=self._loop, daemon=True)
         self._th.start()
@@ -306,41 +312,38 @@ class CpuPfspEngine:
 
         nb = NODE_BYTES
         while True:
-            node = None
             with self._lock:
                 if self._want:
-                    half = len(self._q) // 2
-                    if half:
-                        self._ready = b"".join(self._q[-half:])
-                        del self._q[-half:]
-                    else:
-                        self._ready = b""
+                    half = (len(self._pool) // nb // 2) * nb
+                    self._ready = self._pool[:half]
+                    self._pool = self._pool[half:]
                     self._want = False
-                if self._q:
-                    node = self._q.pop()
-                    self._busy = True
-                else:
-                    self._busy = False
-                    if self._stop:
-                        return
-            if node is None:
+                chunk, self._pool = self._pool, b""
+                self._busy = bool(chunk)
+                self._inflight = len(chunk) // nb
+                best = self._best
+                if not chunk and self._stop:
+                    return
+            if not chunk:
                 _t.sleep(0.001)
                 continue
-            r = self.c.pfsp_seq_from_pool(node, self.inst, self.lb, self.ub, self._best)
+            tree, sol, best_out, left = self.c.pfsp_seq_step(
+                chunk, self.inst, self.lb, self.ub, best, self.STEP_NODES)
             with self._lock:
-                self._tree += r["tree"]
-                self._sol += r["sol"]
-                if r.get("optimum"):
-                    self._best = min(self._best, r["optimum"])
+                self._tree += tree
+                self._sol += sol
+                if 0 < best_out < self._best:
+                    self._best = best_out
+                self._pool = left + self._pool  # submits during the step append
+                self._inflight = 0
+                self._busy = bool(self._pool)
 
     def submit(self, nodes: bytes, best0: int = 0):
-        nb = NODE_BYTES
-        parts = [nodes[i:i + nb] for i in range(0, len(nodes), nb)]
         with self._lock:
-            if best0 > 0:
-                self._best = min(self._best, best0)
-            self._q.extend(parts)
-            if parts:
+            if 0 < best0 < self._best:
+                self._best = best0
+            self._pool += nodes
+            if nodes:
                 self._busy = True
 
     def best(self):
@@ -354,15 +357,17 @@ class CpuPfspEngine:
 
     def done(self):
         with self._lock:
-            return not self._busy and not self._q
+            return not self._busy and not self._pool
 
     def pool_size(self):
         with self._lock:
-            return len(self._q)
+            return len(self._pool) // NODE_BYTES + self._inflight
 
     def request_extract(self):
         with self._lock:
-            if not self._q and not self._busy:
+            if self._want or self._ready is not None:
+                return
+            if not self._pool and not self._busy:
                 self._ready = b""
             else:
                 self._want = True
@@ -381,8 +386,6 @@ class CpuPfspEngine:
             return out or b""
 
     def join(self):
-        import time as _t
-
         with self._lock:
             self._stop = True
         while self._th.is_alive():
@@ -432,7 +435,17 @@ def run_pfsp_live(inst, lb="lb1", ub=1, m=25, M=50000, capacity=1 << 24,
         dev = _backend_device(f"cuda:{local}" if engine == "gpu" else "cpu")
         STEAL_MIN = max(4 * m, 256)  # don't move trivial pools
         gathered = [torch.zeros(5, dtype=torch.int64, device=dev) for _ in range(world)]
+        debug = os.environ.get("GATS_LIVE_DEBUG") == "1"
+        rounds = 0
+        last_dbg = time.perf_counter()
         while True:
+            rounds += 1
+            if debug and time.perf_counter() - last_dbg > 5:
+                last_dbg = time.perf_counter()
+                print(f"[live rank {rank} round {rounds}] done={eng.done()} "
+                      f"pool={eng.pool_size()} pend={eng.extract_pending()} "
+                      f"ready={eng.extract_ready()} steals={steals} best={eng.best()}",
+                      flush=True)
             st = torch.tensor([eng.best(), 1 if eng.done() else 0,
                                int(eng.pool_size()),
                                1 if eng.extract_ready() else 0,

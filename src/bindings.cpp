@@ -197,6 +197,29 @@ py::dict pfsp_seq_from_pool(const py::bytes& nodes, int inst, const std::string&
   return result_to_dict(r);
 }
 
+// Bounded sequential step: explore up to max_nodes tree nodes of the given
+// frontier, then return the remaining pool (checkpointable CPU search; the
+// dist tier's CPU mock engine uses it so a stale incumbent is re-read every
+// few ms instead of once per whole subtree).
+py::tuple pfsp_seq_step(const py::bytes& nodes, int inst, const std::string& lb_str,
+                        int ub, int best0, uint64_t max_nodes) {
+  auto v = nodes_from_bytes<PFSPNode>(nodes);
+  const LbKind lb = lb_from_string(lb_str);
+  uint64_t tree = 0, sol = 0;
+  int best;
+  Pool<PFSPNode> pool;
+  {
+    py::gil_scoped_release rel;
+    PfspInstance I = make_pfsp_instance(inst, ub);
+    best = (best0 > 0) ? best0 : I.init_ub;
+    pool.pushBackBulk(v.data(), v.size());
+    PFSPNode parent;
+    while (tree < max_nodes && pool.popBack(parent))
+      pfsp_decompose(I, lb, parent, tree, sol, best, pool);
+  }
+  return py::make_tuple(tree, sol, best, nodes_to_bytes(pool.data(), pool.size()));
+}
+
 // Exercises pool push/pop/bulk semantics from C++ (unit-test helper).
 py::dict pool_selftest() {
   py::dict d;
@@ -343,6 +366,9 @@ PYBIND11_MODULE(_core, mod) {
           py::arg("device") = 0, py::arg("mode") = "devpool",
           py::arg("capacity") = (1ull << 27));
 
+  mod.def("pfsp_seq_step", &pfsp_seq_step, py::arg("nodes"), py::arg("inst"),
+          py::arg("lb") = "lb1", py::arg("ub") = 1, py::arg("best0") = 0,
+          py::arg("max_nodes") = 50000);
   mod.def("nqueens_seq_from_pool", &nqueens_seq_from_pool, py::arg("nodes"), py::arg("N"),
           py::arg("g") = 1);
   mod.def("pfsp_seq_from_pool", &pfsp_seq_from_pool, py::arg("nodes"), py::arg("inst"),

@@ -150,17 +150,18 @@ struct PfspTablesGuard {
       p1[i] = static_cast<uint8_t>(I.lb2.pairs1[i]);
       p2[i] = static_cast<uint8_t>(I.lb2.pairs2[i]);
     }
-    std::vector<uint64_t> jp(static_cast<size_t>(pairs) * n);
+    // u32 pack (lossless: job < 20, lag <= 20*99 < 2^11, ptm <= 99 < 2^8)
+    std::vector<uint32_t> jp(static_cast<size_t>(pairs) * n);
     for (int k = 0; k < pairs; k++) {
       const int ma0 = I.lb2.pairs1[k], ma1 = I.lb2.pairs2[k];
       for (int j = 0; j < n; j++) {
         const int job = I.lb2.johnson_schedules[static_cast<size_t>(k) * n + j];
-        const uint64_t ptm0 = static_cast<uint64_t>(I.lb1.p_times[ma0 * n + job]);
-        const uint64_t ptm1 = static_cast<uint64_t>(I.lb1.p_times[ma1 * n + job]);
-        const uint64_t lag =
-            static_cast<uint64_t>(I.lb2.lags[static_cast<size_t>(k) * n + job]);
+        const uint32_t ptm0 = static_cast<uint32_t>(I.lb1.p_times[ma0 * n + job]);
+        const uint32_t ptm1 = static_cast<uint32_t>(I.lb1.p_times[ma1 * n + job]);
+        const uint32_t lag =
+            static_cast<uint32_t>(I.lb2.lags[static_cast<size_t>(k) * n + job]);
         jp[static_cast<size_t>(k) * n + j] =
-            (static_cast<uint64_t>(job) << 48) | (lag << 32) | (ptm1 << 16) | ptm0;
+            (static_cast<uint32_t>(job) << 27) | (lag << 16) | (ptm1 << 8) | ptm0;
       }
     }
     tb.p_times = keep(dev_upload(p16.data(), p16.size()), p16.size() * sizeof(p16[0]));

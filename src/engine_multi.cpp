@@ -209,17 +209,18 @@ struct PfspGpuCtx {
       p1[i] = static_cast<uint8_t>(I.lb2.pairs1[i]);
       p2[i] = static_cast<uint8_t>(I.lb2.pairs2[i]);
     }
-    std::vector<uint64_t> jp(static_cast<size_t>(pairs) * jobs);
+    // u32 pack (lossless; see PfspDevTables::johnson_packed)
+    std::vector<uint32_t> jp(static_cast<size_t>(pairs) * jobs);
     for (int k = 0; k < pairs; k++) {
       const int ma0 = I.lb2.pairs1[k], ma1 = I.lb2.pairs2[k];
       for (int j = 0; j < jobs; j++) {
         const int job = I.lb2.johnson_schedules[static_cast<size_t>(k) * jobs + j];
-        const uint64_t ptm0 = static_cast<uint64_t>(I.lb1.p_times[ma0 * jobs + job]);
-        const uint64_t ptm1 = static_cast<uint64_t>(I.lb1.p_times[ma1 * jobs + job]);
-        const uint64_t lag =
-            static_cast<uint64_t>(I.lb2.lags[static_cast<size_t>(k) * jobs + job]);
+        const uint32_t ptm0 = static_cast<uint32_t>(I.lb1.p_times[ma0 * jobs + job]);
+        const uint32_t ptm1 = static_cast<uint32_t>(I.lb1.p_times[ma1 * jobs + job]);
+        const uint32_t lag =
+            static_cast<uint32_t>(I.lb2.lags[static_cast<size_t>(k) * jobs + job]);
         jp[static_cast<size_t>(k) * jobs + j] =
-            (static_cast<uint64_t>(job) << 48) | (lag << 32) | (ptm1 << 16) | ptm0;
+            (static_cast<uint32_t>(job) << 27) | (lag << 16) | (ptm1 << 8) | ptm0;
       }
     }
     tb.p_times = upload(p16);

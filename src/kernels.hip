@@ -15,7 +15,7 @@
 //                     the host only polls a 64 B control block every few
 //                     iterations. No global atomics on the hot path.
 //   * Bound tables staged in LDS (p_times int16; Johnson entries packed into
-//     one u64 per (pair, position): job<<48|lag<<32|ptm1<<16|ptm0): ~31 KB
+//     one u32 per (pair, position): job<<27|lag<<16|ptm1<<8|ptm0): ~16 KB
 //     for 20x20 vs 160 KB per CU.
 //   * No runtime-indexed per-thread arrays (they would spill to scratch on
 //     CDNA4): machine loops are fully unrolled via a machine-count template
@@ -180,7 +180,7 @@ struct LdsLb2 {
   static constexpr int PAIRS = MM * (MM - 1) / 2;
   int16_t p[MM * MAX_JOBS];
   int32_t min_tails[MM];
-  uint64_t jp[PAIRS * MAX_JOBS];  // packed johnson: job<<48|lag<<32|ptm1<<16|ptm0
+  uint32_t jp[PAIRS * MAX_JOBS];  // packed johnson: job<<27|lag<<16|ptm1<<8|ptm0
 };
 
 template <int MM, class LDS>
@@ -315,22 +315,22 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
     const int ma0b = pair_first<MM>(two ? l + 1 : l), ma1b = pair_second<MM>(two ? l + 1 : l);
     int t0a = front[ma0a], t1a = front[ma1a];
     int t0b = front[ma0b], t1b = front[ma1b];
-    const uint64_t* jpa = &lds.jp[l * jobs];
-    const uint64_t* jpb = &lds.jp[(two ? l + 1 : l) * jobs];
+    const uint32_t* jpa = &lds.jp[l * jobs];
+    const uint32_t* jpb = &lds.jp[(two ? l + 1 : l) * jobs];
     for (int j = 0; j < jobs; j++) {
-      const uint64_t va = jpa[j];  // one ds_read_b64 replaces 4 scalar LDS reads
-      const uint64_t vb = jpb[j];
-      const int ja = static_cast<int>(va >> 48);
-      const int jb = static_cast<int>(vb >> 48);
+      const uint32_t va = jpa[j];  // one ds_read_b32 replaces 4 scalar LDS reads
+      const uint32_t vb = jpb[j];
+      const int ja = static_cast<int>(va >> 27);
+      const int jb = static_cast<int>(vb >> 27);
       if (!(scheduled >> ja & 1u)) {
-        t0a += static_cast<int>(va & 0xffff);
-        t1a = max(t1a, t0a + static_cast<int>((va >> 32) & 0xffff));
-        t1a += static_cast<int>((va >> 16) & 0xffff);
+        t0a += static_cast<int>(va & 0xffu);
+        t1a = max(t1a, t0a + static_cast<int>((va >> 16) & 0x7ffu));
+        t1a += static_cast<int>((va >> 8) & 0xffu);
       }
       if (!(scheduled >> jb & 1u)) {
-        t0b += static_cast<int>(vb & 0xffff);
-        t1b = max(t1b, t0b + static_cast<int>((vb >> 32) & 0xffff));
-        t1b += static_cast<int>((vb >> 16) & 0xffff);
+        t0b += static_cast<int>(vb & 0xffu);
+        t1b = max(t1b, t0b + static_cast<int>((vb >> 16) & 0x7ffu));
+        t1b += static_cast<int>((vb >> 8) & 0xffu);
       }
     }
     // merge pair a, check, then pair b: keeps the returned value bit-equal to
@@ -793,7 +793,7 @@ struct LdsLb2w {
   static constexpr int PAIRS = MM * (MM - 1) / 2;
   int16_t p[MM * MAX_JOBS];
   int32_t min_tails[MM];
-  uint64_t jp[PAIRS * MAX_JOBS];
+  uint32_t jp[PAIRS * MAX_JOBS];
   uint8_t pair1[PAIRS], pair2[PAIRS];
   uint16_t fronts[BLOCK][MM + 1];  // child completion times (values <= 20*20*99)
   uint32_t smask[BLOCK];           // scheduled-job bitmask per child
@@ -893,14 +893,14 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
           const int ma1 = lds.pair2[pr];
           int t0 = fr[ma0];
           int t1 = fr[ma1];
-          const uint64_t* jp = &lds.jp[pr * jobs];
+          const uint32_t* jp = &lds.jp[pr * jobs];
           for (int j = 0; j < jobs; j++) {
-            const uint64_t v = jp[j];
-            const int job = static_cast<int>(v >> 48);
+            const uint32_t v = jp[j];
+            const int job = static_cast<int>(v >> 27);
             if (!(sched >> job & 1u)) {
-              t0 += static_cast<int>(v & 0xffff);
-              t1 = max(t1, t0 + static_cast<int>((v >> 32) & 0xffff));
-              t1 += static_cast<int>((v >> 16) & 0xffff);
+              t0 += static_cast<int>(v & 0xffu);
+              t1 = max(t1, t0 + static_cast<int>((v >> 16) & 0x7ffu));
+              t1 += static_cast<int>((v >> 8) & 0xffu);
             }
           }
           mylb = max(mylb, max(t1 + lds.min_tails[ma1], t0 + lds.min_tails[ma0]));
