@@ -195,6 +195,12 @@ def run_pfsp(inst, lb="lb1", ub=1, m=25, M=50000, mode="devpool", capacity=1 << 
              frontier_target=None, engine="gpu"):
     c = gats_amd.core()
     rank, world = init_dist()
+    if (world == 1 and engine == "gpu" and mode == "devpool"
+            and os.environ.get("GATS_NO_ROOTED") != "1"):
+        # single rank: run the whole search device-rooted (no frontier build,
+        # no host marshaling) — the small-search fixed cost drops to the
+        # engine spin-up alone
+        return c.pfsp_gpu_rooted(inst, lb, ub, M, 0, capacity)
     if frontier_target is None:
         # PFSP 20-job trees are small (ta014 lb1 ~2.6M nodes); a deep frontier
         # would move a large share of the search onto the single-threaded CPU

@@ -334,6 +334,20 @@ PYBIND11_MODULE(_core, mod) {
           py::arg("M") = 50000, py::arg("device") = 0, py::arg("mode") = "devpool",
           py::arg("capacity") = (1ull << 27));
 
+  mod.def("pfsp_gpu_rooted",
+          [](int inst, const std::string& lb, int ub, int M, int device,
+             unsigned long long capacity) {
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = pfsp_gpu_rooted(inst, lb, ub, M, device, capacity);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("inst") = 14, py::arg("lb") = "lb1", py::arg("ub") = 1,
+          py::arg("M") = 50000, py::arg("device") = 0,
+          py::arg("capacity") = (1ull << 27));
+
   mod.def("nqueens_gpu_from_pool",
           [](const py::bytes& nodes, int N, int g, int m, int M, int device,
              const std::string& mode, unsigned long long capacity) {

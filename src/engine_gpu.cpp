@@ -1288,6 +1288,38 @@ Result pfsp_gpu(int inst, const std::string& lb_str, int ub, int m, int M, int d
                       nullptr);
 }
 
+Result pfsp_gpu_rooted(int inst, const std::string& lb_str, int ub, int M, int device,
+                       unsigned long long capacity) {
+  const LbKind lb = lb_from_string(lb_str);
+  const PfspInstance& I = pfsp_instance_cached(inst, ub);
+  Result r;
+  Pool<PFSPNode> pool;
+  pool.pushBack(pfsp_root());
+  const int lbk = lbk_of(lb);
+  HIP_CHECK(hipSetDevice(device));
+  const double t0 = now_sec();
+  r.phases.push_back({0, 0, 0.0});  // no CPU phase 1
+  DevpoolMultiOut o =
+      pfsp_devpool_multi(I, pool, lbk, I.init_ub, /*m=*/1, M, {device}, capacity,
+                         nullptr, r);
+  uint64_t tree = o.tree, sol = o.sol;
+  int best = o.best;
+  const double t2 = now_sec();
+  r.phases.push_back({tree, sol, t2 - t0});
+  // leftovers can only exist if a slice aborted; drain defensively
+  uint64_t tree2 = tree, sol2 = sol;
+  PFSPNode parent;
+  while (pool.popBack(parent)) pfsp_decompose(I, lb, parent, tree2, sol2, best, pool);
+  const double t3 = now_sec();
+  r.phases.push_back({tree2 - tree, sol2 - sol, t3 - t2});
+  r.tree = tree2;
+  r.sol = sol2;
+  r.optimum = best;
+  r.gpu_time = t2 - t0;
+  r.time = t3 - t0;
+  return r;
+}
+
 Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst,
                           const std::string& lb_str, int ub, int best0, int m, int M,
                           int device, const std::string& mode, unsigned long long capacity) {

@@ -76,6 +76,16 @@ Result nqueens_gpu_from_pool(const std::vector<NQNode>& nodes, int N, int g, int
 
 Result pfsp_gpu(int inst, const std::string& lb, int ub, int m, int M, int device,
                 const std::string& mode, unsigned long long capacity);
+// Device-rooted search: the WHOLE search runs in the devpool from the root
+// (no CPU phase-1 BFS, no host frontier marshaling, no CPU phase 3) — the
+// devpool's m is 1 so the pool drains on device. One slice thread starts at
+// the root; big searches self-parallelize because idle slice threads take
+// donated half-pools once the pool crosses the donation floor (64k nodes),
+// while small searches (PFSP ta0xx, few ms) never pay slicing overhead.
+// Counts equal the sequential engine's at ub=1 (pruning is
+// decomposition-invariant); the phase split is all-phase-2 by construction.
+Result pfsp_gpu_rooted(int inst, const std::string& lb, int ub, int M, int device,
+                       unsigned long long capacity);
 Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst, const std::string& lb,
                           int ub, int best0, int m, int M, int device,
                           const std::string& mode, unsigned long long capacity);

@@ -143,6 +143,19 @@ def test_async_engine_multiple_submits(gpu):
     assert r["optimum"] == 1377
 
 
+def test_pfsp_rooted_counts(gpu):
+    # device-rooted mode: whole search on the GPU from the root; counts must
+    # equal the sequential engine (small) and the frozen sweep value (ta006)
+    seq = gpu.pfsp_seq(14, "lb2", 1)
+    r = gpu.pfsp_gpu_rooted(14, "lb2", 1, 50000, 0, 1 << 24)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    assert r["optimum"] == 1377
+    r6 = gpu.pfsp_gpu_rooted(6, "lb2", 1, 50000, 0, 1 << 24)
+    assert r6["tree"] == 116837138
+    assert r6["optimum"] == 1195
+
+
 def test_repeated_extraction_exact(gpu):
     # regression for the carve-lifecycle race: rapid repeated extraction out
     # of a running engine (IDLE->WANTED->CARVING->READY) must never lose a
