@@ -251,10 +251,13 @@ static void run_on_pool(int T, MakeTask&& make_task) {
   done_cv.wait(l, [&] { return remaining.load(std::memory_order_acquire) == 0; });
 }
 
-// Stream recycling is OFF by default: reusing streams measured a reproducible
-// 40% N-Queens regression (146 vs 108 ms per N=17 search, same box, ROCm 7.2)
-// — fresh streams apparently schedule better across the slice threads. Keep
-// the cache behind GATS_STREAM_CACHE=1 for experiments.
+// Stream recycling is ON by default (round 2): hipStreamCreate/Destroy costs
+// up to ~3 ms per stream on an otherwise-idle ROCm 7.2 device, which
+// dominated small searches (ta019 lb2: 15 ms -> 1.55 ms with reuse) and cost
+// the N=17 bench ~2 ms/step. Round 1 measured reuse 40% SLOWER, but that
+// regression disappeared with the round-2 loop (parked worker threads +
+// non-blocking streams + async slice copies). GATS_STREAM_CACHE=0 restores
+// fresh streams for experiments.
 struct StreamCache {
   std::mutex mu;
   std::map<int, std::vector<hipStream_t>> free_;
@@ -268,7 +271,8 @@ struct StreamGuard {
   hipStream_t s{};
   int dev = 0;
   static bool caching() {
-    static const bool on = std::getenv("GATS_STREAM_CACHE") != nullptr;
+    static const char* e = std::getenv("GATS_STREAM_CACHE");
+    static const bool on = (e == nullptr) || std::string(e) != "0";
     return on;
   }
   StreamGuard() {
@@ -531,6 +535,25 @@ static int devpool_slices() {
   return 4;
 }
 
+// Internal per-iteration expansion width of the devpool. The reference's M
+// caps how many nodes are COPIED to the GPU per offload round
+// (pfsp_gpu_cuda.c:424-426); a device-resident pool has no copy window, so
+// the devpool widens the chunk to ~512k nodes — one launch then fills the
+// 256 CUs by itself and the iteration count drops ~10x. --M stays a lower
+// bound here (and is honored exactly in hostpool mode). Clamped so one
+// iteration's worst-case children still fit half the pool capacity (the
+// spill path needs that), and so chunk*branching fits u32 child indexing.
+static unsigned long long devpool_chunk_cap(int M, int per, unsigned long long capacity) {
+  unsigned long long c = 1ull << 19;
+  if (const char* e = std::getenv("GATS_DEVPOOL_CHUNK")) c = strtoull(e, nullptr, 10);
+  const unsigned long long fit = capacity / (2 * static_cast<unsigned long long>(per));
+  if (c > fit) c = fit;
+  if (c < static_cast<unsigned long long>(M)) c = M;  // user window is the floor
+  const unsigned long long lim = (1ull << 31) / per;
+  if (c > lim) c = lim;
+  return c;
+}
+
 struct SliceOut {
   DevCtl fin{};
   Result diag;
@@ -675,9 +698,10 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
   StreamGuard stream;
   SliceOut out;
   Result& r = out.diag;
+  const unsigned long long Mc = devpool_chunk_cap(M, N, capacity);
   DevGuard<NQNode> pool_d(capacity);
   DevGuard<DevCtl> ctl_d(2);  // parity-alternating control blocks
-  const int G = devpool_grid(M, N, 1);
+  const int G = devpool_grid(Mc, N, 1);
   const int stride = devpool_stride(1);
   DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G);
@@ -687,10 +711,10 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
-    launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, m, M,
+    launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, m, Mc,
                 stream.s);
     launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, childbuf_d.p, pool_d.p, stride, G,
-                      m, M, capacity, stream.s);
+                      m, Mc, capacity, stream.s);
   };
   ReadbackHook hook = [&](DevCtl* hc, DevCtl* live) {
     donate_if_wanted(share, hc, live, pool_d.p, m, stream.s);
@@ -714,7 +738,7 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
   DevLoopCfg cfg;
   cfg.m = m;
   cfg.capacity = capacity;
-  cfg.growth = static_cast<unsigned long long>(M) * N;  // worst-case children/iter
+  cfg.growth = Mc * N;  // worst-case children/iter
   cfg.allow_graph = allow_graph;
 
   auto run_pool = [&](unsigned long long init_size) {
@@ -802,9 +826,10 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   out.fin.best = best0;
   Result& r = out.diag;
   const int jobs = I.jobs, machines = I.machines;
+  const unsigned long long Mc = devpool_chunk_cap(M, jobs, capacity);
   DevGuard<PFSPNode> pool_d(capacity);
   DevGuard<DevCtl> ctl_d(2);
-  const int G = devpool_grid(M, jobs, lbk);
+  const int G = devpool_grid(Mc, jobs, lbk);
   const int stride = devpool_stride(lbk);
   DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G);
@@ -820,10 +845,10 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
     launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk,
-                  tb, m, M, stream.s);
+                  tb, m, Mc, stream.s);
     if (presum) launch_presum(bc_d.p, gsum_d.p, G, stream.s);
     launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, presum ? gsum_d.p : nullptr,
-                        childbuf_d.p, pool_d.p, stride, G, m, M, capacity, stream.s);
+                        childbuf_d.p, pool_d.p, stride, G, m, Mc, capacity, stream.s);
   };
   ReadbackHook hook = [&](DevCtl* hc, DevCtl* live) {
     donate_if_wanted(share, hc, live, pool_d.p, m, stream.s);
@@ -879,7 +904,7 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   DevLoopCfg cfg;
   cfg.m = m;
   cfg.capacity = capacity;
-  cfg.growth = static_cast<unsigned long long>(M) * jobs;
+  cfg.growth = Mc * jobs;
   cfg.allow_graph = allow_graph;
 
   auto run_pool = [&](unsigned long long init_size, int init_best) {
