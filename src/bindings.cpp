@@ -334,6 +334,18 @@ PYBIND11_MODULE(_core, mod) {
           py::arg("M") = 50000, py::arg("device") = 0, py::arg("mode") = "devpool",
           py::arg("capacity") = (1ull << 27));
 
+  mod.def("nqueens_gpu_rooted",
+          [](int N, int g, int M, int device, unsigned long long capacity) {
+            Result r;
+            {
+              py::gil_scoped_release rel;
+              r = nqueens_gpu_rooted(N, g, M, device, capacity);
+            }
+            return result_to_dict(r);
+          },
+          py::arg("N") = 14, py::arg("g") = 1, py::arg("M") = 50000,
+          py::arg("device") = 0, py::arg("capacity") = (1ull << 27));
+
   mod.def("pfsp_gpu_rooted",
           [](int inst, const std::string& lb, int ub, int M, int device,
              unsigned long long capacity) {

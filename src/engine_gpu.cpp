@@ -1345,6 +1345,28 @@ Result pfsp_gpu_rooted(int inst, const std::string& lb_str, int ub, int M, int d
   return r;
 }
 
+Result nqueens_gpu_rooted(int N, int g, int M, int device, unsigned long long capacity) {
+  Result r;
+  Pool<NQNode> pool;
+  pool.pushBack(nq_root());
+  HIP_CHECK(hipSetDevice(device));
+  const double t0 = now_sec();
+  r.phases.push_back({0, 0, 0.0});  // no CPU phase 1
+  DevpoolMultiOut o = nq_devpool_multi(pool, N, g, /*m=*/1, M, {device}, capacity, r);
+  const double t2 = now_sec();
+  r.phases.push_back({o.tree, o.sol, t2 - t0});
+  uint64_t tree2 = o.tree, sol2 = o.sol;
+  NQNode parent;
+  while (pool.popBack(parent)) nq_decompose(parent, N, g, tree2, sol2, pool);
+  const double t3 = now_sec();
+  r.phases.push_back({tree2 - o.tree, sol2 - o.sol, t3 - t2});
+  r.tree = tree2;
+  r.sol = sol2;
+  r.gpu_time = t2 - t0;
+  r.time = t3 - t0;
+  return r;
+}
+
 Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst,
                           const std::string& lb_str, int ub, int best0, int m, int M,
                           int device, const std::string& mode, unsigned long long capacity) {

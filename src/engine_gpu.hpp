@@ -86,6 +86,7 @@ Result pfsp_gpu(int inst, const std::string& lb, int ub, int m, int M, int devic
 // decomposition-invariant); the phase split is all-phase-2 by construction.
 Result pfsp_gpu_rooted(int inst, const std::string& lb, int ub, int M, int device,
                        unsigned long long capacity);
+Result nqueens_gpu_rooted(int N, int g, int M, int device, unsigned long long capacity);
 Result pfsp_gpu_from_pool(const std::vector<PFSPNode>& nodes, int inst, const std::string& lb,
                           int ub, int best0, int m, int M, int device,
                           const std::string& mode, unsigned long long capacity);

@@ -143,6 +143,16 @@ def test_async_engine_multiple_submits(gpu):
     assert r["optimum"] == 1377
 
 
+def test_nqueens_rooted_counts(gpu):
+    seq = gpu.nqueens_seq(14, 1)
+    r = gpu.nqueens_gpu_rooted(14, 1, 50000, 0, 1 << 26)
+    assert r["tree"] == seq["tree"]
+    assert r["sol"] == seq["sol"]
+    r15 = gpu.nqueens_gpu_rooted(15, 1, 50000, 0, 1 << 26)
+    assert r15["sol"] == 2279184
+    assert r15["tree"] == 171129071
+
+
 def test_pfsp_rooted_counts(gpu):
     # device-rooted mode: whole search on the GPU from the root; counts must
     # equal the sequential engine (small) and the frozen sweep value (ta006)
