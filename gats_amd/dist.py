@@ -157,10 +157,13 @@ def run_nqueens(N, g=1, m=25, M=50000, mode="devpool", capacity=1 << 27,
                 frontier_target=None, engine="gpu"):
     c = gats_amd.core()
     rank, world = init_dist()
-    if (world == 1 and engine == "gpu" and mode == "devpool"
+    if (world == 1 and engine == "gpu"
             and os.environ.get("GATS_NO_ROOTED") != "1"):
-        # single rank: whole search device-rooted (no frontier marshaling)
-        return c.nqueens_gpu_rooted(N, g, M, 0, capacity)
+        # single rank: skip the dist-tier frontier marshaling entirely and run
+        # the single-GPU engine directly. (The device-rooted variant measured
+        # ~3% slower at N=17: donation-based spreading from one chain
+        # balances worse than the pre-split 16-slice queue on big trees.)
+        return c.nqueens_gpu(N, g, m, M, 0, mode, capacity)
     if frontier_target is None:
         # 65536 regardless of world: the parallel block-BFS builds it in ~8 ms
         # (every rank redundantly, per step), and the engine's deep-frontier
