@@ -532,7 +532,12 @@ static int devpool_slices() {
     int v = atoi(e);
     return v < 1 ? 1 : v;
   }
-  return 4;
+  // 2 concurrent slice chains: with the ~512k-node expansion chunk one
+  // launch already fills the 256 CUs, so extra slices only add overhead —
+  // they exist to hide the kernel-boundary bubbles of one chain (S=1 is
+  // ~15% slower; measured N=17: S=1 89.3 ms, S=2 77.4, S=4 79.6, S=6 84.2;
+  // ta021 lb2 and N=18/19 are S-insensitive at wide chunks)
+  return 2;
 }
 
 // Internal per-iteration expansion width of the devpool. The reference's M
