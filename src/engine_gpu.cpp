@@ -548,8 +548,13 @@ static int devpool_slices() {
 // bound here (and is honored exactly in hostpool mode). Clamped so one
 // iteration's worst-case children still fit half the pool capacity (the
 // spill path needs that), and so chunk*branching fits u32 child indexing.
-static unsigned long long devpool_chunk_cap(int M, int per, unsigned long long capacity) {
-  unsigned long long c = 1ull << 19;
+static unsigned long long devpool_chunk_cap(int M, int per, unsigned long long capacity,
+                                            int lbk) {
+  // lb2's wave-cooperative kernel gives each CHILD a whole wave, so M=50000
+  // already launches ~16k waves (chip is full); widening it just multiplies
+  // the gather grid with empty slots and the worst-case growth bound
+  // (measured: spill storms + 687 us gathers on ta006 at a 419k chunk)
+  unsigned long long c = (lbk == 2) ? static_cast<unsigned long long>(M) : (1ull << 19);
   if (const char* e = std::getenv("GATS_DEVPOOL_CHUNK")) c = strtoull(e, nullptr, 10);
   const unsigned long long fit = capacity / (2 * static_cast<unsigned long long>(per));
   if (c > fit) c = fit;
@@ -703,7 +708,7 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
   StreamGuard stream;
   SliceOut out;
   Result& r = out.diag;
-  const unsigned long long Mc = devpool_chunk_cap(M, N, capacity);
+  const unsigned long long Mc = devpool_chunk_cap(M, N, capacity, 1);
   DevGuard<NQNode> pool_d(capacity);
   DevGuard<DevCtl> ctl_d(2);  // parity-alternating control blocks
   const int G = devpool_grid(Mc, N, 1);
@@ -711,6 +716,10 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
   DevGuard<NQNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G);
   DevGuard<unsigned long long> bs_d(G), be_d(G);
+  // group sums keep gather's prefix walk O(G/256): at the wide chunk G is
+  // ~8700 blocks and the linear walk dominated gather (193 us avg at N=17)
+  const bool presum = G > 1024;
+  DevGuard<uint32_t> gsum_d(presum ? (G + 255) / 256 : 1);
   std::vector<NQNode> spilled;  // capacity-pressure spill, re-run after the slice
 
   auto iter = [&](int parity) {
@@ -718,8 +727,9 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
     DevCtl* next = ctl_d.p + (1 - parity);
     launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, m, Mc,
                 stream.s);
-    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, childbuf_d.p, pool_d.p, stride, G,
-                      m, Mc, capacity, stream.s);
+    if (presum) launch_presum(bc_d.p, gsum_d.p, G, stream.s);
+    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, presum ? gsum_d.p : nullptr,
+                      childbuf_d.p, pool_d.p, stride, G, m, Mc, capacity, stream.s);
   };
   ReadbackHook hook = [&](DevCtl* hc, DevCtl* live) {
     donate_if_wanted(share, hc, live, pool_d.p, m, stream.s);
@@ -831,7 +841,7 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   out.fin.best = best0;
   Result& r = out.diag;
   const int jobs = I.jobs, machines = I.machines;
-  const unsigned long long Mc = devpool_chunk_cap(M, jobs, capacity);
+  const unsigned long long Mc = devpool_chunk_cap(M, jobs, capacity, lbk);
   DevGuard<PFSPNode> pool_d(capacity);
   DevGuard<DevCtl> ctl_d(2);
   const int G = devpool_grid(Mc, jobs, lbk);
@@ -839,9 +849,9 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G);
   DevGuard<unsigned long long> bs_d(G);
-  // per-wave counts (lb2) make G large enough that gather's prefix walk needs
-  // group sums; for lb1/lb1_d the walk is short and the extra kernel is skipped
-  const bool presum = (lbk == 2);
+  // group sums keep gather's prefix walk O(G/256) (always needed for lb2's
+  // per-wave counts; needed for lb1/lb1_d once the chunk is wide)
+  const bool presum = (lbk == 2) || G > 1024;
   DevGuard<uint32_t> gsum_d(presum ? (G + 255) / 256 : 1);
 
   std::vector<PFSPNode> spilled;  // capacity-pressure spill, re-run after the slice
@@ -1419,8 +1429,8 @@ std::vector<NQNode> nq_gpu_frontier(int N, int g, size_t target, int device,
     DevCtl* next = ctl_d.p + (1 - parity);
     launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, 1, M,
                 stream.s);
-    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, childbuf_d.p, pool_d.p, stride, G,
-                      1, M, capacity, stream.s);
+    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, nullptr, childbuf_d.p, pool_d.p,
+                      stride, G, 1, M, capacity, stream.s);
   };
   DevLoopCfg cfg;
   cfg.m = 1;

@@ -67,8 +67,8 @@ void launch_pfsp_x(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf, uint32
                    hipStream_t s);
 void launch_gather2_nq(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
                        const unsigned long long* bs, const unsigned long long* be,
-                       const NQNode* childbuf, NQNode* pool, int strideNodes, int G,
-                       unsigned long long m, unsigned long long M,
+                       const uint32_t* groupSums, const NQNode* childbuf, NQNode* pool,
+                       int strideNodes, int G, unsigned long long m, unsigned long long M,
                        unsigned long long capacity, hipStream_t s);
 void launch_presum(const uint32_t* bc, uint32_t* groupSums, int G, hipStream_t s);
 void launch_gather2_pfsp(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,

@@ -1124,12 +1124,11 @@ void launch_presum(const uint32_t* bc, uint32_t* groupSums, int G, hipStream_t s
 
 void launch_gather2_nq(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
                        const unsigned long long* bs, const unsigned long long* be,
-                       const NQNode* childbuf, NQNode* pool, int strideNodes, int G,
-                       unsigned long long m, unsigned long long M,
+                       const uint32_t* groupSums, const NQNode* childbuf, NQNode* pool,
+                       int strideNodes, int G, unsigned long long m, unsigned long long M,
                        unsigned long long capacity, hipStream_t s) {
   hipLaunchKernelGGL(k_gather2<NQNode>, dim3(G), dim3(BLOCK), 0, s, ctl_cur, ctl_next, bc,
-                     bs, be, static_cast<const uint32_t*>(nullptr), childbuf, pool,
-                     strideNodes, G, m, M, capacity);
+                     bs, be, groupSums, childbuf, pool, strideNodes, G, m, M, capacity);
 }
 
 void launch_gather2_pfsp(const DevCtl* ctl_cur, DevCtl* ctl_next, const uint32_t* bc,
