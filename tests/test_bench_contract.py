@@ -57,6 +57,17 @@ def test_bench_json_contract_world1_pfsp():
     assert out["explored_tree_per_step"] == 2573652
 
 
+def test_bench_world8_dry_run():
+    # the driver's SCALE run launches bench.py at --nproc-per-node 8; dry-run
+    # that exact world size on CPU so the rendezvous, 8-way slicing and
+    # reductions cannot fail on plumbing when a real 8-GPU node appears
+    out = run_bench(8, ["--steps", "1", "--warmup", "0", "--N", "11",
+                        "--inst", "14", "--lb", "lb1_d"])
+    assert out["n_gpus"] == 8
+    assert out["config"]["nqueens_tree_per_step"] == 166925
+    assert out["config"]["pfsp_tree_per_step"] == 2573652
+
+
 def test_bench_headline_combines_both_configs():
     # default --problem headline runs BOTH BASELINE.json configs per step
     # (small stand-ins here: N=10 + ta014 lb1_d)
