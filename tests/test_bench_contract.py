@@ -21,11 +21,13 @@ def run_bench(nproc, extra):
            "--nnodes=1", f"--nproc-per-node={nproc}",
            "--standalone", "--local-addr", "127.0.0.1",
            os.path.join(ROOT, "bench.py"), "--gpus", str(nproc)] + extra
-    for attempt in range(3):  # --standalone rendezvous can transiently fail
+    import time
+    for attempt in range(4):  # --standalone rendezvous can transiently fail
         r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env,
                            cwd=ROOT)
         if r.returncode == 0:
             break
+        time.sleep(2 * (attempt + 1))
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
     lines = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
     assert len(lines) == 1, r.stdout  # exactly ONE JSON line, from rank 0

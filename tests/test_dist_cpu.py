@@ -11,15 +11,18 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 import pytest
 
 
-def run_torchrun(cmd, env, tries=3):
+def run_torchrun(cmd, env, tries=4):
     """torchrun --standalone can transiently fail its rendezvous bind when
-    other tests' agents are tearing down; retry a couple of times before
+    other tests' agents are tearing down; retry with a pause before
     declaring failure."""
+    import time
+
     for attempt in range(tries):
         r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, env=env,
                            cwd=os.path.dirname(HERE))
         if r.returncode == 0:
             return r
+        time.sleep(2 * (attempt + 1))
     return r
 
 
