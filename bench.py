@@ -30,8 +30,10 @@ from gats_amd import dist as gdist  # noqa: E402
 def parse_args():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=5)
-    ap.add_argument("--warmup", type=int, default=2)
+    # default timed region ~2 s (25 x ~80 ms headline steps): long enough for
+    # the driver's SMI gpu_busy sampler to land inside it, still < minutes
+    ap.add_argument("--steps", type=int, default=25)
+    ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--problem", default="headline",
                     choices=["headline", "nqueens", "pfsp"])
     ap.add_argument("--N", type=int, default=17)
