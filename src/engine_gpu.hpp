@@ -14,6 +14,9 @@
 namespace gats {
 
 int gpu_device_count();
+// hipSetDevice only if this thread isn't already on `device` (the raw call
+// costs ~0.8 ms on ROCm 7.2; see engine_gpu.cpp)
+void set_device_cached(int device);
 
 // Phase-2+3 engine cores (phase-1 pool supplied by the caller); used by the
 // CLI engines, the distributed tier and the multi-GPU tier's devpool workers.

@@ -156,7 +156,7 @@ struct NqGpuCtx {
   uint8_t* labels_d = nullptr;
   int N, g, M;
   NqGpuCtx(int device, int N_, int g_, int M_) : N(N_), g(g_), M(M_) {
-    HIP_CHECK_M(hipSetDevice(device));
+    set_device_cached(device);
     HIP_CHECK_M(hipStreamCreate(&stream));
     HIP_CHECK_M(hipHostMalloc(reinterpret_cast<void**>(&parents_h), M * sizeof(NQNode)));
     HIP_CHECK_M(hipHostMalloc(reinterpret_cast<void**>(&labels_h), size_t(M) * N));
@@ -192,7 +192,7 @@ struct PfspGpuCtx {
 
   PfspGpuCtx(int device, const PfspInstance& I, int M_)
       : jobs(I.jobs), machines(I.machines), M(M_) {
-    HIP_CHECK_M(hipSetDevice(device));
+    set_device_cached(device);
     HIP_CHECK_M(hipStreamCreate(&stream));
     HIP_CHECK_M(hipMalloc(reinterpret_cast<void**>(&parents_d), M * sizeof(PFSPNode)));
     HIP_CHECK_M(hipMalloc(reinterpret_cast<void**>(&bounds_d),
