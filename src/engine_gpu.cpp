@@ -150,12 +150,6 @@ struct PfspTablesGuard {
     std::vector<int16_t> p16(static_cast<size_t>(m) * n);
     for (size_t i = 0; i < p16.size(); i++) p16[i] = static_cast<int16_t>(I.lb1.p_times[i]);
     std::vector<int32_t> mt(I.lb1.min_tails.begin(), I.lb1.min_tails.end());
-    std::vector<int16_t> lags16(static_cast<size_t>(pairs) * n);
-    std::vector<uint8_t> js8(static_cast<size_t>(pairs) * n);
-    for (size_t i = 0; i < lags16.size(); i++) {
-      lags16[i] = static_cast<int16_t>(I.lb2.lags[i]);
-      js8[i] = static_cast<uint8_t>(I.lb2.johnson_schedules[i]);
-    }
     std::vector<uint8_t> p1(pairs), p2(pairs);
     for (int i = 0; i < pairs; i++) {
       p1[i] = static_cast<uint8_t>(I.lb2.pairs1[i]);
@@ -177,8 +171,6 @@ struct PfspTablesGuard {
     }
     tb.p_times = keep(dev_upload(p16.data(), p16.size()), p16.size() * sizeof(p16[0]));
     tb.min_tails = keep(dev_upload(mt.data(), mt.size()), mt.size() * sizeof(mt[0]));
-    tb.lags = keep(dev_upload(lags16.data(), lags16.size()), lags16.size() * sizeof(lags16[0]));
-    tb.johnson_schedules = keep(dev_upload(js8.data(), js8.size()), js8.size() * sizeof(js8[0]));
     tb.johnson_packed = keep(dev_upload(jp.data(), jp.size()), jp.size() * sizeof(jp[0]));
     tb.pairs1 = keep(dev_upload(p1.data(), p1.size()), p1.size() * sizeof(p1[0]));
     tb.pairs2 = keep(dev_upload(p2.data(), p2.size()), p2.size() * sizeof(p2[0]));

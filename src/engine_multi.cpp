@@ -202,8 +202,6 @@ struct PfspGpuCtx {
     const int pairs = I.lb2.nb_pairs;
     std::vector<int16_t> p16(I.lb1.p_times.begin(), I.lb1.p_times.end());
     std::vector<int32_t> mt(I.lb1.min_tails.begin(), I.lb1.min_tails.end());
-    std::vector<int16_t> lags16(I.lb2.lags.begin(), I.lb2.lags.end());
-    std::vector<uint8_t> js8(I.lb2.johnson_schedules.begin(), I.lb2.johnson_schedules.end());
     std::vector<uint8_t> p1(pairs), p2(pairs);
     for (int i = 0; i < pairs; i++) {
       p1[i] = static_cast<uint8_t>(I.lb2.pairs1[i]);
@@ -225,8 +223,6 @@ struct PfspGpuCtx {
     }
     tb.p_times = upload(p16);
     tb.min_tails = upload(mt);
-    tb.lags = upload(lags16);
-    tb.johnson_schedules = upload(js8);
     tb.johnson_packed = upload(jp);
     tb.pairs1 = upload(p1);
     tb.pairs2 = upload(p2);

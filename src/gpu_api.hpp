@@ -33,12 +33,12 @@ struct DevCtl {
 struct PfspDevTables {
   const int16_t* p_times;            // [machines * jobs]
   const int32_t* min_tails;          // [machines]
-  const int16_t* lags;               // [pairs * jobs]
-  const uint8_t* johnson_schedules;  // [pairs * jobs]
   // packed per (pair, johnson position): job<<27 | lag<<16 | ptm1<<8 | ptm0
   // (one ds_read_b32 per inner step instead of 4 scalar LDS reads; lossless:
   // job < 20 -> 5 bits, lag <= 20*99 -> 11 bits, ptm <= 99 -> 8 bits each;
-  // round-1 used a u64 pack — u32 halves the LDS table and its bandwidth)
+  // round-1 used a u64 pack — u32 halves the LDS table and its bandwidth.
+  // The raw lags/johnson_schedules arrays the reference ships separately,
+  // c_bound_johnson.h, live only on the host now.)
   const uint32_t* johnson_packed;    // [pairs * jobs]
   const uint8_t* pairs1;             // [pairs]
   const uint8_t* pairs2;             // [pairs]
