@@ -796,10 +796,17 @@ struct LdsLb2w {
   uint32_t jp[PAIRS * MAX_JOBS];
   uint8_t pair1[PAIRS], pair2[PAIRS];
   uint16_t fronts[BLOCK][MM + 1];  // child completion times (values <= 20*20*99)
-  uint32_t smask[BLOCK];           // scheduled-job bitmask per child
   uint8_t lpar[BLOCK];             // child's parent index in snodes
   int8_t ck[BLOCK];                // child's k, or -1 invalid, <=-2 leaf (-2-k)
   PFSPNode snodes[BLOCK / 5 + 2];
+  // per-PARENT prefix state (phase A0): every child of a parent shares the
+  // parent's front schedule and scheduled mask, so they are computed once
+  // per parent and each child's phase A is a single O(m) forward step.
+  // (The per-child scheduled mask is pmask[parent] | bit(job_k), derived in
+  // phase B — keeping a per-child copy would cost 1 KB of LDS and a
+  // block/CU of occupancy.)
+  uint16_t pfront[BLOCK / 5 + 2][MM + 1];
+  uint32_t pmask[BLOCK / 5 + 2];
 };
 
 __device__ inline int wave_max_i32(int v) {
@@ -836,7 +843,35 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
   }
   __syncthreads();
 
-  // ---- phase A: one thread per child slot ----
+  // ---- phase A0: one thread per PARENT (prefix front + scheduled mask) ----
+  if (c0 < total) {
+    uint32_t c1 = c0 + BLOCK;
+    if (c1 > total) c1 = total;
+    const int nblk = static_cast<int>((c1 - 1) / jobs - first + 1);
+    for (int pi = threadIdx.x; pi < nblk; pi += blockDim.x) {
+      const PFSPNode& p = lds.snodes[pi];
+      int front[MM];
+#pragma unroll
+      for (int i = 0; i < MM; i++) front[i] = 0;
+      uint32_t sched = 0;
+      for (int i = 0; i < p.depth; i++) {
+        const int job = p.prmu[i];
+        sched |= 1u << job;
+        front[0] += lds.p[job];
+#pragma unroll
+        for (int j = 1; j < MM; j++)
+          front[j] = max(front[j - 1], front[j]) + lds.p[j * jobs + job];
+      }
+#pragma unroll
+      for (int i = 0; i < MM; i++)
+        lds.pfront[pi][i] = static_cast<uint16_t>(front[i]);
+      lds.pmask[pi] = sched;
+    }
+  }
+  __syncthreads();
+
+  // ---- phase A: one thread per child slot — a single O(m) step from the
+  // parent's front ----
   {
     const uint32_t t = c0 + threadIdx.x;
     int8_t state = -1;
@@ -846,22 +881,26 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
       const PFSPNode& p = lds.snodes[pid - first];
       lds.lpar[threadIdx.x] = static_cast<uint8_t>(pid - first);
       if (k >= p.limit1 + 1) {
-        int front[MM];
-        child_front<MM>(lds, p.prmu, p.depth, p.prmu[k], jobs, front, 1);
-        uint32_t sched = 0;
-        for (int i = 0; i < p.depth; i++) sched |= 1u << p.prmu[i];
-        sched |= 1u << p.prmu[k];
-        lds.smask[threadIdx.x] = sched;
+        const int job_k = p.prmu[k];
+        const uint16_t* pf = lds.pfront[pid - first];
+        int prev = pf[0] + lds.p[job_k];
+        lds.fronts[threadIdx.x][0] = static_cast<uint16_t>(prev);
 #pragma unroll
-        for (int i = 0; i < MM; i++)
-          lds.fronts[threadIdx.x][i] = static_cast<uint16_t>(front[i]);
+        for (int j = 1; j < MM; j++) {
+          prev = max(prev, static_cast<int>(pf[j])) + lds.p[j * jobs + job_k];
+          lds.fronts[threadIdx.x][j] = static_cast<uint16_t>(prev);
+        }
         state = (p.depth + 1 == jobs) ? static_cast<int8_t>(-2 - k)
                                       : static_cast<int8_t>(k);
       }
     }
     lds.ck[threadIdx.x] = state;
   }
-  __syncthreads();
+  // phase B consumes only THIS wave's phase-A slots (ct = wid*64 + cc, all
+  // written by lanes of the same wave), so no block barrier is needed — just
+  // order this wave's own LDS traffic
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_wave_barrier();
 
   // ---- phase B: one child per wave at a time, pairs across lanes ----
   const int wid = threadIdx.x >> 6;
@@ -879,7 +918,8 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
     const bool leaf = state <= -2;
     const int k = leaf ? (-2 - state) : state;
     const uint16_t* fr = lds.fronts[ct];
-    const uint32_t sched = lds.smask[ct];
+    const int lp = lds.lpar[ct];
+    const uint32_t sched = lds.pmask[lp] | (1u << lds.snodes[lp].prmu[k]);
 
     int mylb = 0;
     bool exited = false;
