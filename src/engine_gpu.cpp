@@ -138,43 +138,27 @@ T* dev_upload(const T* src, size_t n) {
 }
 
 
-// Owns the int16/uint8-compressed PFSP bound tables on device (identity pair
-// order — the kernels' pair map is the compile-time lexicographic one).
+// Owns the int16/uint8-compressed PFSP bound tables on device.
 struct PfspTablesGuard {
   PfspDevTables tb{};
   std::vector<void*> allocs;
 
   PfspTablesGuard(const PfspInstance& I) {
     const int n = I.jobs, m = I.machines;
-    const int pairs = I.lb2.nb_pairs;
     std::vector<int16_t> p16(static_cast<size_t>(m) * n);
     for (size_t i = 0; i < p16.size(); i++) p16[i] = static_cast<int16_t>(I.lb1.p_times[i]);
     std::vector<int32_t> mt(I.lb1.min_tails.begin(), I.lb1.min_tails.end());
-    std::vector<uint8_t> p1(pairs), p2(pairs);
-    for (int i = 0; i < pairs; i++) {
-      p1[i] = static_cast<uint8_t>(I.lb2.pairs1[i]);
-      p2[i] = static_cast<uint8_t>(I.lb2.pairs2[i]);
-    }
-    // u32 pack (lossless: job < 20, lag <= 20*99 < 2^11, ptm <= 99 < 2^8)
-    std::vector<uint32_t> jp(static_cast<size_t>(pairs) * n);
-    for (int k = 0; k < pairs; k++) {
-      const int ma0 = I.lb2.pairs1[k], ma1 = I.lb2.pairs2[k];
-      for (int j = 0; j < n; j++) {
-        const int job = I.lb2.johnson_schedules[static_cast<size_t>(k) * n + j];
-        const uint32_t ptm0 = static_cast<uint32_t>(I.lb1.p_times[ma0 * n + job]);
-        const uint32_t ptm1 = static_cast<uint32_t>(I.lb1.p_times[ma1 * n + job]);
-        const uint32_t lag =
-            static_cast<uint32_t>(I.lb2.lags[static_cast<size_t>(k) * n + job]);
-        jp[static_cast<size_t>(k) * n + j] =
-            (static_cast<uint32_t>(job) << 27) | (lag << 16) | (ptm1 << 8) | ptm0;
-      }
-    }
+    const PfspPackedTables pk = build_packed_johnson(I);
     tb.p_times = keep(dev_upload(p16.data(), p16.size()), p16.size() * sizeof(p16[0]));
     tb.min_tails = keep(dev_upload(mt.data(), mt.size()), mt.size() * sizeof(mt[0]));
-    tb.johnson_packed = keep(dev_upload(jp.data(), jp.size()), jp.size() * sizeof(jp[0]));
-    tb.pairs1 = keep(dev_upload(p1.data(), p1.size()), p1.size() * sizeof(p1[0]));
-    tb.pairs2 = keep(dev_upload(p2.data(), p2.size()), p2.size() * sizeof(p2[0]));
-
+    tb.johnson_packed =
+        keep(dev_upload(pk.jp.data(), pk.jp.size()), pk.jp.size() * sizeof(uint32_t));
+    tb.pairs1 = keep(dev_upload(pk.p1.data(), pk.p1.size()), pk.p1.size());
+    tb.pairs2 = keep(dev_upload(pk.p2.data(), pk.p2.size()), pk.p2.size());
+    tb.johnson_packed_w =
+        keep(dev_upload(pk.jp_w.data(), pk.jp_w.size()), pk.jp_w.size() * sizeof(uint32_t));
+    tb.pairs1_w = keep(dev_upload(pk.p1_w.data(), pk.p1_w.size()), pk.p1_w.size());
+    tb.pairs2_w = keep(dev_upload(pk.p2_w.data(), pk.p2_w.size()), pk.p2_w.size());
   }
   std::vector<size_t> alloc_bytes;
   template <typename T>
@@ -334,6 +318,61 @@ const PfspDevTables& pfsp_tables_cached(const PfspInstance& I, int device) {
     it = cache.emplace(key, new PfspTablesGuard(I)).first;  // lives for the process
   return it->second->tb;
 }
+
+}  // namespace
+
+PfspPackedTables build_packed_johnson(const PfspInstance& I) {
+  const int n = I.jobs;
+  const int pairs = I.lb2.nb_pairs;
+  PfspPackedTables pk;
+  pk.p1.resize(pairs);
+  pk.p2.resize(pairs);
+  // u32 pack (lossless: job < 20, lag <= 20*99 < 2^11, ptm <= 99 < 2^8)
+  pk.jp.resize(static_cast<size_t>(pairs) * n);
+  for (int k = 0; k < pairs; k++) {
+    const int ma0 = I.lb2.pairs1[k], ma1 = I.lb2.pairs2[k];
+    pk.p1[k] = static_cast<uint8_t>(ma0);
+    pk.p2[k] = static_cast<uint8_t>(ma1);
+    for (int j = 0; j < n; j++) {
+      const int job = I.lb2.johnson_schedules[static_cast<size_t>(k) * n + j];
+      const uint32_t ptm0 = static_cast<uint32_t>(I.lb1.p_times[ma0 * n + job]);
+      const uint32_t ptm1 = static_cast<uint32_t>(I.lb1.p_times[ma1 * n + job]);
+      const uint32_t lag =
+          static_cast<uint32_t>(I.lb2.lags[static_cast<size_t>(k) * n + job]);
+      pk.jp[static_cast<size_t>(k) * n + j] =
+          (static_cast<uint32_t>(job) << 27) | (lag << 16) | (ptm1 << 8) | ptm0;
+    }
+  }
+  // wave-kernel order: widest machine span first (strongest pairs in the
+  // first 64-pair round -> the collective early exit fires sooner); counts
+  // are order-invariant — the bound VALUE is the max over all pairs and the
+  // exit decision (partial max > best) implies the full max's decision
+  std::vector<int> perm(pairs);
+  for (int i = 0; i < pairs; i++) perm[i] = i;
+  const char* ord = std::getenv("GATS_LB2_ORDER");
+  if (ord == nullptr || std::string(ord) != "lex") {
+    std::sort(perm.begin(), perm.end(), [&](int a, int b) {
+      const int sa = I.lb2.pairs2[a] - I.lb2.pairs1[a];
+      const int sb = I.lb2.pairs2[b] - I.lb2.pairs1[b];
+      if (sa != sb) return sa > sb;
+      return a < b;
+    });
+  }
+  pk.p1_w.resize(pairs);
+  pk.p2_w.resize(pairs);
+  pk.jp_w.resize(pk.jp.size());
+  for (int l = 0; l < pairs; l++) {
+    const int k = perm[l];
+    pk.p1_w[l] = pk.p1[k];
+    pk.p2_w[l] = pk.p2[k];
+    std::copy(pk.jp.begin() + static_cast<size_t>(k) * n,
+              pk.jp.begin() + static_cast<size_t>(k + 1) * n,
+              pk.jp_w.begin() + static_cast<size_t>(l) * n);
+  }
+  return pk;
+}
+
+namespace {
 
 int lbk_of(LbKind lb) {
   switch (lb) {

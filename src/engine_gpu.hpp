@@ -71,6 +71,15 @@ DevpoolMultiOut pfsp_devpool_multi(const PfspInstance& I, Pool<PFSPNode>& pool, 
                                    std::atomic<int>* shared_best, Result& diag,
                                    ExtractShare* extract = nullptr);
 
+// Host-side build of the packed Johnson tables: lexicographic order (per-lane
+// kernels, compile-time pair map) and strength order (wave kernel; see
+// PfspDevTables). Shared by the engine and multigpu table uploaders.
+struct PfspPackedTables {
+  std::vector<uint32_t> jp, jp_w;
+  std::vector<uint8_t> p1, p2, p1_w, p2_w;
+};
+PfspPackedTables build_packed_johnson(const PfspInstance& I);
+
 Result nqueens_gpu(int N, int g, int m, int M, int device, const std::string& mode,
                    unsigned long long capacity);
 Result nqueens_gpu_from_pool(const std::vector<NQNode>& nodes, int N, int g, int m, int M,

@@ -190,6 +190,15 @@ struct PfspGpuCtx {
     return static_cast<T*>(p);
   }
 
+  template <typename T>
+  T* upload_n(const T* src, size_t nelem) {
+    void* p = nullptr;
+    HIP_CHECK_M(hipMalloc(&p, nelem * sizeof(T)));
+    HIP_CHECK_M(hipMemcpy(p, src, nelem * sizeof(T), hipMemcpyHostToDevice));
+    tb_allocs.push_back(p);
+    return static_cast<T*>(p);
+  }
+
   PfspGpuCtx(int device, const PfspInstance& I, int M_)
       : jobs(I.jobs), machines(I.machines), M(M_) {
     set_device_cached(device);
@@ -199,33 +208,17 @@ struct PfspGpuCtx {
                           size_t(M) * jobs * sizeof(int32_t)));
     HIP_CHECK_M(hipHostMalloc(reinterpret_cast<void**>(&bounds_h),
                               size_t(M) * jobs * sizeof(int32_t)));
-    const int pairs = I.lb2.nb_pairs;
     std::vector<int16_t> p16(I.lb1.p_times.begin(), I.lb1.p_times.end());
     std::vector<int32_t> mt(I.lb1.min_tails.begin(), I.lb1.min_tails.end());
-    std::vector<uint8_t> p1(pairs), p2(pairs);
-    for (int i = 0; i < pairs; i++) {
-      p1[i] = static_cast<uint8_t>(I.lb2.pairs1[i]);
-      p2[i] = static_cast<uint8_t>(I.lb2.pairs2[i]);
-    }
-    // u32 pack (lossless; see PfspDevTables::johnson_packed)
-    std::vector<uint32_t> jp(static_cast<size_t>(pairs) * jobs);
-    for (int k = 0; k < pairs; k++) {
-      const int ma0 = I.lb2.pairs1[k], ma1 = I.lb2.pairs2[k];
-      for (int j = 0; j < jobs; j++) {
-        const int job = I.lb2.johnson_schedules[static_cast<size_t>(k) * jobs + j];
-        const uint32_t ptm0 = static_cast<uint32_t>(I.lb1.p_times[ma0 * jobs + job]);
-        const uint32_t ptm1 = static_cast<uint32_t>(I.lb1.p_times[ma1 * jobs + job]);
-        const uint32_t lag =
-            static_cast<uint32_t>(I.lb2.lags[static_cast<size_t>(k) * jobs + job]);
-        jp[static_cast<size_t>(k) * jobs + j] =
-            (static_cast<uint32_t>(job) << 27) | (lag << 16) | (ptm1 << 8) | ptm0;
-      }
-    }
+    const PfspPackedTables pk = build_packed_johnson(I);
     tb.p_times = upload(p16);
     tb.min_tails = upload(mt);
-    tb.johnson_packed = upload(jp);
-    tb.pairs1 = upload(p1);
-    tb.pairs2 = upload(p2);
+    tb.johnson_packed = upload_n(pk.jp.data(), pk.jp.size());
+    tb.pairs1 = upload_n(pk.p1.data(), pk.p1.size());
+    tb.pairs2 = upload_n(pk.p2.data(), pk.p2.size());
+    tb.johnson_packed_w = upload_n(pk.jp_w.data(), pk.jp_w.size());
+    tb.pairs1_w = upload_n(pk.p1_w.data(), pk.p1_w.size());
+    tb.pairs2_w = upload_n(pk.p2_w.data(), pk.p2_w.size());
   }
   ~PfspGpuCtx() {
     (void)hipStreamDestroy(stream);

@@ -39,9 +39,18 @@ struct PfspDevTables {
   // round-1 used a u64 pack — u32 halves the LDS table and its bandwidth.
   // The raw lags/johnson_schedules arrays the reference ships separately,
   // c_bound_johnson.h, live only on the host now.)
-  const uint32_t* johnson_packed;    // [pairs * jobs]
+  const uint32_t* johnson_packed;    // [pairs * jobs] (lexicographic order)
   const uint8_t* pairs1;             // [pairs]
   const uint8_t* pairs2;             // [pairs]
+  // wave-kernel copies in STRENGTH order (widest machine span first): the
+  // collective early exit checks a shfl-max after every 64-pair round, so
+  // putting the strongest pairs in round 0 exits sooner. The per-lane
+  // kernels keep the compile-time lexicographic map (their `front` is a
+  // register array, so pair machine ids must be compile-time constants).
+  // GATS_LB2_ORDER=lex disables the reorder (host-side, at table build).
+  const uint32_t* johnson_packed_w;  // [pairs * jobs]
+  const uint8_t* pairs1_w;           // [pairs]
+  const uint8_t* pairs2_w;           // [pairs]
 };
 
 

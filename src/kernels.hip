@@ -824,11 +824,13 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
   constexpr int PAIRS = LdsLb2w<MM>::PAIRS;
   for (int i = threadIdx.x; i < MM * jobs; i += blockDim.x) lds.p[i] = tb.p_times[i];
   if (threadIdx.x < MM) lds.min_tails[threadIdx.x] = tb.min_tails[threadIdx.x];
+  // strength-ordered tables: strongest pairs land in the first 64-pair round
+  // so the collective early exit fires there (see PfspDevTables)
   for (int i = threadIdx.x; i < PAIRS * jobs; i += blockDim.x)
-    lds.jp[i] = tb.johnson_packed[i];
+    lds.jp[i] = tb.johnson_packed_w[i];
   if (threadIdx.x < PAIRS) {
-    lds.pair1[threadIdx.x] = static_cast<uint8_t>(pair_first<MM>(threadIdx.x));
-    lds.pair2[threadIdx.x] = static_cast<uint8_t>(pair_second<MM>(threadIdx.x));
+    lds.pair1[threadIdx.x] = tb.pairs1_w[threadIdx.x];
+    lds.pair2[threadIdx.x] = tb.pairs2_w[threadIdx.x];
   }
 
   const unsigned long long c = derive_chunk(ctl, m, M);
