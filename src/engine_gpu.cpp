@@ -349,8 +349,19 @@ PfspPackedTables build_packed_johnson(const PfspInstance& I) {
   // exit decision (partial max > best) implies the full max's decision
   std::vector<int> perm(pairs);
   for (int i = 0; i < pairs; i++) perm[i] = i;
-  const char* ord = std::getenv("GATS_LB2_ORDER");
-  if (ord == nullptr || std::string(ord) != "lex") {
+  const char* orde = std::getenv("GATS_LB2_ORDER");
+  const std::string ord = orde ? orde : "span";
+  if (ord == "lagsum") {
+    // tie-break proxy: pairs whose relaxation carries the most lag mass
+    std::vector<long long> ls(pairs, 0);
+    for (int k = 0; k < pairs; k++)
+      for (int j = 0; j < n; j++)
+        ls[k] += I.lb2.lags[static_cast<size_t>(k) * n + j];
+    std::sort(perm.begin(), perm.end(), [&](int a, int b) {
+      if (ls[a] != ls[b]) return ls[a] > ls[b];
+      return a < b;
+    });
+  } else if (ord != "lex") {  // default: widest machine span first
     std::sort(perm.begin(), perm.end(), [&](int a, int b) {
       const int sa = I.lb2.pairs2[a] - I.lb2.pairs1[a];
       const int sb = I.lb2.pairs2[b] - I.lb2.pairs1[b];
