@@ -11,14 +11,14 @@ python build.py
 echo "== GPU test suite (kernel numerics vs CPU oracle + engine count parity) =="
 python -m pytest tests -m gpu -q
 
-echo "== N-Queens N=17 (BASELINE config 2: ~85-95 ms/search, ~90 Gnodes/s) =="
-python bench.py --gpus 1 --steps 5 --warmup 2
+echo "== headline (N=17 search + ta014 lb1 proof per step; ~77 ms, ~104 Gnodes/s) =="
+python bench.py --gpus 1 --steps 25 --warmup 3
 
-echo "== PFSP ta014 lb1 (BASELINE config 3) =="
-python bench.py --gpus 1 --steps 5 --warmup 2 --problem pfsp --inst 14 --lb lb1
+echo "== PFSP ta014 lb1 alone (BASELINE config 3: ~2.1 ms/search) =="
+python bench.py --gpus 1 --steps 25 --warmup 5 --problem pfsp --inst 14 --lb lb1
 
 if [ "$MODE" = "full" ]; then
-  echo "== N-Queens N=18 (~0.9 s) =="
+  echo "== N-Queens N=18 (~0.6 s) =="
   python -m gats_amd.cli nqueens --N 18 --tier gpu
   echo "== PFSP ta001-ta020 lb2 sweep, optima proven from ub=1 (~31 s total) =="
   for i in $(seq 1 20); do
