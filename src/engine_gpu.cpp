@@ -580,17 +580,18 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, const DevLoopCfg& c
 // finish. Counts are slice-order independent.
 // ---------------------------------------------------------------------------
 
-static int devpool_slices() {
+// Concurrent slice chains per device. Wide-chunk paths (N-Queens, lb1,
+// lb1_d) fill the chip from one ~512k-node launch, so 2 chains suffice to
+// hide kernel-boundary bubbles (measured N=17: S=1 89.3 ms, S=2 77.4,
+// S=4 79.6, S=6 84.2). lb2 keeps the narrow --M chunk (per-wave grid), so
+// its kernels are short and need 4 chains (ta005 20x5: S=2 41.4 s,
+// S=4 31.5 s; ta021 20x20 is S-insensitive).
+static int devpool_slices(int lbk = 1) {
   if (const char* e = std::getenv("GATS_SLICES")) {
     int v = atoi(e);
     return v < 1 ? 1 : v;
   }
-  // 2 concurrent slice chains: with the ~512k-node expansion chunk one
-  // launch already fills the 256 CUs, so extra slices only add overhead —
-  // they exist to hide the kernel-boundary bubbles of one chain (S=1 is
-  // ~15% slower; measured N=17: S=1 89.3 ms, S=2 77.4, S=4 79.6, S=6 84.2;
-  // ta021 lb2 and N=18/19 are S-insensitive at wide chunks)
-  return 2;
+  return (lbk == 2) ? 4 : 2;
 }
 
 // Internal per-iteration expansion width of the devpool. The reference's M
@@ -1135,7 +1136,7 @@ DevpoolMultiOut pfsp_devpool_multi(const PfspInstance& I, Pool<PFSPNode>& pool, 
   if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
     throw std::invalid_argument("devpool requires M * jobs <= 2^31");
   const int D = static_cast<int>(devices.size());
-  int S = devpool_slices();
+  int S = devpool_slices(lbk);
   int maxd = 0;  // same deep-frontier rule as N-Queens (12+ open jobs)
   for (size_t i = 0; i < pool.size(); i++)
     maxd = std::max(maxd, static_cast<int>(pool.data()[i].depth));
