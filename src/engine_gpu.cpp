@@ -895,17 +895,18 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   out.fin.best = best0;
   Result& r = out.diag;
   const int jobs = I.jobs, machines = I.machines;
-  const unsigned long long Mc = devpool_chunk_cap(M, jobs, capacity, lbk);
+  const int lbg = devpool_lbk_geom(lbk, machines);  // 3 = per-lane lb2
+  const unsigned long long Mc = devpool_chunk_cap(M, jobs, capacity, lbg);
   DevGuard<PFSPNode> pool_d(capacity);
   DevGuard<DevCtl> ctl_d(2);
-  const int G = devpool_grid(Mc, jobs, lbk);
-  const int stride = devpool_stride(lbk);
+  const int G = devpool_grid(Mc, jobs, lbg);
+  const int stride = devpool_stride(lbg);
   DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G);
   DevGuard<unsigned long long> bs_d(G);
   // group sums keep gather's prefix walk O(G/256) (always needed for lb2's
-  // per-wave counts; needed for lb1/lb1_d once the chunk is wide)
-  const bool presum = (lbk == 2) || G > 1024;
+  // per-wave counts; needed for thread-per-child paths once the chunk is wide)
+  const bool presum = (lbg == 2) || G > 1024;
   DevGuard<uint32_t> gsum_d(presum ? (G + 255) / 256 : 1);
 
   std::vector<PFSPNode> spilled;  // capacity-pressure spill, re-run after the slice
@@ -913,7 +914,7 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
-    launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk,
+    launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbg,
                   tb, m, Mc, stream.s);
     if (presum) launch_presum(bc_d.p, gsum_d.p, G, stream.s);
     launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, presum ? gsum_d.p : nullptr,
@@ -1136,7 +1137,7 @@ DevpoolMultiOut pfsp_devpool_multi(const PfspInstance& I, Pool<PFSPNode>& pool, 
   if (static_cast<unsigned long long>(M) * jobs > (1ull << 31))
     throw std::invalid_argument("devpool requires M * jobs <= 2^31");
   const int D = static_cast<int>(devices.size());
-  int S = devpool_slices(lbk);
+  int S = devpool_slices(devpool_lbk_geom(lbk, I.machines));
   int maxd = 0;  // same deep-frontier rule as N-Queens (12+ open jobs)
   for (size_t i = 0; i < pool.size(); i++)
     maxd = std::max(maxd, static_cast<int>(pool.data()[i].depth));
@@ -1515,16 +1516,17 @@ std::vector<PFSPNode> pfsp_gpu_frontier(const PfspInstance& I, int lbk, size_t t
   StreamGuard stream;
   Result r;
   const int jobs = I.jobs, machines = I.machines;
+  const int lbg = devpool_lbk_geom(lbk, machines);
   const unsigned long long M = target;
   const unsigned long long capacity = target * (MAX_JOBS + 1) + 64;
   DevGuard<PFSPNode> pool_d(capacity);
   DevGuard<DevCtl> ctl_d(2);
-  const int G = devpool_grid(M, jobs, lbk);
-  const int stride = devpool_stride(lbk);
+  const int G = devpool_grid(M, jobs, lbg);
+  const int stride = devpool_stride(lbg);
   DevGuard<PFSPNode> childbuf_d(static_cast<size_t>(G) * stride);
   DevGuard<uint32_t> bc_d(G);
   DevGuard<unsigned long long> bs_d(G);
-  const bool presum = (lbk == 2);
+  const bool presum = (lbg == 2);
   DevGuard<uint32_t> gsum_d(presum ? (G + 255) / 256 : 1);
 
   PFSPNode root = pfsp_root();
@@ -1540,7 +1542,7 @@ std::vector<PFSPNode> pfsp_gpu_frontier(const PfspInstance& I, int lbk, size_t t
   auto iter = [&](int parity) {
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
-    launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbk, tb, 1,
+    launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbg, tb, 1,
                   M, stream.s);
     if (presum) launch_presum(bc_d.p, gsum_d.p, G, stream.s);
     launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, presum ? gsum_d.p : nullptr,

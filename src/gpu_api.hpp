@@ -64,6 +64,9 @@ void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, in
                       const PfspDevTables& tb, int best, int32_t* bounds, hipStream_t s);
 
 // devpool v3 (see kernels.hip): expand -> scan -> gather, 3 kernels/iteration.
+// lbk geometry codes: 0 lb1_d, 1 thread-per-child, 2 wave-coop lb2,
+// 3 per-lane lb2 with thread-per-child geometry (machines <= 10).
+int devpool_lbk_geom(int lbk, int machines);
 int devpool_grid(unsigned long long M, int per, int lbk);
 int devpool_stride(int lbk);
 void launch_nq_x(const DevCtl* ctl, const NQNode* pool, NQNode* childbuf,

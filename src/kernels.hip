@@ -1099,6 +1099,16 @@ void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, in
 
 // ---- devpool v3 launchers ----
 
+// lbk geometry codes: 0 = lb1_d (thread per parent), 1 = thread-per-child
+// (lb1), 2 = wave-cooperative lb2 (per-wave slabs), 3 = PER-LANE lb2 with
+// thread-per-child geometry — used when machines <= 10: with only 10-45
+// machine pairs the wave-cooperative kernel leaves most of a wave idle and
+// its collective early exit buys little, while the per-lane full sweep is a
+// short fully-unrolled register loop (ta005 20x5 measured 3x faster).
+int devpool_lbk_geom(int lbk, int machines) {
+  return (lbk == 2 && machines <= 10) ? 3 : lbk;
+}
+
 int devpool_grid(unsigned long long M, int per, int lbk) {
   if (lbk == 0)  // lb1_d: one thread per parent
     return static_cast<int>((M + BLOCK - 1) / BLOCK);
@@ -1138,6 +1148,9 @@ static void launch_pfsp_x_mm(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childb
                        ctl, pool, childbuf, bc, bs, jobs, tb, m, M);
   } else if (lbk == 1) {
     hipLaunchKernelGGL((k_pfsp_x<MM, 1>), dim3(devpool_grid(M, jobs, 1)), dim3(BLOCK), 0, s,
+                       ctl, pool, childbuf, bc, bs, jobs, tb, m, M);
+  } else if (lbk == 3) {  // per-lane lb2, thread-per-child geometry
+    hipLaunchKernelGGL((k_pfsp_x<MM, 2>), dim3(devpool_grid(M, jobs, 1)), dim3(BLOCK), 0, s,
                        ctl, pool, childbuf, bc, bs, jobs, tb, m, M);
   } else {
     // 256 child slots per block (4 waves x 64); count arrays are per wave
