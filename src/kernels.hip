@@ -24,6 +24,7 @@
 //   * Wavefront = 64 idioms throughout (__ballot is 64-bit).
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
 #include <type_traits>
 
 #include "gpu_api.hpp"
@@ -1106,7 +1107,10 @@ void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, in
 // its collective early exit buys little, while the per-lane full sweep is a
 // short fully-unrolled register loop (ta005 20x5 measured 3x faster).
 int devpool_lbk_geom(int lbk, int machines) {
-  return (lbk == 2 && machines <= 10) ? 3 : lbk;
+  if (lbk != 2) return lbk;
+  int cut = 10;  // per-lane for <= cut machines (GATS_LB2_LANE_MAX overrides)
+  if (const char* e = std::getenv("GATS_LB2_LANE_MAX")) cut = atoi(e);
+  return (machines <= cut) ? 3 : lbk;
 }
 
 int devpool_grid(unsigned long long M, int per, int lbk) {
