@@ -1108,7 +1108,10 @@ void launch_pfsp_eval(const PFSPNode* parents, int n, int jobs, int machines, in
 // short fully-unrolled register loop (ta005 20x5 measured 3x faster).
 int devpool_lbk_geom(int lbk, int machines) {
   if (lbk != 2) return lbk;
-  int cut = 10;  // per-lane for <= cut machines (GATS_LB2_LANE_MAX overrides)
+  // per-lane for m=5 (10 pairs: the wave kernel idles 54/64 lanes, ta005
+  // 31.5 -> 10.9 s); wave for m=10+ (45+ pairs: measured 25% faster on
+  // ta018 at m=10, and the collective exit dominates at m=20)
+  int cut = 5;
   if (const char* e = std::getenv("GATS_LB2_LANE_MAX")) cut = atoi(e);
   return (machines <= cut) ? 3 : lbk;
 }
