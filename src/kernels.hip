@@ -816,13 +816,6 @@ __device__ inline int wave_max_i32(int v) {
   return v;
 }
 
-// max over one 32-lane half of the wave (width=32 confines the shuffle)
-__device__ inline int half_max_i32(int v) {
-#pragma unroll
-  for (int d = 16; d >= 1; d >>= 1) v = max(v, __shfl_xor(v, d, 32));
-  return v;
-}
-
 template <int MM>
 __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* childbuf,
                               uint32_t* waveCounts, unsigned long long* waveSols, int jobs,
@@ -912,74 +905,66 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_wave_barrier();
 
-  // ---- phase B: TWO children per wave at a time (32 pair-lanes each) —
-  // halves the wave's serial child chain; with strength-ordered pairs most
-  // exits still fire in the first 32-pair round. Count/slab granularity is
-  // the HALF-wave (gw = wave*2 + half, 32-node slabs; devpool_grid/stride).
+  // ---- phase B: one child per wave at a time, pairs across lanes ----
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int half = lane >> 5;   // which child of the running pair
-  const int hlane = lane & 31;  // lane within the half-wave
   const int best = ctl->best;
   uint32_t mycnt = 0;
   unsigned long long mysols = 0;
-  const unsigned long long half_slab =
-      ((static_cast<unsigned long long>(blockIdx.x) * 4 + wid) * 2 + half) * 32ull;
+  const unsigned long long wave_slab =
+      (static_cast<unsigned long long>(blockIdx.x) * 4 + wid) * 64ull;
 
-  for (int cc = 0; cc < 64; cc += 2) {
-    const int ct = wid * 64 + cc + half;
-    const int8_t state = lds.ck[ct];  // same address across the half: broadcast
-    const bool active = (state != -1);
-    const bool leaf = active && state <= -2;
+  for (int cc = 0; cc < 64; cc++) {
+    const int ct = wid * 64 + cc;
+    const int8_t state = lds.ck[ct];  // same address across the wave: broadcast
+    if (state == -1) continue;
+    const bool leaf = state <= -2;
     const int k = leaf ? (-2 - state) : state;
     const uint16_t* fr = lds.fronts[ct];
     const int lp = lds.lpar[ct];
+    const uint32_t sched = lds.pmask[lp] | (1u << lds.snodes[lp].prmu[k]);
 
     int mylb = 0;
-    bool exited = !active;
-    if (active) {
-      const uint32_t sched = lds.pmask[lp] | (1u << lds.snodes[lp].prmu[k]);
-      constexpr int HROUNDS = (PAIRS + 31) / 32;
+    bool exited = false;
+    constexpr int ROUNDS = (PAIRS + 63) / 64;
 #pragma unroll
-      for (int r = 0; r < HROUNDS; r++) {
-        if (!exited) {
-          const int pr = hlane + r * 32;
-          if (pr < PAIRS) {
-            const int ma0 = lds.pair1[pr];
-            const int ma1 = lds.pair2[pr];
-            int t0 = fr[ma0];
-            int t1 = fr[ma1];
-            const uint32_t* jp = &lds.jp[pr * jobs];
-            for (int j = 0; j < jobs; j++) {
-              const uint32_t v = jp[j];
-              const int job = static_cast<int>(v >> 27);
-              if (!(sched >> job & 1u)) {
-                t0 += static_cast<int>(v & 0xffu);
-                t1 = max(t1, t0 + static_cast<int>((v >> 16) & 0x7ffu));
-                t1 += static_cast<int>((v >> 8) & 0xffu);
-              }
+    for (int r = 0; r < ROUNDS; r++) {
+      if (!exited) {
+        const int pr = lane + r * 64;
+        if (pr < PAIRS) {
+          const int ma0 = lds.pair1[pr];
+          const int ma1 = lds.pair2[pr];
+          int t0 = fr[ma0];
+          int t1 = fr[ma1];
+          const uint32_t* jp = &lds.jp[pr * jobs];
+          for (int j = 0; j < jobs; j++) {
+            const uint32_t v = jp[j];
+            const int job = static_cast<int>(v >> 27);
+            if (!(sched >> job & 1u)) {
+              t0 += static_cast<int>(v & 0xffu);
+              t1 = max(t1, t0 + static_cast<int>((v >> 16) & 0x7ffu));
+              t1 += static_cast<int>((v >> 8) & 0xffu);
             }
-            mylb = max(mylb, max(t1 + lds.min_tails[ma1], t0 + lds.min_tails[ma0]));
           }
-          if (half_max_i32(mylb) > best) exited = true;  // collective early exit
+          mylb = max(mylb, max(t1 + lds.min_tails[ma1], t0 + lds.min_tails[ma0]));
         }
+        if (wave_max_i32(mylb) > best) exited = true;  // collective early exit
       }
     }
-    const int lb = half_max_i32(mylb);
-    if (hlane == 0 && active) {
+    const int lb = wave_max_i32(mylb);
+    if (lane == 0) {
       if (leaf) {
         mysols++;
         if (!exited && lb < best) atomicMin(&ctl->best, lb);
       } else if (!exited && lb < best) {
-        const PFSPNode& p = lds.snodes[lp];
-        emit_pfsp_child(childbuf, half_slab + mycnt, p, p.depth, p.limit1, k);
+        const PFSPNode& p = lds.snodes[lds.lpar[ct]];
+        emit_pfsp_child(childbuf, wave_slab + mycnt, p, p.depth, p.limit1, k);
         mycnt++;
       }
     }
   }
-  if (hlane == 0) {
-    const unsigned long long gw =
-        (static_cast<unsigned long long>(blockIdx.x) * 4 + wid) * 2 + half;
+  if (lane == 0) {
+    const unsigned long long gw = static_cast<unsigned long long>(blockIdx.x) * 4 + wid;
     waveCounts[gw] = mycnt;
     waveSols[gw] = mysols;
   }
@@ -1134,16 +1119,15 @@ int devpool_lbk_geom(int lbk, int machines) {
 int devpool_grid(unsigned long long M, int per, int lbk) {
   if (lbk == 0)  // lb1_d: one thread per parent
     return static_cast<int>((M + BLOCK - 1) / BLOCK);
-  if (lbk == 2)  // wave-cooperative lb2: counts/slabs are per HALF-WAVE
-                 // (32 slots), padded to whole 4-wave blocks
-                 // (gw index = (blockIdx*4 + wid)*2 + half)
-    return static_cast<int>((M * per + BLOCK - 1) / BLOCK) * 8;
+  if (lbk == 2)  // wave-cooperative lb2: counts/slabs are per WAVE (64 slots),
+                 // padded to whole 4-wave blocks (gw index = blockIdx*4 + wid)
+    return static_cast<int>((M * per + BLOCK - 1) / BLOCK) * 4;
   return static_cast<int>((M * per + EMIT_TILE - 1) / EMIT_TILE);
 }
 
 int devpool_stride(int lbk) {
   if (lbk == 0) return BLOCK * MAX_JOBS;
-  if (lbk == 2) return 32;  // one half-wave's child slab
+  if (lbk == 2) return 64;  // one wave's child slab
   return EMIT_TILE;
 }
 
