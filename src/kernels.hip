@@ -792,9 +792,14 @@ __global__ void k_pfsp_x_lb1d(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
 template <int MM>
 struct LdsLb2w {
   static constexpr int PAIRS = MM * (MM - 1) / 2;
+  // only the ROUND-0 pair rows live in LDS: with the collective early exit
+  // and strength-ordered pairs most children never touch rounds 1-2, so
+  // those rows read from global (L2-resident, the table is ~15 KB) — the
+  // 10 KB of LDS saved lifts occupancy from 5 to 7 blocks/CU at MM=20
+  static constexpr int JP_LDS = PAIRS < 64 ? PAIRS : 64;
   int16_t p[MM * MAX_JOBS];
   int32_t min_tails[MM];
-  uint32_t jp[PAIRS * MAX_JOBS];
+  uint32_t jp[JP_LDS * MAX_JOBS];
   uint8_t pair1[PAIRS], pair2[PAIRS];
   uint16_t fronts[BLOCK][MM + 1];  // child completion times (values <= 20*20*99)
   uint8_t lpar[BLOCK];             // child's parent index in snodes
@@ -827,7 +832,7 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
   if (threadIdx.x < MM) lds.min_tails[threadIdx.x] = tb.min_tails[threadIdx.x];
   // strength-ordered tables: strongest pairs land in the first 64-pair round
   // so the collective early exit fires there (see PfspDevTables)
-  for (int i = threadIdx.x; i < PAIRS * jobs; i += blockDim.x)
+  for (int i = threadIdx.x; i < LdsLb2w<MM>::JP_LDS * jobs; i += blockDim.x)
     lds.jp[i] = tb.johnson_packed_w[i];
   if (threadIdx.x < PAIRS) {
     lds.pair1[threadIdx.x] = tb.pairs1_w[threadIdx.x];
@@ -936,7 +941,9 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
           const int ma1 = lds.pair2[pr];
           int t0 = fr[ma0];
           int t1 = fr[ma1];
-          const uint32_t* jp = &lds.jp[pr * jobs];
+          const uint32_t* jp = (pr < LdsLb2w<MM>::JP_LDS)
+                                   ? &lds.jp[pr * jobs]
+                                   : &tb.johnson_packed_w[pr * jobs];
           for (int j = 0; j < jobs; j++) {
             const uint32_t v = jp[j];
             const int job = static_cast<int>(v >> 27);
