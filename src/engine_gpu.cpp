@@ -11,9 +11,9 @@
 //       one HIP stream, host-side generate_children.
 //   "devpool": MI355X-native fast path — pools live in HBM3E; each offload
 //       round is an expand-compact + gather kernel pair (kernels.hip), with
-//       ~16 frontier slices pulled off a queue by 4 worker threads on
-//       concurrent streams to fill the chip; the host polls 48 B control
-//       blocks every few iterations.
+//       frontier slices pulled off a queue by a few worker threads on
+//       concurrent streams (see devpool_slices); the host polls 48 B
+//       control blocks every few iterations.
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
@@ -571,13 +571,13 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, const DevLoopCfg& c
 
 // ---------------------------------------------------------------------------
 // Multi-slice devpool: S independent device pools driven on S HIP streams by
-// S host threads fill the chip — ONE devpool loop is a serial expand->gather
-// dependency chain whose kernels leave the MI355X roughly half idle at
-// M = 50000 (measured: two ranks time-sharing one GPU nearly doubled
-// throughput). The frontier is round-robin split exactly like the multi-GPU
-// partition (nqueens_multigpu_chpl.chpl:221-226), just inside one device.
-// No stealing between slices; S = 4 keeps the device busy while stragglers
-// finish. Counts are slice-order independent.
+// S host threads — ONE devpool loop is a serial expand->gather dependency
+// chain, so concurrent chains hide its kernel-boundary bubbles (and, before
+// the wide-chunk change, filled the otherwise half-idle chip). The frontier
+// is round-robin split exactly like the multi-GPU partition
+// (nqueens_multigpu_chpl.chpl:221-226), just inside one device; a drained
+// thread takes donated half-pools (SliceShare below). Counts are
+// slice-order independent. S default: see devpool_slices().
 // ---------------------------------------------------------------------------
 
 // Concurrent slice chains per device. Wide-chunk paths (N-Queens, lb1,
