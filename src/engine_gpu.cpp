@@ -428,6 +428,7 @@ struct DevLoopCfg {
   unsigned long long capacity = 0;   // pool capacity in nodes
   unsigned long long stop_size = 0;  // frontier builder: stop once size >= this
   int kernels_per_iter = 2;
+  int per = 1;                       // max children per node (grid-bound growth)
   bool allow_graph = true;
 };
 
@@ -455,6 +456,11 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, const DevLoopCfg& c
   int batches = 0;
   int par = 0;  // parity of the NEXT iteration == index of the live ctl block
   unsigned long long last_size = cfg.init_size;
+  // chunk BOUND handed to each launch so its grid covers only the pool that
+  // can possibly exist: exact at readbacks, multiplied by the branching
+  // factor per blind iteration. Full-size grids for tiny chunks were the
+  // small-search floor (an 80-node ta019 tree dispatched 15k-block lb2
+  // grids: 1.53 ms; the engine infra itself costs 0.025 ms).
   hipGraph_t graph = nullptr;
   hipGraphExec_t exec = nullptr;
   bool overflow = false;
@@ -462,8 +468,11 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, const DevLoopCfg& c
   ctl_h.p->size = cfg.init_size;
   while (true) {
     // capacity-aware batch: never launch more iterations than worst-case
-    // growth allows; spill when even one iteration might not fit
+    // growth allows; spill when even one iteration might not fit. Small
+    // pools use short batches so the chunk bound (and with it the launch
+    // grids) is re-tightened every couple of iterations.
     int b = (cfg.stop_size > 0) ? 1 : BATCH;
+    if (cfg.per > 1 && last_size < 4096 && b > 2) b = 2;
     if (cfg.growth > 0 && cfg.capacity > 0) {
       const unsigned long long room =
           cfg.capacity > last_size ? cfg.capacity - last_size : 0;
@@ -492,7 +501,7 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, const DevLoopCfg& c
       std::lock_guard<std::mutex> lock(capture_mu);
       hipError_t ce = hipStreamBeginCapture(s, hipStreamCaptureModeRelaxed);
       if (ce == hipSuccess) {
-        for (int i = 0; i < BATCH; i++) enqueue_iter(i & 1);
+        for (int i = 0; i < BATCH; i++) enqueue_iter(i & 1, ~0ull);
         ce = hipStreamEndCapture(s, &graph);
         if (ce == hipSuccess) {
           if (hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0) != hipSuccess) {
@@ -513,9 +522,11 @@ static DevCtl run_devpool_loop(hipStream_t s, DevCtl* ctl_d, const DevLoopCfg& c
     if (exec != nullptr && b == BATCH && par == 0) {
       HIP_CHECK(hipGraphLaunch(exec, s));  // even count: parity unchanged
     } else {
+      unsigned long long bound = last_size ? last_size : 1;
       for (int i = 0; i < b; i++) {
-        enqueue_iter(par);
+        enqueue_iter(par, bound);
         par ^= 1;
+        if (bound < (1ull << 40)) bound *= cfg.per > 1 ? cfg.per : 2;
       }
     }
     batches++;
@@ -776,14 +787,19 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
   DevGuard<uint32_t> gsum_d(presum ? (G + 255) / 256 : 1);
   std::vector<NQNode> spilled;  // capacity-pressure spill, re-run after the slice
 
-  auto iter = [&](int parity) {
+  auto iter = [&](int parity, unsigned long long bound) {
+    // grid covers only the chunk that can exist (bound >= pool size, so
+    // min(size, Mi) == min(size, Mc): the pop rule is unchanged)
+    const unsigned long long Mi = std::min(Mc, std::max<unsigned long long>(bound, 1));
+    const int Gi = devpool_grid(Mi, N, 1);
+    const bool ps = presum && Gi > 256;
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
-    launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, m, Mc,
+    launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, m, Mi,
                 stream.s);
-    if (presum) launch_presum(bc_d.p, gsum_d.p, G, stream.s);
-    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, presum ? gsum_d.p : nullptr,
-                      childbuf_d.p, pool_d.p, stride, G, m, Mc, capacity, stream.s);
+    if (ps) launch_presum(bc_d.p, gsum_d.p, Gi, stream.s);
+    launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, ps ? gsum_d.p : nullptr,
+                      childbuf_d.p, pool_d.p, stride, Gi, m, Mi, capacity, stream.s);
   };
   ReadbackHook hook = [&](DevCtl* hc, DevCtl* live) {
     donate_if_wanted(share, hc, live, pool_d.p, m, stream.s);
@@ -808,6 +824,7 @@ static SliceOut devpool_thread_nq(const std::vector<std::vector<NQNode>>& slices
   cfg.m = m;
   cfg.capacity = capacity;
   cfg.growth = Mc * N;  // worst-case children/iter
+  cfg.per = N;
   cfg.allow_graph = allow_graph;
 
   auto run_pool = [&](unsigned long long init_size) {
@@ -911,14 +928,17 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
 
   std::vector<PFSPNode> spilled;  // capacity-pressure spill, re-run after the slice
 
-  auto iter = [&](int parity) {
+  auto iter = [&](int parity, unsigned long long bound) {
+    const unsigned long long Mi = std::min(Mc, std::max<unsigned long long>(bound, 1));
+    const int Gi = devpool_grid(Mi, jobs, lbg);
+    const bool ps = presum && (lbg == 2 || Gi > 256);
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
     launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbg,
-                  tb, m, Mc, stream.s);
-    if (presum) launch_presum(bc_d.p, gsum_d.p, G, stream.s);
-    launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, presum ? gsum_d.p : nullptr,
-                        childbuf_d.p, pool_d.p, stride, G, m, Mc, capacity, stream.s);
+                  tb, m, Mi, stream.s);
+    if (ps) launch_presum(bc_d.p, gsum_d.p, Gi, stream.s);
+    launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, ps ? gsum_d.p : nullptr,
+                        childbuf_d.p, pool_d.p, stride, Gi, m, Mi, capacity, stream.s);
   };
   ReadbackHook hook = [&](DevCtl* hc, DevCtl* live) {
     donate_if_wanted(share, hc, live, pool_d.p, m, stream.s);
@@ -975,6 +995,7 @@ static SliceOut devpool_thread_pfsp(const std::vector<std::vector<PFSPNode>>& sl
   cfg.m = m;
   cfg.capacity = capacity;
   cfg.growth = Mc * jobs;
+  cfg.per = jobs;
   cfg.allow_graph = allow_graph;
 
   auto run_pool = [&](unsigned long long init_size, int init_best) {
@@ -1479,18 +1500,22 @@ std::vector<NQNode> nq_gpu_frontier(int N, int g, size_t target, int device,
   HIP_CHECK(hipMemcpyAsync(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
   HIP_CHECK(hipStreamSynchronize(stream.s));
 
-  auto iter = [&](int parity) {
+  auto iter = [&](int parity, unsigned long long bound) {
+    const unsigned long long Mi =
+        std::min<unsigned long long>(M, std::max<unsigned long long>(bound, 1));
+    const int Gi = devpool_grid(Mi, N, 1);
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
-    launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, 1, M,
+    launch_nq_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, be_d.p, N, g, finish, 1, Mi,
                 stream.s);
     launch_gather2_nq(cur, next, bc_d.p, bs_d.p, be_d.p, nullptr, childbuf_d.p, pool_d.p,
-                      stride, G, 1, M, capacity, stream.s);
+                      stride, Gi, 1, Mi, capacity, stream.s);
   };
   DevLoopCfg cfg;
   cfg.m = 1;
   cfg.init_size = 1;
   cfg.stop_size = target;
+  cfg.per = N;
   cfg.allow_graph = false;
   const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, cfg, iter, r);
   tree = fin.tree;
@@ -1539,19 +1564,24 @@ std::vector<PFSPNode> pfsp_gpu_frontier(const PfspInstance& I, int lbk, size_t t
   HIP_CHECK(hipMemcpyAsync(ctl_d.p + 1, &ctl, sizeof(DevCtl), hipMemcpyHostToDevice, stream.s));
   HIP_CHECK(hipStreamSynchronize(stream.s));
 
-  auto iter = [&](int parity) {
+  auto iter = [&](int parity, unsigned long long bound) {
+    const unsigned long long Mi =
+        std::min<unsigned long long>(M, std::max<unsigned long long>(bound, 1));
+    const int Gi = devpool_grid(Mi, jobs, lbg);
+    const bool ps = presum && (lbg == 2 || Gi > 256);
     DevCtl* cur = ctl_d.p + parity;
     DevCtl* next = ctl_d.p + (1 - parity);
     launch_pfsp_x(cur, pool_d.p, childbuf_d.p, bc_d.p, bs_d.p, jobs, machines, lbg, tb, 1,
-                  M, stream.s);
-    if (presum) launch_presum(bc_d.p, gsum_d.p, G, stream.s);
-    launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, presum ? gsum_d.p : nullptr,
-                        childbuf_d.p, pool_d.p, stride, G, 1, M, capacity, stream.s);
+                  Mi, stream.s);
+    if (ps) launch_presum(bc_d.p, gsum_d.p, Gi, stream.s);
+    launch_gather2_pfsp(cur, next, bc_d.p, bs_d.p, ps ? gsum_d.p : nullptr,
+                        childbuf_d.p, pool_d.p, stride, Gi, 1, Mi, capacity, stream.s);
   };
   DevLoopCfg cfg;
   cfg.m = 1;
   cfg.init_size = 1;
   cfg.stop_size = target;
+  cfg.per = jobs;
   cfg.allow_graph = false;
   const DevCtl fin = run_devpool_loop(stream.s, ctl_d.p, cfg, iter, r);
   tree = fin.tree;
