@@ -143,6 +143,14 @@ def test_async_engine_multiple_submits(gpu):
     assert r["optimum"] == 1377
 
 
+def test_nqueens_n18_frozen_counts(gpu):
+    # deep-tree regression at the wide expansion chunk (~0.6 s on MI355X);
+    # tree count frozen in BASELINE.md, solution count is the public value
+    r = gpu.nqueens_gpu(18, 1, 25, 50000, 0, "devpool", 1 << 27)
+    assert r["sol"] == 666090624
+    assert r["tree"] == 59365844490
+
+
 def test_nqueens_rooted_counts(gpu):
     seq = gpu.nqueens_seq(14, 1)
     r = gpu.nqueens_gpu_rooted(14, 1, 50000, 0, 1 << 26)
