@@ -407,10 +407,14 @@ int lbk_of(LbKind lb) {
 // the engine's best into it (atomic min) and adopts a lower value published by
 // other ranks/threads by writing ctl->best on the (synchronized) stream —
 // the RCCL incumbent-UB exchange of the distributed tier plugs in here.
-// enqueue_iter(parity): parity alternates 0/1 per iteration — iteration i
-// reads ctl[parity] and gather2 writes ctl[1-parity]. The loop tracks the
-// live parity so hooks and readbacks always touch the block the last gather
-// wrote.
+// enqueue_iter(parity, chunk_bound): parity alternates 0/1 per iteration —
+// iteration i reads ctl[parity] and gather2 writes ctl[1-parity]; the loop
+// tracks the live parity so hooks and readbacks always touch the block the
+// last gather wrote. chunk_bound is an upper bound on the pool size at that
+// iteration (exact at readbacks, x branching per blind iteration) so the
+// callee can size its launch grids to work that can actually exist; pop
+// semantics are unchanged because bound >= size implies
+// min(size, min(Mc, bound)) == min(size, Mc).
 // `readback_hook(host_ctl, live_ctl_d)`: called after every batch readback
 // with the stream idle; may reduce host_ctl->size after carving the pool (the
 // donor path of SliceShare) — it must then also write the device ctl itself.
