@@ -944,13 +944,29 @@ __global__ void k_pfsp_x_lb2w(DevCtl* ctl, const PFSPNode* pool, PFSPNode* child
           const uint32_t* jp = (pr < LdsLb2w<MM>::JP_LDS)
                                    ? &lds.jp[pr * jobs]
                                    : &tb.johnson_packed_w[pr * jobs];
-          for (int j = 0; j < jobs; j++) {
-            const uint32_t v = jp[j];
-            const int job = static_cast<int>(v >> 27);
-            if (!(sched >> job & 1u)) {
-              t0 += static_cast<int>(v & 0xffu);
-              t1 = max(t1, t0 + static_cast<int>((v >> 16) & 0x7ffu));
-              t1 += static_cast<int>((v >> 8) & 0xffu);
+          // jobs == MAX_JOBS for every GPU-supported Taillard instance:
+          // full unrolling lets the compiler batch the 20 dependent
+          // ds_read_b32s instead of serializing load->use 20 times
+          if (jobs == MAX_JOBS) {
+#pragma unroll
+            for (int j = 0; j < MAX_JOBS; j++) {
+              const uint32_t v = jp[j];
+              const int job = static_cast<int>(v >> 27);
+              if (!(sched >> job & 1u)) {
+                t0 += static_cast<int>(v & 0xffu);
+                t1 = max(t1, t0 + static_cast<int>((v >> 16) & 0x7ffu));
+                t1 += static_cast<int>((v >> 8) & 0xffu);
+              }
+            }
+          } else {
+            for (int j = 0; j < jobs; j++) {
+              const uint32_t v = jp[j];
+              const int job = static_cast<int>(v >> 27);
+              if (!(sched >> job & 1u)) {
+                t0 += static_cast<int>(v & 0xffu);
+                t1 = max(t1, t0 + static_cast<int>((v >> 16) & 0x7ffu));
+                t1 += static_cast<int>((v >> 8) & 0xffu);
+              }
             }
           }
           mylb = max(mylb, max(t1 + lds.min_tails[ma1], t0 + lds.min_tails[ma0]));
