@@ -318,20 +318,42 @@ __device__ inline int lb2_child_bound(const LdsLb2<MM>& lds, const uint8_t* prmu
     int t0b = front[ma0b], t1b = front[ma1b];
     const uint32_t* jpa = &lds.jp[l * jobs];
     const uint32_t* jpb = &lds.jp[(two ? l + 1 : l) * jobs];
-    for (int j = 0; j < jobs; j++) {
-      const uint32_t va = jpa[j];  // one ds_read_b32 replaces 4 scalar LDS reads
-      const uint32_t vb = jpb[j];
-      const int ja = static_cast<int>(va >> 27);
-      const int jb = static_cast<int>(vb >> 27);
-      if (!(scheduled >> ja & 1u)) {
-        t0a += static_cast<int>(va & 0xffu);
-        t1a = max(t1a, t0a + static_cast<int>((va >> 16) & 0x7ffu));
-        t1a += static_cast<int>((va >> 8) & 0xffu);
+    // jobs == MAX_JOBS for every GPU-supported instance: full unrolling
+    // batches the dependent ds_read_b32s (see the wave kernel's note)
+    if (jobs == MAX_JOBS) {
+#pragma unroll
+      for (int j = 0; j < MAX_JOBS; j++) {
+        const uint32_t va = jpa[j];  // one ds_read_b32 replaces 4 scalar LDS reads
+        const uint32_t vb = jpb[j];
+        const int ja = static_cast<int>(va >> 27);
+        const int jb = static_cast<int>(vb >> 27);
+        if (!(scheduled >> ja & 1u)) {
+          t0a += static_cast<int>(va & 0xffu);
+          t1a = max(t1a, t0a + static_cast<int>((va >> 16) & 0x7ffu));
+          t1a += static_cast<int>((va >> 8) & 0xffu);
+        }
+        if (!(scheduled >> jb & 1u)) {
+          t0b += static_cast<int>(vb & 0xffu);
+          t1b = max(t1b, t0b + static_cast<int>((vb >> 16) & 0x7ffu));
+          t1b += static_cast<int>((vb >> 8) & 0xffu);
+        }
       }
-      if (!(scheduled >> jb & 1u)) {
-        t0b += static_cast<int>(vb & 0xffu);
-        t1b = max(t1b, t0b + static_cast<int>((vb >> 16) & 0x7ffu));
-        t1b += static_cast<int>((vb >> 8) & 0xffu);
+    } else {
+      for (int j = 0; j < jobs; j++) {
+        const uint32_t va = jpa[j];
+        const uint32_t vb = jpb[j];
+        const int ja = static_cast<int>(va >> 27);
+        const int jb = static_cast<int>(vb >> 27);
+        if (!(scheduled >> ja & 1u)) {
+          t0a += static_cast<int>(va & 0xffu);
+          t1a = max(t1a, t0a + static_cast<int>((va >> 16) & 0x7ffu));
+          t1a += static_cast<int>((va >> 8) & 0xffu);
+        }
+        if (!(scheduled >> jb & 1u)) {
+          t0b += static_cast<int>(vb & 0xffu);
+          t1b = max(t1b, t0b + static_cast<int>((vb >> 16) & 0x7ffu));
+          t1b += static_cast<int>((vb >> 8) & 0xffu);
+        }
       }
     }
     // merge pair a, check, then pair b: keeps the returned value bit-equal to
