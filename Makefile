@@ -14,7 +14,7 @@ test-gpu: build
 	python -m pytest tests -q -m gpu
 
 bench: build
-	python bench.py --gpus 1 --steps 5 --warmup 2
+	python bench.py --gpus 1 --steps 25 --warmup 3
 
 repro: build
 	bash scripts/reproduce.sh quick

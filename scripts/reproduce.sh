@@ -20,7 +20,7 @@ python bench.py --gpus 1 --steps 25 --warmup 5 --problem pfsp --inst 14 --lb lb1
 if [ "$MODE" = "full" ]; then
   echo "== N-Queens N=18 (~0.6 s) =="
   python -m gats_amd.cli nqueens --N 18 --tier gpu
-  echo "== PFSP ta001-ta020 lb2 sweep, optima proven from ub=1 (~31 s total) =="
+  echo "== PFSP ta001-ta020 lb2 sweep, optima proven from ub=1 (~15-20 s total via the CLI) =="
   for i in $(seq 1 20); do
     python -m gats_amd.cli pfsp --inst "$i" --lb lb2 --ub 1 --tier gpu \
         --capacity $((1 << 25)) --stats-file sweep_lb2.dat
