@@ -7,6 +7,7 @@
 #include <stdexcept>
 
 #include "engine_gpu.hpp"
+#include "gpu_api.hpp"
 #include "nodes.hpp"
 #include "pool.hpp"
 #include "search_host.hpp"
@@ -428,6 +429,11 @@ PYBIND11_MODULE(_core, mod) {
           py::arg("device") = 0);
 
   mod.def("pool_selftest", &pool_selftest);
+  // pure policy helpers (CPU-unit-testable): kernel geometry selection and
+  // the devpool expansion-chunk clamp
+  mod.def("devpool_lbk_geom", &devpool_lbk_geom, py::arg("lbk"), py::arg("machines"));
+  mod.def("devpool_grid", &devpool_grid, py::arg("M"), py::arg("per"), py::arg("lbk"));
+  mod.def("devpool_stride", &devpool_stride, py::arg("lbk"));
 
   py::class_<PfspAsyncEngine>(mod, "PfspAsyncEngine",
                               "Persistent per-rank PFSP devpool engine: successive "
