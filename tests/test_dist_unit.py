@@ -22,3 +22,17 @@ def test_slice_frontier_world_larger_than_nodes():
     nodes = bytes(NODE_BYTES)  # one node
     assert slice_frontier(nodes, 0, 4) == nodes
     assert slice_frontier(nodes, 3, 4) == b""
+
+
+def test_frontier_cache_memoizes_per_key():
+    from gats_amd import dist
+
+    calls = []
+    dist._frontier_cache.clear()
+    out1 = dist._cached_frontier(("k", 1), lambda: calls.append(1) or (b"a", 1, 2))
+    out2 = dist._cached_frontier(("k", 1), lambda: calls.append(1) or (b"b", 9, 9))
+    out3 = dist._cached_frontier(("k", 2), lambda: calls.append(1) or (b"c", 3, 4))
+    assert out1 == out2 == (b"a", 1, 2)  # second build never ran
+    assert out3 == (b"c", 3, 4)
+    assert len(calls) == 2
+    dist._frontier_cache.clear()
